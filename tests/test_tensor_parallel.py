@@ -1,0 +1,154 @@
+"""TP numerics on gloo world_size=2: parallel layers vs plain nn.Linear,
+vocab-parallel CE vs dense CE, broadcast_data."""
+import pytest
+import torch
+import torch.nn as nn
+
+from tests.distributed_utils import run_distributed
+
+
+def _tp_linear_worker(rank, world_size):
+    import torch.distributed as dist
+    from fengshen_amd.parallel.groups import (
+        init_distributed, initialize_model_parallel)
+    from fengshen_amd.parallel.layers import (
+        ColumnParallelLinear, RowParallelLinear, VocabParallelEmbedding)
+
+    init_distributed(backend="gloo")
+    initialize_model_parallel(tensor_model_parallel_size=world_size)
+    torch.manual_seed(5)
+    # reference dense layer
+    ref_col = nn.Linear(16, 32, bias=True)
+    ref_row = nn.Linear(32, 16, bias=True)
+    ref_emb = nn.Embedding(50, 16)
+    x = torch.randn(4, 7, 16)
+    ids = torch.randint(0, 50, (4, 7))
+
+    col = ColumnParallelLinear(16, 32, bias=True, gather_output=True,
+                               dtype=torch.float32)
+    row = RowParallelLinear(32, 16, bias=True, input_is_parallel=False,
+                            dtype=torch.float32)
+    emb = VocabParallelEmbedding(50, 16, dtype=torch.float32)
+    # copy reference weights into the shards
+    out_per = 32 // world_size
+    in_per = 32 // world_size
+    with torch.no_grad():
+        col.weight.copy_(ref_col.weight[rank * out_per:(rank + 1) * out_per])
+        col.bias.copy_(ref_col.bias[rank * out_per:(rank + 1) * out_per])
+        row.weight.copy_(ref_row.weight[:, rank * in_per:(rank + 1) * in_per])
+        row.bias.copy_(ref_row.bias)
+        vper = 50 // world_size
+        emb.weight.copy_(ref_emb.weight[rank * vper:(rank + 1) * vper])
+
+    y_col = col(x)
+    y_ref_col = ref_col(x)
+    col_ok = torch.allclose(y_col, y_ref_col, atol=1e-5)
+
+    h = torch.randn(4, 7, 32)
+    y_row = row(h)
+    y_ref_row = ref_row(h)
+    row_ok = torch.allclose(y_row, y_ref_row, atol=1e-5)
+
+    y_emb = emb(ids)
+    emb_ok = torch.allclose(y_emb, ref_emb(ids), atol=1e-5)
+
+    # backward through col: grads of input must match dense
+    x1 = x.clone().requires_grad_(True)
+    x2 = x.clone().requires_grad_(True)
+    col(x1).sum().backward()
+    ref_col(x2).sum().backward()
+    grad_ok = torch.allclose(x1.grad, x2.grad, atol=1e-5)
+
+    dist.destroy_process_group()
+    return dict(col=col_ok, row=row_ok, emb=emb_ok, grad=grad_ok)
+
+
+def test_tp_layers_match_dense():
+    results = run_distributed(_tp_linear_worker, world_size=2)
+    for r in results:
+        assert all(r.values()), r
+
+
+def _vocab_ce_worker(rank, world_size):
+    import torch.distributed as dist
+    from fengshen_amd.parallel.groups import (
+        init_distributed, initialize_model_parallel)
+    from fengshen_amd.parallel.cross_entropy import vocab_parallel_cross_entropy
+
+    init_distributed(backend="gloo")
+    initialize_model_parallel(tensor_model_parallel_size=world_size)
+    torch.manual_seed(9)
+    b, s, v = 3, 5, 64
+    logits = torch.randn(b, s, v)
+    target = torch.randint(0, v, (b, s))
+    vper = v // world_size
+    shard = logits[:, :, rank * vper:(rank + 1) * vper].clone().requires_grad_(True)
+    loss = vocab_parallel_cross_entropy(shard, target)
+    ref = torch.nn.functional.cross_entropy(
+        logits.view(-1, v), target.view(-1), reduction="none").view(b, s)
+    fwd_ok = torch.allclose(loss, ref, atol=1e-5)
+
+    loss.mean().backward()
+    full = logits.clone().requires_grad_(True)
+    torch.nn.functional.cross_entropy(
+        full.view(-1, v), target.view(-1)).backward()
+    ref_grad_shard = full.grad[:, :, rank * vper:(rank + 1) * vper]
+    bwd_ok = torch.allclose(shard.grad, ref_grad_shard, atol=1e-5)
+    dist.destroy_process_group()
+    return dict(fwd=fwd_ok, bwd=bwd_ok)
+
+
+def test_vocab_parallel_cross_entropy():
+    results = run_distributed(_vocab_ce_worker, world_size=2)
+    for r in results:
+        assert r["fwd"] and r["bwd"], r
+
+
+def _broadcast_worker(rank, world_size):
+    import torch.distributed as dist
+    from fengshen_amd.parallel.groups import (
+        init_distributed, initialize_model_parallel)
+    from fengshen_amd.parallel.data import broadcast_data
+
+    init_distributed(backend="gloo")
+    initialize_model_parallel(tensor_model_parallel_size=world_size)
+    if rank == 0:
+        data = {"input_ids": torch.arange(12).view(3, 4),
+                "labels": torch.ones(3, 4, dtype=torch.long)}
+    else:
+        data = {}  # non-src ranks have no data
+    out = broadcast_data(["input_ids", "labels"], data, torch.long)
+    ok = torch.equal(out["input_ids"].cpu(), torch.arange(12).view(3, 4))
+    dist.destroy_process_group()
+    return ok
+
+
+def test_broadcast_data():
+    assert all(run_distributed(_broadcast_worker, world_size=2))
+
+
+def _mappings_worker(rank, world_size):
+    import torch.distributed as dist
+    from fengshen_amd.parallel.groups import (
+        init_distributed, initialize_model_parallel)
+    from fengshen_amd.parallel import mappings
+
+    init_distributed(backend="gloo")
+    initialize_model_parallel(tensor_model_parallel_size=world_size)
+    x = torch.full((2, 3), float(rank + 1))
+    red = mappings.reduce_from_tensor_model_parallel_region(x.clone())
+    reduce_ok = torch.allclose(red, torch.full((2, 3), 3.0))  # 1+2
+
+    y = torch.full((2, 2), float(rank))
+    gat = mappings.gather_from_tensor_model_parallel_region(y)
+    gather_ok = gat.shape == (2, 4) and gat[0, 0] == 0 and gat[0, 3] == 1
+
+    z = torch.arange(8.0).view(2, 4)
+    sc = mappings.scatter_to_tensor_model_parallel_region(z)
+    scatter_ok = sc.shape == (2, 2) and sc[0, 0] == (0.0 if rank == 0 else 2.0)
+    dist.destroy_process_group()
+    return reduce_ok and gather_ok and scatter_ok
+
+
+def test_region_mappings():
+    assert all(run_distributed(_mappings_worker, world_size=2))

@@ -1,0 +1,186 @@
+"""End-to-end CPU tests of the trainer stack (module -> trainer -> optimizer
+-> checkpoint/resume) on a toy regression problem."""
+import argparse
+import math
+import os
+
+import pytest
+import torch
+import torch.nn as nn
+from torch.utils.data import DataLoader, Dataset
+
+from fengshen_amd import FengshenModule, Trainer
+from fengshen_amd.models.model_utils import (
+    add_module_args,
+    configure_optimizers,
+    get_scheduler,
+)
+from fengshen_amd.utils.universal_checkpoint import UniversalCheckpoint
+
+
+class ToyDataset(Dataset):
+    def __init__(self, n=256, d=8, seed=0):
+        g = torch.Generator().manual_seed(seed)
+        self.x = torch.randn(n, d, generator=g)
+        w = torch.arange(1, d + 1, dtype=torch.float32)
+        self.y = self.x @ w
+
+    def __len__(self):
+        return len(self.x)
+
+    def __getitem__(self, i):
+        return {"x": self.x[i], "y": self.y[i]}
+
+
+class ToyModule(FengshenModule):
+    def __init__(self, args, d=8):
+        super().__init__()
+        self.save_hyperparameters(args)
+        self.net = nn.Sequential(nn.Linear(d, 32), nn.Tanh(), nn.Linear(32, 1))
+        self.losses = []
+
+    def training_step(self, batch, batch_idx):
+        pred = self.net(batch["x"]).squeeze(-1)
+        loss = torch.nn.functional.mse_loss(pred, batch["y"])
+        self.log("train_loss", loss)
+        self.losses.append(loss.item())
+        return loss
+
+    def validation_step(self, batch, batch_idx):
+        pred = self.net(batch["x"]).squeeze(-1)
+        loss = torch.nn.functional.mse_loss(pred, batch["y"])
+        self.log("val_loss", loss, sync_dist=True)
+        return loss
+
+    def configure_optimizers(self):
+        return configure_optimizers(self)
+
+
+def _make_args(**over):
+    parser = argparse.ArgumentParser()
+    add_module_args(parser)
+    args = parser.parse_args([])
+    args.learning_rate = 1e-2
+    args.warmup_steps = 2
+    for k, v in over.items():
+        setattr(args, k, v)
+    return args
+
+
+def test_fit_loss_decreases(tmp_path):
+    args = _make_args()
+    model = ToyModule(args)
+    loader = DataLoader(ToyDataset(), batch_size=32, shuffle=True)
+    trainer = Trainer(max_steps=60, precision="fp32",
+                      default_root_dir=str(tmp_path), log_every_n_steps=20)
+    trainer.fit(model, train_dataloaders=loader)
+    assert trainer.global_step == 60
+    first = sum(model.losses[:5]) / 5
+    last = sum(model.losses[-5:]) / 5
+    assert last < first * 0.5, f"loss did not decrease: {first} -> {last}"
+
+
+def test_grad_accumulation_counts(tmp_path):
+    args = _make_args()
+    model = ToyModule(args)
+    loader = DataLoader(ToyDataset(n=64), batch_size=8)
+    trainer = Trainer(max_steps=4, precision="fp32", accumulate_grad_batches=2,
+                      default_root_dir=str(tmp_path))
+    trainer.fit(model, train_dataloaders=loader)
+    assert trainer.global_step == 4
+    # 4 optimizer steps * 2 micro * 8 samples
+    assert trainer.global_samples == 64
+
+
+def _toy_datamodule(args, n=512):
+    from fengshen_amd import UniversalDataModule
+    args.train_batchsize = 16
+    args.val_batchsize = 16
+    args.sampler_type = "single"  # sequential => deterministic order
+    args.num_workers = 0
+    args.train_datasets_field = "train"
+    args.val_datasets_field = "validation"
+    return UniversalDataModule(
+        tokenizer=None, collate_fn=None, args=args,
+        datasets={"train": ToyDataset(n=n)})
+
+
+def test_checkpoint_resume_exact(tmp_path):
+    # run 1: 10 steps uninterrupted; run 2: 5 steps -> ckpt -> resume 5 more.
+    # exact data-order resume comes from consumed_samples + PretrainingSampler.
+    args = _make_args()
+    args.lr_decay_steps = 10  # fix schedule horizon across the interrupted run
+    torch.manual_seed(7)
+    model_a = ToyModule(args)
+    tr_a = Trainer(max_steps=10, precision="fp32",
+                   default_root_dir=str(tmp_path / "a"))
+    tr_a.fit(model_a, datamodule=_toy_datamodule(args))
+
+    torch.manual_seed(7)
+    model_b = ToyModule(args)
+    tr_b = Trainer(max_steps=5, precision="fp32",
+                   default_root_dir=str(tmp_path / "b"))
+    tr_b.fit(model_b, datamodule=_toy_datamodule(args))
+    ckpt = str(tmp_path / "ckpt5")
+    tr_b.save_checkpoint(ckpt)
+
+    model_c = ToyModule(args)
+    tr_c = Trainer(max_steps=10, precision="fp32",
+                   default_root_dir=str(tmp_path / "c"))
+    tr_c.fit(model_c, datamodule=_toy_datamodule(args), ckpt_path=ckpt)
+    assert tr_c.global_step == 10
+
+    # same final weights as the uninterrupted run (same data order: sequential
+    # 512-sample dataset consumed deterministically per fit call)
+    for pa, pc in zip(model_a.parameters(), model_c.parameters()):
+        assert torch.allclose(pa, pc, atol=1e-5), "resume diverged"
+
+
+def test_universal_checkpoint_callback(tmp_path):
+    args = _make_args()
+    args.save_ckpt_path = str(tmp_path / "ckpts")
+    args.every_n_train_steps = 5
+    args.save_top_k = 2
+    args.save_last = False
+    args.monitor = "step"
+    args.mode = "max"
+    args.filename = "model-{step:02d}"
+    args.save_weights_only = False
+    args.every_n_epochs = None
+    cb = UniversalCheckpoint(args)
+    model = ToyModule(args)
+    loader = DataLoader(ToyDataset(n=640), batch_size=16)
+    trainer = Trainer(max_steps=20, precision="fp32", callbacks=[cb],
+                      default_root_dir=str(tmp_path))
+    trainer.fit(model, train_dataloaders=loader)
+    saved = sorted(os.listdir(args.save_ckpt_path))
+    assert len(saved) == 2  # top-k pruning
+    assert "model-15.ckpt" in saved and "model-20.ckpt" in saved
+
+
+def test_schedulers_shapes():
+    m = nn.Linear(2, 2)
+    opt = torch.optim.AdamW(m.parameters(), lr=1.0)
+    for name in ["constant", "constant_with_warmup", "linear", "cosine",
+                 "polynomial", "inverse_sqrt", "direct"]:
+        sch = get_scheduler(name, opt, num_warmup_steps=5,
+                            num_training_steps=20, lr_init=1.0, lr_end=0.1)
+        lrs = []
+        for _ in range(25):
+            lrs.append(opt.param_groups[0]["lr"])
+            opt.step()
+            sch.step()
+        assert all(not math.isnan(x) for x in lrs), name
+        if name != "constant":
+            assert lrs[1] < lrs[5], f"{name}: warmup missing"
+
+
+def test_validation_loop(tmp_path):
+    args = _make_args()
+    model = ToyModule(args)
+    train = DataLoader(ToyDataset(n=128), batch_size=16)
+    val = DataLoader(ToyDataset(n=64, seed=1), batch_size=16)
+    trainer = Trainer(max_steps=8, precision="fp32", val_check_interval=4,
+                      default_root_dir=str(tmp_path))
+    trainer.fit(model, train_dataloaders=train, val_dataloaders=val)
+    assert "val_loss" in trainer._metrics
