@@ -1,0 +1,136 @@
+"""Model-zoo numerics on CPU: forward/backward/generate for the flagship
+models, plus TP=2 parity vs TP=1 via shard_state_dict (gloo)."""
+import os
+
+import pytest
+import torch
+
+from tests.distributed_utils import run_distributed
+
+
+def _ids(vocab=256, b=2, s=16, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    return torch.randint(3, vocab, (b, s), generator=g)
+
+
+def test_gpt2_forward_backward_generate():
+    from fengshen_amd.models.gpt2.configuration_gpt2 import gpt2_tiny_config
+    from fengshen_amd.models.gpt2.modeling_gpt2 import GPT2LMHeadModel
+    torch.manual_seed(0)
+    m = GPT2LMHeadModel(gpt2_tiny_config())
+    ids = _ids()
+    out = m(ids, labels=ids)
+    assert out.loss.isfinite()
+    out.loss.backward()
+    assert m.transformer.wte.weight.grad is not None
+    m.eval()
+    gen = m.generate(ids[:, :4], max_new_tokens=6, do_sample=False)
+    assert gen.shape[1] == 10
+
+
+def test_gpt2_cache_consistency():
+    from fengshen_amd.models.gpt2.configuration_gpt2 import gpt2_tiny_config
+    from fengshen_amd.models.gpt2.modeling_gpt2 import GPT2LMHeadModel
+    from transformers.cache_utils import DynamicCache
+    torch.manual_seed(0)
+    m = GPT2LMHeadModel(gpt2_tiny_config()).eval()
+    ids = _ids()
+    with torch.no_grad():
+        full = m(ids).logits
+        c = DynamicCache()
+        m(ids[:, :8], use_cache=True, past_key_values=c)
+        o2 = m(ids[:, 8:9], past_key_values=c, use_cache=True)
+    assert torch.allclose(full[:, 8], o2.logits[:, 0], atol=1e-4)
+
+
+def test_megatron_bert_pretraining():
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.models.megatron_bert.modeling_megatron_bert import (
+        MegatronBertForPreTraining)
+    torch.manual_seed(0)
+    m = MegatronBertForPreTraining(bert_tiny_config())
+    ids = _ids()
+    labels = ids.clone()
+    labels[:, ::3] = -100  # unmasked positions ignored
+    mask = torch.ones_like(ids)
+    mask[:, -3:] = 0
+    sop = torch.randint(0, 2, (2,))
+    out = m(ids, attention_mask=mask, labels=labels, next_sentence_label=sop)
+    assert out.loss.isfinite()
+    out.loss.backward()
+    assert m.bert.embeddings.word_embeddings.weight.grad is not None
+
+
+def test_bert_mlm_loss_matches_dense_ce():
+    """vocab-parallel CE path (tp=1) equals plain CE with ignore_index."""
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.models.megatron_bert.modeling_megatron_bert import (
+        MegatronBertForMaskedLM)
+    torch.manual_seed(0)
+    m = MegatronBertForMaskedLM(bert_tiny_config()).eval()
+    ids = _ids()
+    labels = ids.clone()
+    labels[:, ::2] = -100
+    with torch.no_grad():
+        out = m(ids, labels=labels)
+        ref = torch.nn.functional.cross_entropy(
+            out.logits.view(-1, 256).float(), labels.view(-1), ignore_index=-100)
+    assert torch.allclose(out.loss, ref, atol=1e-5)
+
+
+def _llama_tp_worker(rank, world_size, full_sd_cpu):
+    import torch.distributed as dist
+    from fengshen_amd.parallel.groups import (
+        init_distributed, initialize_model_parallel)
+    from fengshen_amd.models.llama.configuration_llama import llama_tiny_config
+    from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+    from fengshen_amd.utils.tp_convert import shard_state_dict
+
+    init_distributed(backend="gloo")
+    initialize_model_parallel(tensor_model_parallel_size=world_size)
+    torch.manual_seed(123)
+    m = LlamaForCausalLM(llama_tiny_config())
+    shard = shard_state_dict(m, full_sd_cpu, world_size, rank)
+    m.load_state_dict(shard)
+    ids = _ids(seed=5)
+    out = m(ids, labels=ids)
+    loss = out.loss.item()
+    out.loss.backward()
+    # TP grads of a duplicated param must match across ranks after autograd
+    gnorm = m.model.norm.weight.grad.norm().item()
+    dist.destroy_process_group()
+    return {"loss": loss, "gnorm": gnorm}
+
+
+def test_llama_tp2_matches_tp1():
+    from fengshen_amd.models.llama.configuration_llama import llama_tiny_config
+    from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+    torch.manual_seed(123)
+    ref = LlamaForCausalLM(llama_tiny_config())
+    full_sd = {k: v.clone() for k, v in ref.state_dict().items()}
+    ids = _ids(seed=5)
+    out = ref(ids, labels=ids)
+    ref_loss = out.loss.item()
+    out.loss.backward()
+    ref_gnorm = ref.model.norm.weight.grad.norm().item()
+
+    results = run_distributed(_llama_tp_worker, world_size=2, args=(full_sd,))
+    for r in results:
+        assert abs(r["loss"] - ref_loss) < 1e-4, (r["loss"], ref_loss)
+        assert abs(r["gnorm"] - ref_gnorm) / max(ref_gnorm, 1e-8) < 1e-3
+
+
+def test_tp_shard_merge_roundtrip():
+    from fengshen_amd.models.llama.configuration_llama import llama_tiny_config
+    from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+    from fengshen_amd.utils.tp_convert import shard_state_dict, merge_state_dicts
+    torch.manual_seed(1)
+    m = LlamaForCausalLM(llama_tiny_config())
+    full = m.state_dict()
+    shards = [shard_state_dict(m, full, 2, r) for r in range(2)]
+    # note: rules derived from a tp=1 model mark partition dims identically
+    merged = merge_state_dicts(m, shards)
+    for k in full:
+        assert torch.equal(full[k], merged[k]), k
