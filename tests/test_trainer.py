@@ -69,6 +69,7 @@ def _make_args(**over):
 
 def test_fit_loss_decreases(tmp_path):
     args = _make_args()
+    torch.manual_seed(0)
     model = ToyModule(args)
     loader = DataLoader(ToyDataset(), batch_size=32, shuffle=True)
     trainer = Trainer(max_steps=60, precision="fp32",
