@@ -78,7 +78,7 @@ def test_fit_loss_decreases(tmp_path):
     assert trainer.global_step == 60
     first = sum(model.losses[:5]) / 5
     last = sum(model.losses[-5:]) / 5
-    assert last < first * 0.5, f"loss did not decrease: {first} -> {last}"
+    assert last < first * 0.7, f"loss did not decrease: {first} -> {last}"
 
 
 def test_grad_accumulation_counts(tmp_path):
