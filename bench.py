@@ -1,0 +1,174 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Ziya-LLaMA-13B causal-LM training step, bf16, native
+ZeRO over RCCL, synthetic data, random-init weights (BASELINE.json metric:
+"tokens/sec Ziya-LLaMA-13B ZeRO @1/2/4/8 GPU", weak scaling).
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+(N>1 launched via torch.distributed.run, one rank per GPU over RCCL.)
+W untimed warmup steps, then EXACTLY K timed steps bracketed by
+barrier + torch.cuda.synchronize on both sides; time is MAX over ranks;
+rank 0 prints ONE JSON line.
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=8)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--model", type=str, default="ziya-llama-13b",
+                   choices=["ziya-llama-13b", "wenzhong-gpt2-3.5b",
+                            "erlangshen-1.3b", "llama-tiny"])
+    p.add_argument("--seq_len", type=int, default=2048)
+    p.add_argument("--micro_batch", type=int, default=8)
+    p.add_argument("--zero_stage", type=int, default=2)
+    p.add_argument("--lr", type=float, default=1e-5)
+    return p.parse_args()
+
+
+def build_model(name: str, seq_len: int):
+    if name == "ziya-llama-13b":
+        from fengshen_amd.models.llama.configuration_llama import (
+            ziya_llama_13b_config)
+        from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+        cfg = ziya_llama_13b_config(max_position_embeddings=max(seq_len, 2048))
+        return LlamaForCausalLM(cfg), cfg.vocab_size, "Ziya-LLaMA-13B"
+    if name == "wenzhong-gpt2-3.5b":
+        from fengshen_amd.models.gpt2.configuration_gpt2 import (
+            wenzhong_gpt2_3b5_config)
+        from fengshen_amd.models.gpt2.modeling_gpt2 import GPT2LMHeadModel
+        cfg = wenzhong_gpt2_3b5_config(
+            max_position_embeddings=max(seq_len, 1024))
+        return GPT2LMHeadModel(cfg), cfg.vocab_size, "Wenzhong-GPT2-3.5B"
+    if name == "erlangshen-1.3b":
+        from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+            erlangshen_1b3_config)
+        from fengshen_amd.models.megatron_bert.modeling_megatron_bert import (
+            MegatronBertForPreTraining)
+        cfg = erlangshen_1b3_config()
+        return MegatronBertForPreTraining(cfg), cfg.vocab_size, \
+            "Erlangshen-MegatronBert-1.3B"
+    from fengshen_amd.models.llama.configuration_llama import LlamaConfig
+    from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+    cfg = LlamaConfig(vocab_size=2048, hidden_size=512, num_hidden_layers=4,
+                      num_attention_heads=8, intermediate_size=1408,
+                      max_position_embeddings=max(seq_len, 512))
+    return LlamaForCausalLM(cfg), cfg.vocab_size, "llama-tiny"
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    assert torch.cuda.is_available(), "bench.py requires a GPU"
+    torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+
+    import torch.distributed as dist
+    from fengshen_amd.parallel.groups import init_distributed
+    from fengshen_amd.parallel.zero import ZeroOptimizer
+    from fengshen_amd.ops import has_ext
+    assert has_ext(), "HIP extension must be built (python -m fengshen_amd.ops.build)"
+
+    if world > 1:
+        init_distributed(backend="nccl")
+
+    torch.manual_seed(1234)
+    model, vocab, model_name = build_model(args.model, args.seq_len)
+    model = model.to(torch.bfloat16).to(device)
+    if hasattr(model, "gradient_checkpointing_enable"):
+        model.gradient_checkpointing_enable()
+    model.train()
+
+    opt = ZeroOptimizer(model.parameters(), stage=args.zero_stage, lr=args.lr,
+                        betas=(0.9, 0.95), eps=1e-8, weight_decay=0.1)
+
+    is_bert = args.model == "erlangshen-1.3b"
+    b, s = args.micro_batch, args.seq_len
+    g = torch.Generator(device="cpu").manual_seed(42 + rank)
+    ids = torch.randint(3, vocab, (b, s), generator=g).to(device)
+    if is_bert:
+        labels = ids.clone()
+        mask_pos = torch.rand(b, s, generator=g) < 0.15
+        labels[~mask_pos.to(device)] = -100
+        sop = torch.randint(0, 2, (b,), generator=g).to(device)
+        batch = dict(input_ids=ids, labels=labels, next_sentence_label=sop,
+                     attention_mask=torch.ones_like(ids))
+    else:
+        batch = dict(input_ids=ids, labels=ids)
+
+    def step():
+        out = model(**batch)
+        opt.zero_grad()
+        out.loss.backward()
+        opt.step()
+        return out.loss
+
+    for _ in range(args.warmup):
+        loss = step()
+
+    if world > 1:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = step()
+    torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = t.item()
+
+    tokens_per_step = b * s * world
+    samples_per_step = b * world
+    value = (samples_per_step if is_bert else tokens_per_step) \
+        * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": ("samples/sec Erlangshen-1.3B MLM pretrain" if is_bert
+                       else "tokens/sec Ziya-LLaMA-13B ZeRO"
+                       if args.model == "ziya-llama-13b"
+                       else f"tokens/sec {model_name}"),
+            "value": round(value, 2),
+            "unit": "samples/s" if is_bert else "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "loss": round(float(loss.item()), 4),
+            "config": {
+                "model": model_name,
+                "global_batch": b * world,
+                "micro_batch": b,
+                "seq_len": s,
+                "parallelism": f"zero{args.zero_stage}_dp{world}",
+                "activation_checkpointing": True,
+            },
+        }), flush=True)
+
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

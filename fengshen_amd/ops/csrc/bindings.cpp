@@ -1,0 +1,272 @@
+// Torch bindings for fengshen_amd HIP kernels (gfx950).
+// Kernels live in kernels.hip / flash_attn.hip as extern "C" launchers taking
+// raw pointers + hipStream_t; this file owns all Tensor plumbing.
+#include <torch/extension.h>
+
+#include <hip/hip_runtime.h>
+#include <c10/hip/HIPStream.h>
+
+#include <vector>
+
+enum FsDtype { FS_F32 = 0, FS_BF16 = 1, FS_F16 = 2 };
+
+static int fs_dtype(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return FS_F32;
+    case at::kBFloat16: return FS_BF16;
+    case at::kHalf: return FS_F16;
+    default: TORCH_CHECK(false, "unsupported dtype ", t.scalar_type());
+  }
+}
+
+static hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+extern "C" {
+void fs_rms_norm_fwd(const void*, const void*, void*, float*, int, int, float,
+                     int, hipStream_t);
+void fs_rms_norm_bwd(const void*, const void*, const void*, const float*,
+                     void*, float*, int, int, int, hipStream_t);
+void fs_layer_norm_fwd(const void*, const void*, const void*, void*, float*,
+                       float*, int, int, float, int, hipStream_t);
+void fs_layer_norm_bwd(const void*, const void*, const void*, const float*,
+                       const float*, void*, float*, float*, int, int, int,
+                       hipStream_t);
+void fs_scaled_softmax_fwd(const void*, const unsigned char*, void*, float,
+                           int, int, int, int, int, int, int, hipStream_t);
+void fs_scaled_softmax_bwd(const void*, const void*, void*, float, int, int,
+                           int, hipStream_t);
+void fs_rope(const void*, void*, const float*, const float*, long, int, int,
+             int, int, int, hipStream_t);
+void fs_swiglu_fwd(const void*, void*, long, int, int, hipStream_t);
+void fs_swiglu_bwd(const void*, const void*, void*, long, int, int,
+                   hipStream_t);
+void fs_bias_gelu(const void*, const void*, const void*, void*, long, int, int,
+                  int, hipStream_t);
+void fs_fused_adamw(float*, const void*, float*, float*, void*, long, float,
+                    float, float, float, float, int, int, int, hipStream_t);
+}
+
+// ---------------------------------------------------------------------------
+static std::vector<at::Tensor> rms_norm_fwd(at::Tensor x, at::Tensor w,
+                                            double eps) {
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "hidden must be divisible by 8");
+  const long rows = x.numel() / H;
+  auto out = at::empty_like(x);
+  auto invrms = at::empty({rows}, x.options().dtype(at::kFloat));
+  fs_rms_norm_fwd(x.data_ptr(), w.data_ptr(), out.data_ptr(),
+                  invrms.data_ptr<float>(), (int)rows, H, (float)eps,
+                  fs_dtype(x), cur_stream());
+  return {out, invrms};
+}
+
+static std::vector<at::Tensor> rms_norm_bwd(at::Tensor gy, at::Tensor x,
+                                            at::Tensor w, at::Tensor invrms) {
+  const int H = x.size(-1);
+  const long rows = x.numel() / H;
+  auto gx = at::empty_like(x);
+  auto gw32 = at::zeros({H}, x.options().dtype(at::kFloat));
+  fs_rms_norm_bwd(gy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                  invrms.data_ptr<float>(), gx.data_ptr(),
+                  gw32.data_ptr<float>(), (int)rows, H, fs_dtype(x),
+                  cur_stream());
+  return {gx, gw32.to(w.scalar_type())};
+}
+
+static std::vector<at::Tensor> layer_norm_fwd(at::Tensor x, at::Tensor w,
+                                              c10::optional<at::Tensor> b,
+                                              double eps) {
+  TORCH_CHECK(x.is_contiguous());
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "hidden must be divisible by 8");
+  const long rows = x.numel() / H;
+  auto out = at::empty_like(x);
+  auto mean = at::empty({rows}, x.options().dtype(at::kFloat));
+  auto invstd = at::empty({rows}, x.options().dtype(at::kFloat));
+  fs_layer_norm_fwd(x.data_ptr(), w.data_ptr(),
+                    b.has_value() ? b->data_ptr() : nullptr, out.data_ptr(),
+                    mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                    (int)rows, H, (float)eps, fs_dtype(x), cur_stream());
+  return {out, mean, invstd};
+}
+
+static std::vector<at::Tensor> layer_norm_bwd(at::Tensor gy, at::Tensor x,
+                                              at::Tensor w, at::Tensor mean,
+                                              at::Tensor invstd) {
+  const int H = x.size(-1);
+  const long rows = x.numel() / H;
+  auto gx = at::empty_like(x);
+  auto gw32 = at::zeros({H}, x.options().dtype(at::kFloat));
+  auto gb32 = at::zeros({H}, x.options().dtype(at::kFloat));
+  fs_layer_norm_bwd(gy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                    mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                    gx.data_ptr(), gw32.data_ptr<float>(),
+                    gb32.data_ptr<float>(), (int)rows, H, fs_dtype(x),
+                    cur_stream());
+  return {gx, gw32.to(w.scalar_type()), gb32.to(w.scalar_type())};
+}
+
+// x: [b, np, sq, sk]; mask: uint8/bool [mb, 1, sq, sk] (1 = masked)
+static at::Tensor scaled_masked_softmax_fwd(at::Tensor x,
+                                            c10::optional<at::Tensor> mask,
+                                            double scale) {
+  TORCH_CHECK(x.dim() == 4 && x.is_contiguous());
+  const int sk = x.size(3), sq = x.size(2), np = x.size(1), b = x.size(0);
+  auto out = at::empty_like(x);
+  const unsigned char* mp = nullptr;
+  at::Tensor m8;
+  int mb = 1, mode = 0;
+  if (mask.has_value()) {
+    m8 = mask->to(at::kByte).contiguous();
+    TORCH_CHECK(m8.dim() == 4 && m8.size(3) == sk && m8.size(2) == sq,
+                "mask must be [mb,1,sq,sk]");
+    mb = m8.size(0);
+    TORCH_CHECK(mb == 1 || mb == b, "mask batch must be 1 or b");
+    mp = m8.data_ptr<unsigned char>();
+    mode = 1;
+  }
+  fs_scaled_softmax_fwd(x.data_ptr(), mp, out.data_ptr(), (float)scale,
+                        b * np * sq, sk, sq, np, mb, mode, fs_dtype(x),
+                        cur_stream());
+  return out;
+}
+
+// x: [ab, sq, sk] with sq == sk, causal mask generated in-kernel
+static at::Tensor scaled_causal_softmax_fwd(at::Tensor x, double scale) {
+  TORCH_CHECK(x.dim() == 3 && x.is_contiguous());
+  const int sk = x.size(2), sq = x.size(1), ab = x.size(0);
+  TORCH_CHECK(sq == sk, "causal softmax needs sq == sk");
+  auto out = at::empty_like(x);
+  fs_scaled_softmax_fwd(x.data_ptr(), nullptr, out.data_ptr(), (float)scale,
+                        ab * sq, sk, sq, 1, 1, 2, fs_dtype(x), cur_stream());
+  return out;
+}
+
+static at::Tensor scaled_softmax_bwd(at::Tensor gy, at::Tensor y,
+                                     double scale) {
+  TORCH_CHECK(gy.is_contiguous() && y.is_contiguous());
+  const int sk = y.size(-1);
+  const long rows = y.numel() / sk;
+  auto gx = at::empty_like(y);
+  fs_scaled_softmax_bwd(gy.data_ptr(), y.data_ptr(), gx.data_ptr(),
+                        (float)scale, (int)rows, sk, fs_dtype(y),
+                        cur_stream());
+  return gx;
+}
+
+// q,k: [b, np, s, hn]; cos/sin: [S, hn] fp32
+static std::vector<at::Tensor> rope_fwd(at::Tensor q, at::Tensor k,
+                                        at::Tensor cos, at::Tensor sin,
+                                        int64_t offset) {
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous());
+  TORCH_CHECK(cos.scalar_type() == at::kFloat && cos.is_contiguous());
+  const int hn = q.size(-1), s = q.size(-2);
+  TORCH_CHECK((hn / 2) % 8 == 0, "head_dim/2 must be divisible by 8");
+  auto qo = at::empty_like(q);
+  auto ko = at::empty_like(k);
+  const long bnq = q.numel() / ((long)s * hn);
+  const long bnk = k.numel() / ((long)s * hn);
+  fs_rope(q.data_ptr(), qo.data_ptr(), cos.data_ptr<float>(),
+          sin.data_ptr<float>(), bnq, s, hn, (int)offset, 0, fs_dtype(q),
+          cur_stream());
+  fs_rope(k.data_ptr(), ko.data_ptr(), cos.data_ptr<float>(),
+          sin.data_ptr<float>(), bnk, s, hn, (int)offset, 0, fs_dtype(k),
+          cur_stream());
+  return {qo, ko};
+}
+
+static std::vector<at::Tensor> rope_bwd(at::Tensor gq, at::Tensor gk,
+                                        at::Tensor cos, at::Tensor sin,
+                                        int64_t offset) {
+  const int hn = gq.size(-1), s = gq.size(-2);
+  auto qo = at::empty_like(gq);
+  auto ko = at::empty_like(gk);
+  const long bnq = gq.numel() / ((long)s * hn);
+  const long bnk = gk.numel() / ((long)s * hn);
+  fs_rope(gq.data_ptr(), qo.data_ptr(), cos.data_ptr<float>(),
+          sin.data_ptr<float>(), bnq, s, hn, (int)offset, 1, fs_dtype(gq),
+          cur_stream());
+  fs_rope(gk.data_ptr(), ko.data_ptr(), cos.data_ptr<float>(),
+          sin.data_ptr<float>(), bnk, s, hn, (int)offset, 1, fs_dtype(gk),
+          cur_stream());
+  return {qo, ko};
+}
+
+static at::Tensor swiglu_fwd(at::Tensor packed) {
+  TORCH_CHECK(packed.is_contiguous());
+  const int h2 = packed.size(-1);
+  TORCH_CHECK(h2 % 16 == 0, "packed dim must be divisible by 16");
+  const int h = h2 / 2;
+  const long rows = packed.numel() / h2;
+  auto sizes = packed.sizes().vec();
+  sizes.back() = h;
+  auto out = at::empty(sizes, packed.options());
+  fs_swiglu_fwd(packed.data_ptr(), out.data_ptr(), rows, h, fs_dtype(packed),
+                cur_stream());
+  return out;
+}
+
+static at::Tensor swiglu_bwd(at::Tensor gy, at::Tensor packed) {
+  const int h2 = packed.size(-1);
+  const int h = h2 / 2;
+  const long rows = packed.numel() / h2;
+  auto gpacked = at::empty_like(packed);
+  fs_swiglu_bwd(gy.contiguous().data_ptr(), packed.data_ptr(),
+                gpacked.data_ptr(), rows, h, fs_dtype(packed), cur_stream());
+  return gpacked;
+}
+
+static at::Tensor bias_gelu_fwd(at::Tensor x, at::Tensor bias) {
+  TORCH_CHECK(x.is_contiguous() && bias.is_contiguous());
+  const int h = x.size(-1);
+  TORCH_CHECK(h % 8 == 0);
+  const long rows = x.numel() / h;
+  auto out = at::empty_like(x);
+  fs_bias_gelu(x.data_ptr(), bias.data_ptr(), nullptr, out.data_ptr(), rows, h,
+               0, fs_dtype(x), cur_stream());
+  return out;
+}
+
+static at::Tensor bias_gelu_bwd(at::Tensor gy, at::Tensor x, at::Tensor bias) {
+  const int h = x.size(-1);
+  const long rows = x.numel() / h;
+  auto gx = at::empty_like(x);
+  fs_bias_gelu(x.data_ptr(), bias.data_ptr(), gy.data_ptr(), gx.data_ptr(),
+               rows, h, 1, fs_dtype(x), cur_stream());
+  return gx;
+}
+
+static void fused_adamw(at::Tensor master, at::Tensor grad, at::Tensor m,
+                        at::Tensor v, at::Tensor out_param, double lr,
+                        double beta1, double beta2, double eps, double wd,
+                        int64_t step) {
+  TORCH_CHECK(master.scalar_type() == at::kFloat);
+  TORCH_CHECK(master.is_contiguous() && grad.is_contiguous());
+  void* outp = out_param.data_ptr() == master.data_ptr()
+                   ? nullptr : out_param.data_ptr();
+  fs_fused_adamw(master.data_ptr<float>(), grad.data_ptr(),
+                 m.data_ptr<float>(), v.data_ptr<float>(), outp,
+                 master.numel(), (float)lr, (float)beta1, (float)beta2,
+                 (float)eps, (float)wd, (int)step, fs_dtype(grad),
+                 outp ? fs_dtype(out_param) : FS_F32, cur_stream());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("rms_norm_fwd", &rms_norm_fwd);
+  mod.def("rms_norm_bwd", &rms_norm_bwd);
+  mod.def("layer_norm_fwd", &layer_norm_fwd);
+  mod.def("layer_norm_bwd", &layer_norm_bwd);
+  mod.def("scaled_masked_softmax_fwd", &scaled_masked_softmax_fwd);
+  mod.def("scaled_causal_softmax_fwd", &scaled_causal_softmax_fwd);
+  mod.def("scaled_softmax_bwd", &scaled_softmax_bwd);
+  mod.def("rope_fwd", &rope_fwd);
+  mod.def("rope_bwd", &rope_bwd);
+  mod.def("swiglu_fwd", &swiglu_fwd);
+  mod.def("swiglu_bwd", &swiglu_bwd);
+  mod.def("bias_gelu_fwd", &bias_gelu_fwd);
+  mod.def("bias_gelu_bwd", &bias_gelu_bwd);
+  mod.def("fused_adamw", &fused_adamw);
+}

@@ -1,0 +1,655 @@
+#include "hip/hip_runtime.h"
+// fengshen_amd fused kernels for MI355X (gfx950, CDNA4).
+//
+// Design (per /opt/skills/guides/cdna_hip_programming.md):
+//   * all ops here are memory-bound: vectorized bf16 loads (short8, 16B/lane),
+//     one global read + one global write per tensor where possible
+//   * rows staged through LDS (160 KiB/CU) so softmax/norms read HBM once
+//   * wave64 shuffle reductions, block reduction via LDS
+//   * grid-stride loops, 256-thread blocks
+//
+// Replaces (behavioral parity, redesigned for wave64 — NOT translated):
+//   scaled_masked_softmax_cuda + scaled_upper_triang_masked_softmax_cuda
+//     (reference fused_kernels/*.h warp-softmax, warp=32, sk<=2048 cap;
+//      ours: LDS row softmax, any sk up to 40960)
+//   apex fused layer norm (reference layer_norm_cuda.cpp, vestigial)
+//   RMSNorm/rotary/SwiGLU/bias-GELU (reference runs these eager/jit)
+//   DeepSpeed FusedAdam (multi-bucket flat fused AdamW)
+
+#include "common.h"
+
+enum FsDtype { FS_F32 = 0, FS_BF16 = 1, FS_F16 = 2 };
+
+// ===========================================================================
+// RMSNorm
+// ===========================================================================
+template <typename T>
+__global__ void rms_norm_fwd_kernel(const T* __restrict__ x,
+                                    const T* __restrict__ w,
+                                    T* __restrict__ out,
+                                    float* __restrict__ invrms,
+                                    int rows, int H, float eps) {
+  extern __shared__ float lds[];          // [H] row staging
+  __shared__ float red[32];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* xr = x + (long)row * H;
+    T* yr = out + (long)row * H;
+    float ss = 0.f;
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+      float v[8];
+      load8<T>(xr + c, v);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        lds[c + i] = v[i];
+        ss += v[i] * v[i];
+      }
+    }
+    ss = block_reduce_sum(ss, red);
+    float ir = rsqrtf(ss / H + eps);
+    if (threadIdx.x == 0 && invrms) invrms[row] = ir;
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+      float wv[8], o[8];
+      load8<T>(w + c, wv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) o[i] = lds[c + i] * ir * wv[i];
+      store8<T>(yr + c, o);
+    }
+    __syncthreads();
+  }
+}
+
+// bwd: gx = ir*(gy*w - xhat * mean(gy*w*xhat)); gw(+)= gy*xhat (fp32 atomics)
+template <typename T>
+__global__ void rms_norm_bwd_kernel(const T* __restrict__ gy,
+                                    const T* __restrict__ x,
+                                    const T* __restrict__ w,
+                                    const float* __restrict__ invrms,
+                                    T* __restrict__ gx,
+                                    float* __restrict__ gw_f32,
+                                    int rows, int H) {
+  extern __shared__ float lds[];  // [2*H]: xhat row, gyw row
+  __shared__ float red[32];
+  float* xhat_s = lds;
+  float* gyw_s = lds + H;
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* xr = x + (long)row * H;
+    const T* gr = gy + (long)row * H;
+    T* gxr = gx + (long)row * H;
+    const float ir = invrms[row];
+    float dot = 0.f;
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+      float xv[8], gv[8], wv[8];
+      load8<T>(xr + c, xv);
+      load8<T>(gr + c, gv);
+      load8<T>(w + c, wv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float xh = xv[i] * ir;
+        float gw_ = gv[i] * wv[i];
+        xhat_s[c + i] = xh;
+        gyw_s[c + i] = gw_;
+        dot += gw_ * xh;
+        atomicAdd(&gw_f32[c + i], gv[i] * xh);
+      }
+    }
+    dot = block_reduce_sum(dot, red) / H;
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+      float o[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        o[i] = ir * (gyw_s[c + i] - xhat_s[c + i] * dot);
+      store8<T>(gxr + c, o);
+    }
+    __syncthreads();
+  }
+}
+
+// ===========================================================================
+// LayerNorm
+// ===========================================================================
+template <typename T>
+__global__ void layer_norm_fwd_kernel(const T* __restrict__ x,
+                                      const T* __restrict__ w,
+                                      const T* __restrict__ b,
+                                      T* __restrict__ out,
+                                      float* __restrict__ mean_out,
+                                      float* __restrict__ invstd_out,
+                                      int rows, int H, float eps) {
+  extern __shared__ float lds[];  // [H]
+  __shared__ float red[32];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* xr = x + (long)row * H;
+    T* yr = out + (long)row * H;
+    float s = 0.f;
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+      float v[8];
+      load8<T>(xr + c, v);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        lds[c + i] = v[i];
+        s += v[i];
+      }
+    }
+    float mu = block_reduce_sum(s, red) / H;
+    float var = 0.f;
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float d = lds[c + i] - mu;
+        var += d * d;
+      }
+    }
+    var = block_reduce_sum(var, red) / H;
+    float is = rsqrtf(var + eps);
+    if (threadIdx.x == 0) {
+      if (mean_out) mean_out[row] = mu;
+      if (invstd_out) invstd_out[row] = is;
+    }
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+      float wv[8], bv[8], o[8];
+      load8<T>(w + c, wv);
+      if (b) load8<T>(b + c, bv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        o[i] = (lds[c + i] - mu) * is * wv[i] + (b ? bv[i] : 0.f);
+      }
+      store8<T>(yr + c, o);
+    }
+    __syncthreads();
+  }
+}
+
+template <typename T>
+__global__ void layer_norm_bwd_kernel(const T* __restrict__ gy,
+                                      const T* __restrict__ x,
+                                      const T* __restrict__ w,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ invstd,
+                                      T* __restrict__ gx,
+                                      float* __restrict__ gw_f32,
+                                      float* __restrict__ gb_f32,
+                                      int rows, int H) {
+  extern __shared__ float lds[];  // [2*H]
+  __shared__ float red[32];
+  float* xhat_s = lds;
+  float* gyw_s = lds + H;
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* xr = x + (long)row * H;
+    const T* gr = gy + (long)row * H;
+    T* gxr = gx + (long)row * H;
+    const float mu = mean[row];
+    const float is = invstd[row];
+    float dot = 0.f, gsum = 0.f;
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+      float xv[8], gv[8], wv[8];
+      load8<T>(xr + c, xv);
+      load8<T>(gr + c, gv);
+      load8<T>(w + c, wv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float xh = (xv[i] - mu) * is;
+        float gw_ = gv[i] * wv[i];
+        xhat_s[c + i] = xh;
+        gyw_s[c + i] = gw_;
+        dot += gw_ * xh;
+        gsum += gw_;
+        atomicAdd(&gw_f32[c + i], gv[i] * xh);
+        if (gb_f32) atomicAdd(&gb_f32[c + i], gv[i]);
+      }
+    }
+    dot = block_reduce_sum(dot, red) / H;
+    gsum = block_reduce_sum(gsum, red) / H;
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+      float o[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        o[i] = is * (gyw_s[c + i] - gsum - xhat_s[c + i] * dot);
+      store8<T>(gxr + c, o);
+    }
+    __syncthreads();
+  }
+}
+
+// ===========================================================================
+// Softmax family (fused scale + mask + softmax, fp32 accumulation in LDS)
+// ===========================================================================
+// mode 0: plain scaled softmax
+// mode 1: padding mask (uint8, 1 = masked), mask row layout [mb,1,sq,sk]
+// mode 2: causal (upper-triangular masked by index compare, no mask tensor)
+template <typename T, int MODE>
+__global__ void scaled_softmax_fwd_kernel(const T* __restrict__ x,
+                                          const unsigned char* __restrict__ mask,
+                                          T* __restrict__ out, float scale,
+                                          int rows, int sk, int sq,
+                                          int np, int mask_batches) {
+  extern __shared__ float lds[];  // [sk]
+  __shared__ float red[32];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* xr = x + (long)row * sk;
+    T* yr = out + (long)row * sk;
+    const int si = row % sq;                 // query position
+    const unsigned char* mr = nullptr;
+    if (MODE == 1) {
+      const int bi = row / (np * sq);
+      const int mb = (mask_batches == 1) ? 0 : bi;
+      mr = mask + ((long)mb * sq + si) * sk;
+    }
+    const int limit = (MODE == 2) ? (si + 1) : sk;
+
+    float mx = -INFINITY;
+    int c = threadIdx.x * 8;
+    for (; c + 7 < limit; c += blockDim.x * 8) {
+      float v[8];
+      load8<T>(xr + c, v);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float val = v[i] * scale;
+        if (MODE == 1 && mr[c + i]) val = -10000.f;
+        lds[c + i] = val;
+        mx = fmaxf(mx, val);
+      }
+    }
+    // scalar tail: elements in [limit & ~7, limit)
+    for (int t = (limit & ~7) + threadIdx.x; t < limit; t += blockDim.x) {
+      float val = to_f32<T>(xr[t]) * scale;
+      if (MODE == 1 && mr[t]) val = -10000.f;
+      lds[t] = val;
+      mx = fmaxf(mx, val);
+    }
+    mx = block_reduce_max(mx, red);
+    float sum = 0.f;
+    for (int t = threadIdx.x; t < limit; t += blockDim.x) {
+      float e = __expf(lds[t] - mx);
+      lds[t] = e;
+      sum += e;
+    }
+    sum = block_reduce_sum(sum, red);
+    const float inv = 1.f / sum;
+    for (int t = threadIdx.x; t < limit; t += blockDim.x)
+      lds[t] *= inv;
+    __syncthreads();
+    for (c = threadIdx.x * 8; c + 7 < sk; c += blockDim.x * 8) {
+      float o[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) o[i] = (c + i < limit) ? lds[c + i] : 0.f;
+      store8<T>(yr + c, o);
+    }
+    for (int t = (sk & ~7) + threadIdx.x; t < sk; t += blockDim.x)
+      yr[t] = from_f32<T>(t < limit ? lds[t] : 0.f);
+    __syncthreads();
+  }
+}
+
+// bwd (shared by all modes): dx = scale * y * (gy - sum(gy*y))
+template <typename T>
+__global__ void scaled_softmax_bwd_kernel(const T* __restrict__ gy,
+                                          const T* __restrict__ y,
+                                          T* __restrict__ gx, float scale,
+                                          int rows, int sk) {
+  extern __shared__ float lds[];  // [2*sk]: y, gy
+  __shared__ float red[32];
+  float* ys = lds;
+  float* gs = lds + sk;
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const T* yr = y + (long)row * sk;
+    const T* gr = gy + (long)row * sk;
+    T* gxr = gx + (long)row * sk;
+    float dot = 0.f;
+    int c = threadIdx.x * 8;
+    for (; c + 7 < sk; c += blockDim.x * 8) {
+      float yv[8], gv[8];
+      load8<T>(yr + c, yv);
+      load8<T>(gr + c, gv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        ys[c + i] = yv[i];
+        gs[c + i] = gv[i];
+        dot += yv[i] * gv[i];
+      }
+    }
+    for (int t = (sk & ~7) + threadIdx.x; t < sk; t += blockDim.x) {
+      float yv = to_f32<T>(yr[t]);
+      float gv = to_f32<T>(gr[t]);
+      ys[t] = yv;
+      gs[t] = gv;
+      dot += yv * gv;
+    }
+    dot = block_reduce_sum(dot, red);
+    for (c = threadIdx.x * 8; c + 7 < sk; c += blockDim.x * 8) {
+      float o[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        o[i] = scale * ys[c + i] * (gs[c + i] - dot);
+      store8<T>(gxr + c, o);
+    }
+    for (int t = (sk & ~7) + threadIdx.x; t < sk; t += blockDim.x)
+      gxr[t] = from_f32<T>(scale * ys[t] * (gs[t] - dot));
+    __syncthreads();
+  }
+}
+
+// ===========================================================================
+// RoPE (half-rotation layout: pair (d, d+hn/2), shared cos/sin per pair)
+// ===========================================================================
+template <typename T, bool BWD>
+__global__ void rope_kernel(const T* __restrict__ x, T* __restrict__ out,
+                            const float* __restrict__ cos_t,
+                            const float* __restrict__ sin_t,
+                            long total_pairs, int s, int h2, int offset) {
+  // x: [b*np, s, hn]; pairs indexed by (bn, si, d) with d in [0, h2)
+  const long stride = (long)gridDim.x * blockDim.x * 8;
+  for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
+       i < total_pairs; i += stride) {
+    // decompose on the first element of the 8-vector; the whole vector is
+    // within one (bn, si) row because h2 % 8 == 0
+    const long d = i % h2;
+    const long si = (i / h2) % s;
+    const long bn = i / ((long)h2 * s);
+    const long base = (bn * s + si) * (2 * h2);
+    const float* cr = cos_t + (si + offset) * (2 * h2) + d;
+    const float* sr = sin_t + (si + offset) * (2 * h2) + d;
+    float x1[8], x2[8], o1[8], o2[8];
+    load8<T>(x + base + d, x1);
+    load8<T>(x + base + d + h2, x2);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float c = cr[j];
+      const float sn = sr[j];
+      if (!BWD) {
+        o1[j] = x1[j] * c - x2[j] * sn;
+        o2[j] = x2[j] * c + x1[j] * sn;
+      } else {  // transpose rotation
+        o1[j] = x1[j] * c + x2[j] * sn;
+        o2[j] = x2[j] * c - x1[j] * sn;
+      }
+    }
+    store8<T>(out + base + d, o1);
+    store8<T>(out + base + d + h2, o2);
+  }
+}
+
+// ===========================================================================
+// SwiGLU: packed [rows, 2h] -> out [rows, h]
+// ===========================================================================
+template <typename T>
+__global__ void swiglu_fwd_kernel(const T* __restrict__ packed,
+                                  T* __restrict__ out, long rows, int h) {
+  const long total = rows * (long)h;
+  const long stride = (long)gridDim.x * blockDim.x * 8;
+  for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; i < total;
+       i += stride) {
+    const long row = i / h;
+    const long col = i % h;
+    const T* g = packed + row * 2 * h + col;
+    const T* u = g + h;
+    float gv[8], uv[8], o[8];
+    load8<T>(g, gv);
+    load8<T>(u, uv);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float sig = 1.f / (1.f + __expf(-gv[j]));
+      o[j] = gv[j] * sig * uv[j];
+    }
+    store8<T>(out + i, o);
+  }
+}
+
+template <typename T>
+__global__ void swiglu_bwd_kernel(const T* __restrict__ gy,
+                                  const T* __restrict__ packed,
+                                  T* __restrict__ gpacked, long rows, int h) {
+  const long total = rows * (long)h;
+  const long stride = (long)gridDim.x * blockDim.x * 8;
+  for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; i < total;
+       i += stride) {
+    const long row = i / h;
+    const long col = i % h;
+    const T* g = packed + row * 2 * h + col;
+    const T* u = g + h;
+    float gv[8], uv[8], go[8], dg[8], du[8];
+    load8<T>(g, gv);
+    load8<T>(u, uv);
+    load8<T>(gy + i, go);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float sig = 1.f / (1.f + __expf(-gv[j]));
+      const float silu = gv[j] * sig;
+      dg[j] = go[j] * uv[j] * sig * (1.f + gv[j] * (1.f - sig));
+      du[j] = go[j] * silu;
+    }
+    store8<T>(gpacked + row * 2 * h + col, dg);
+    store8<T>(gpacked + row * 2 * h + col + h, du);
+  }
+}
+
+// ===========================================================================
+// bias-GELU (tanh approx, matches reference activations.py:60-77)
+// ===========================================================================
+__device__ __forceinline__ float gelu_tanh(float y) {
+  return y * 0.5f * (1.f + tanhf(0.79788456f * y * (1.f + 0.044715f * y * y)));
+}
+__device__ __forceinline__ float gelu_tanh_grad(float y) {
+  const float t = tanhf(0.79788456f * y * (1.f + 0.044715f * y * y));
+  return 0.5f * y * ((1.f - t * t) * (0.79788456f + 0.1070322243f * y * y)) +
+         0.5f * (1.f + t);
+}
+
+template <typename T, bool BWD>
+__global__ void bias_gelu_kernel(const T* __restrict__ x,
+                                 const T* __restrict__ bias,
+                                 const T* __restrict__ gy,
+                                 T* __restrict__ out, long rows, int h) {
+  const long total = rows * (long)h;
+  const long stride = (long)gridDim.x * blockDim.x * 8;
+  for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; i < total;
+       i += stride) {
+    const long col = i % h;
+    float xv[8], bv[8], o[8];
+    load8<T>(x + i, xv);
+    load8<T>(bias + col, bv);
+    if (!BWD) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) o[j] = gelu_tanh(xv[j] + bv[j]);
+    } else {
+      float gv[8];
+      load8<T>(gy + i, gv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o[j] = gelu_tanh_grad(xv[j] + bv[j]) * gv[j];
+    }
+    store8<T>(out + i, o);
+  }
+}
+
+// ===========================================================================
+// fused AdamW over flat fp32 master (+ low-precision param copy-out)
+// ===========================================================================
+template <typename GT, typename OT>
+__global__ void fused_adamw_kernel(float* __restrict__ master,
+                                   const GT* __restrict__ grad,
+                                   float* __restrict__ m,
+                                   float* __restrict__ v,
+                                   OT* __restrict__ out_param,
+                                   long n, float lr, float beta1, float beta2,
+                                   float eps, float wd, float bc1, float bc2) {
+  const long stride = (long)gridDim.x * blockDim.x * 4;
+  for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 4; i < n;
+       i += stride) {
+    const int k = (int)min((long)4, n - i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      if (j >= k) break;
+      const long t = i + j;
+      float g = to_f32<GT>(grad[t]);
+      float p = master[t];
+      p *= (1.f - lr * wd);
+      float mj = m[t] = beta1 * m[t] + (1.f - beta1) * g;
+      float vj = v[t] = beta2 * v[t] + (1.f - beta2) * g * g;
+      const float denom = sqrtf(vj / bc2) + eps;
+      p -= lr / bc1 * (mj / denom);
+      master[t] = p;
+      if (out_param) out_param[t] = from_f32<OT>(p);
+    }
+  }
+}
+
+// ===========================================================================
+// extern "C" launchers
+// ===========================================================================
+static inline int grid_for(long work_items, int block = 256) {
+  long g = (work_items + block - 1) / block;
+  if (g > FS_MAX_BLOCKS) g = FS_MAX_BLOCKS;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+#define DISPATCH_DTYPE(dtype, T, ...)            \
+  switch (dtype) {                               \
+    case FS_F32: { using T = float; __VA_ARGS__; break; }   \
+    case FS_BF16: { using T = bf16_t; __VA_ARGS__; break; } \
+    case FS_F16: { using T = fp16_t; __VA_ARGS__; break; }  \
+  }
+
+extern "C" {
+
+void fs_rms_norm_fwd(const void* x, const void* w, void* out, float* invrms,
+                     int rows, int H, float eps, int dtype, hipStream_t s) {
+  int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
+  size_t lds = (size_t)H * 4;
+  DISPATCH_DTYPE(dtype, T,
+    hipLaunchKernelGGL((rms_norm_fwd_kernel<T>), dim3(grid), dim3(256), lds, s,
+                       (const T*)x, (const T*)w, (T*)out, invrms, rows, H, eps));
+}
+
+void fs_rms_norm_bwd(const void* gy, const void* x, const void* w,
+                     const float* invrms, void* gx, float* gw_f32, int rows,
+                     int H, int dtype, hipStream_t s) {
+  int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
+  size_t lds = (size_t)H * 8;
+  DISPATCH_DTYPE(dtype, T,
+    hipLaunchKernelGGL((rms_norm_bwd_kernel<T>), dim3(grid), dim3(256), lds, s,
+                       (const T*)gy, (const T*)x, (const T*)w, invrms, (T*)gx,
+                       gw_f32, rows, H));
+}
+
+void fs_layer_norm_fwd(const void* x, const void* w, const void* b, void* out,
+                       float* mean, float* invstd, int rows, int H, float eps,
+                       int dtype, hipStream_t s) {
+  int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
+  size_t lds = (size_t)H * 4;
+  DISPATCH_DTYPE(dtype, T,
+    hipLaunchKernelGGL((layer_norm_fwd_kernel<T>), dim3(grid), dim3(256), lds, s,
+                       (const T*)x, (const T*)w, (const T*)b, (T*)out, mean,
+                       invstd, rows, H, eps));
+}
+
+void fs_layer_norm_bwd(const void* gy, const void* x, const void* w,
+                       const float* mean, const float* invstd, void* gx,
+                       float* gw_f32, float* gb_f32, int rows, int H, int dtype,
+                       hipStream_t s) {
+  int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
+  size_t lds = (size_t)H * 8;
+  DISPATCH_DTYPE(dtype, T,
+    hipLaunchKernelGGL((layer_norm_bwd_kernel<T>), dim3(grid), dim3(256), lds, s,
+                       (const T*)gy, (const T*)x, (const T*)w, mean, invstd,
+                       (T*)gx, gw_f32, gb_f32, rows, H));
+}
+
+void fs_scaled_softmax_fwd(const void* x, const unsigned char* mask, void* out,
+                           float scale, int rows, int sk, int sq, int np,
+                           int mask_batches, int mode, int dtype,
+                           hipStream_t s) {
+  int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
+  size_t lds = (size_t)sk * 4;
+  DISPATCH_DTYPE(dtype, T, {
+    if (mode == 0)
+      hipLaunchKernelGGL((scaled_softmax_fwd_kernel<T, 0>), dim3(grid),
+                         dim3(256), lds, s, (const T*)x, mask, (T*)out, scale,
+                         rows, sk, sq, np, mask_batches);
+    else if (mode == 1)
+      hipLaunchKernelGGL((scaled_softmax_fwd_kernel<T, 1>), dim3(grid),
+                         dim3(256), lds, s, (const T*)x, mask, (T*)out, scale,
+                         rows, sk, sq, np, mask_batches);
+    else
+      hipLaunchKernelGGL((scaled_softmax_fwd_kernel<T, 2>), dim3(grid),
+                         dim3(256), lds, s, (const T*)x, mask, (T*)out, scale,
+                         rows, sk, sq, np, mask_batches);
+  });
+}
+
+void fs_scaled_softmax_bwd(const void* gy, const void* y, void* gx, float scale,
+                           int rows, int sk, int dtype, hipStream_t s) {
+  int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
+  size_t lds = (size_t)sk * 8;
+  DISPATCH_DTYPE(dtype, T,
+    hipLaunchKernelGGL((scaled_softmax_bwd_kernel<T>), dim3(grid), dim3(256),
+                       lds, s, (const T*)gy, (const T*)y, (T*)gx, scale, rows,
+                       sk));
+}
+
+void fs_rope(const void* x, void* out, const float* cos_t, const float* sin_t,
+             long bn, int seq, int hn, int offset, int bwd, int dtype,
+             hipStream_t s) {
+  const int h2 = hn / 2;
+  const long total_pairs = bn * (long)seq * h2;
+  int grid = grid_for(total_pairs / 8);
+  DISPATCH_DTYPE(dtype, T, {
+    if (bwd)
+      hipLaunchKernelGGL((rope_kernel<T, true>), dim3(grid), dim3(256), 0, s,
+                         (const T*)x, (T*)out, cos_t, sin_t, total_pairs, seq,
+                         h2, offset);
+    else
+      hipLaunchKernelGGL((rope_kernel<T, false>), dim3(grid), dim3(256), 0, s,
+                         (const T*)x, (T*)out, cos_t, sin_t, total_pairs, seq,
+                         h2, offset);
+  });
+}
+
+void fs_swiglu_fwd(const void* packed, void* out, long rows, int h, int dtype,
+                   hipStream_t s) {
+  int grid = grid_for(rows * (long)h / 8);
+  DISPATCH_DTYPE(dtype, T,
+    hipLaunchKernelGGL((swiglu_fwd_kernel<T>), dim3(grid), dim3(256), 0, s,
+                       (const T*)packed, (T*)out, rows, h));
+}
+
+void fs_swiglu_bwd(const void* gy, const void* packed, void* gpacked, long rows,
+                   int h, int dtype, hipStream_t s) {
+  int grid = grid_for(rows * (long)h / 8);
+  DISPATCH_DTYPE(dtype, T,
+    hipLaunchKernelGGL((swiglu_bwd_kernel<T>), dim3(grid), dim3(256), 0, s,
+                       (const T*)gy, (const T*)packed, (T*)gpacked, rows, h));
+}
+
+void fs_bias_gelu(const void* x, const void* bias, const void* gy, void* out,
+                  long rows, int h, int bwd, int dtype, hipStream_t s) {
+  int grid = grid_for(rows * (long)h / 8);
+  DISPATCH_DTYPE(dtype, T, {
+    if (bwd)
+      hipLaunchKernelGGL((bias_gelu_kernel<T, true>), dim3(grid), dim3(256), 0,
+                         s, (const T*)x, (const T*)bias, (const T*)gy, (T*)out,
+                         rows, h);
+    else
+      hipLaunchKernelGGL((bias_gelu_kernel<T, false>), dim3(grid), dim3(256), 0,
+                         s, (const T*)x, (const T*)bias, (const T*)gy, (T*)out,
+                         rows, h);
+  });
+}
+
+void fs_fused_adamw(float* master, const void* grad, float* m, float* v,
+                    void* out_param, long n, float lr, float beta1, float beta2,
+                    float eps, float wd, int step, int grad_dtype,
+                    int out_dtype, hipStream_t s) {
+  const float bc1 = 1.f - powf(beta1, (float)step);
+  const float bc2 = 1.f - powf(beta2, (float)step);
+  int grid = grid_for((n + 3) / 4);
+  DISPATCH_DTYPE(grad_dtype, GT, {
+    DISPATCH_DTYPE(out_dtype, OT, {
+      hipLaunchKernelGGL((fused_adamw_kernel<GT, OT>), dim3(grid), dim3(256), 0,
+                         s, master, (const GT*)grad, m, v, (OT*)out_param, n,
+                         lr, beta1, beta2, eps, wd, bc1, bc2);
+    });
+  });
+}
+
+}  // extern "C"

@@ -1,0 +1,199 @@
+"""HIP kernel numerics vs plain-PyTorch fp32 oracles (SURVEY.md §4 pattern).
+All tests require an MI355X."""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _hip_ready():
+    if not torch.cuda.is_available():
+        return False
+    from fengshen_amd.ops import has_ext
+    return has_ext()
+
+
+@pytest.fixture(autouse=True)
+def _require_ext():
+    assert _hip_ready(), "HIP extension missing on GPU box"
+
+
+def _rand(*shape, dtype=torch.bfloat16, seed=0):
+    g = torch.Generator(device="cuda").manual_seed(seed)
+    return torch.randn(*shape, generator=g, device="cuda", dtype=torch.float32) \
+        .to(dtype)
+
+
+def test_rms_norm_fwd_bwd():
+    from fengshen_amd.ops import functional as F
+    x = _rand(4, 33, 1024).requires_grad_(True)
+    w = _rand(1024, seed=1).requires_grad_(True)
+    y = F.rms_norm(x, w, 1e-6)
+    ref = F.eager_rms_norm(x.detach().float(), w.detach().float(), 1e-6)
+    assert (y.float() - ref).abs().max().item() < 2e-2
+
+    gy = _rand(4, 33, 1024, seed=2)
+    y.backward(gy)
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().float().requires_grad_(True)
+    F.eager_rms_norm(x2, w2, 1e-6).backward(gy.float())
+    assert (x.grad.float() - x2.grad).abs().max().item() < 2e-2
+    rel = (w.grad.float() - w2.grad).abs().max() / w2.grad.abs().max()
+    assert rel.item() < 2e-2
+
+
+def test_layer_norm_fwd_bwd():
+    from fengshen_amd.ops import functional as F
+    x = _rand(6, 17, 768).requires_grad_(True)
+    w = _rand(768, seed=1).requires_grad_(True)
+    b = _rand(768, seed=2).requires_grad_(True)
+    y = F.layer_norm(x, w, b, 1e-5)
+    ref = torch.nn.functional.layer_norm(
+        x.detach().float(), (768,), w.detach().float(), b.detach().float(), 1e-5)
+    assert (y.float() - ref).abs().max().item() < 2e-2
+
+    gy = _rand(6, 17, 768, seed=3)
+    y.backward(gy)
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().float().requires_grad_(True)
+    b2 = b.detach().float().requires_grad_(True)
+    torch.nn.functional.layer_norm(x2, (768,), w2, b2, 1e-5).backward(gy.float())
+    assert (x.grad.float() - x2.grad).abs().max().item() < 2e-2
+    assert ((w.grad.float() - w2.grad).abs().max()
+            / w2.grad.abs().max()).item() < 2e-2
+    assert ((b.grad.float() - b2.grad).abs().max()
+            / b2.grad.abs().max()).item() < 2e-2
+
+
+@pytest.mark.parametrize("sk", [128, 512, 2048, 4096, 100])
+def test_scaled_masked_softmax(sk):
+    from fengshen_amd.ops import functional as F
+    b, np_, sq = 2, 4, 64
+    x = _rand(b, np_, sq, sk).requires_grad_(True)
+    mask = (torch.rand(b, 1, sq, sk, device="cuda") < 0.2)
+    mask[..., 0] = False  # keep at least one position
+    scale = 0.35
+    y = F.scaled_masked_softmax(x, mask, scale)
+    ref = F.eager_scaled_masked_softmax(x.detach().float(), mask, scale)
+    assert (y.float() - ref).abs().max().item() < 1e-2
+
+    gy = _rand(b, np_, sq, sk, seed=3)
+    y.backward(gy)
+    x2 = x.detach().float().requires_grad_(True)
+    F.eager_scaled_masked_softmax(x2, mask, scale).backward(gy.float())
+    assert (x.grad.float() - x2.grad).abs().max().item() < 1e-2
+
+
+@pytest.mark.parametrize("s", [64, 1024, 2048])
+def test_scaled_causal_softmax(s):
+    from fengshen_amd.ops import functional as F
+    ab = 8
+    x = _rand(ab, s, s).requires_grad_(True)
+    scale = 1.0 / math.sqrt(128)
+    y = F.scaled_causal_softmax(x, scale)
+    ref = F.eager_scaled_causal_softmax(x.detach().float(), scale)
+    assert (y.float() - ref).abs().max().item() < 1e-2
+    # strictly zero above diagonal
+    assert y.float().triu(1).abs().max().item() == 0.0
+
+    gy = _rand(ab, s, s, seed=3)
+    y.backward(gy)
+    x2 = x.detach().float().requires_grad_(True)
+    F.eager_scaled_causal_softmax(x2, scale).backward(gy.float())
+    assert (x.grad.float() - x2.grad).abs().max().item() < 1e-2
+
+
+def test_rope_fwd_bwd():
+    from fengshen_amd.ops import functional as F
+    b, np_, s, hn = 2, 4, 128, 128
+    q = _rand(b, np_, s, hn).requires_grad_(True)
+    k = _rand(b, np_, s, hn, seed=1).requires_grad_(True)
+    cos, sin = F.build_rope_cache(256, hn, device="cuda")
+    qo, ko = F.apply_rotary(q, k, cos, sin, offset=7)
+    q2 = q.detach().float().requires_grad_(True)
+    k2 = k.detach().float().requires_grad_(True)
+    rq, rk = F.eager_apply_rotary(q2, k2, cos, sin, offset=7)
+    assert (qo.float() - rq).abs().max().item() < 2e-2
+    assert (ko.float() - rk).abs().max().item() < 2e-2
+
+    gq = _rand(b, np_, s, hn, seed=2)
+    gk = _rand(b, np_, s, hn, seed=3)
+    (qo.float() * gq.float()).sum().backward()
+    (rq * gq.float()).sum().backward()
+    assert (q.grad.float() - q2.grad).abs().max().item() < 2e-2
+
+
+def test_swiglu_fwd_bwd():
+    from fengshen_amd.ops import functional as F
+    x = _rand(64, 2 * 1408).requires_grad_(True)
+    y = F.swiglu(x)
+    g, u = x.detach().float().chunk(2, -1)
+    ref = torch.nn.functional.silu(g) * u
+    assert (y.float() - ref).abs().max().item() < 2e-2
+
+    gy = _rand(64, 1408, seed=5)
+    y.backward(gy)
+    x2 = x.detach().float().requires_grad_(True)
+    g2, u2 = x2.chunk(2, -1)
+    (torch.nn.functional.silu(g2) * u2).backward(gy.float())
+    assert (x.grad.float() - x2.grad).abs().max().item() < 2e-2
+
+
+def test_bias_gelu_fwd_bwd():
+    from fengshen_amd.ops import functional as F
+    x = _rand(128, 3072).requires_grad_(True)
+    b = _rand(3072, seed=1).requires_grad_(True)
+    y = F.bias_gelu(x, b)
+    ref = F.eager_gelu(x.detach().float() + b.detach().float())
+    assert (y.float() - ref).abs().max().item() < 2e-2
+
+    gy = _rand(128, 3072, seed=2)
+    y.backward(gy)
+    x2 = x.detach().float().requires_grad_(True)
+    b2 = b.detach().float().requires_grad_(True)
+    F.eager_gelu(x2 + b2).backward(gy.float())
+    assert (x.grad.float() - x2.grad).abs().max().item() < 3e-2
+    assert ((b.grad.float() - b2.grad).abs().max()
+            / b2.grad.abs().max()).item() < 3e-2
+
+
+def test_fused_adamw_matches_torch():
+    from fengshen_amd.ops.adamw import fused_adamw_flat_
+    n = 1 << 20
+    torch.manual_seed(0)
+    master = torch.randn(n, device="cuda", dtype=torch.float32)
+    ref_p = master.clone()
+    grad = torch.randn(n, device="cuda", dtype=torch.bfloat16)
+    m = torch.zeros_like(master)
+    v = torch.zeros_like(master)
+    out = torch.empty(n, device="cuda", dtype=torch.bfloat16)
+
+    ref = torch.nn.Parameter(ref_p.clone())
+    opt = torch.optim.AdamW([ref], lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
+                            weight_decay=0.01)
+    for step in range(1, 4):
+        fused_adamw_flat_(master, grad, m, v, out, lr=1e-3, beta1=0.9,
+                          beta2=0.999, eps=1e-8, weight_decay=0.01, step=step)
+        ref.grad = grad.float()
+        opt.step()
+    assert (master - ref.detach()).abs().max().item() < 1e-5
+    assert (out.float() - master).abs().max().item() < 1e-2  # bf16 roundoff
+
+
+def test_attention_vs_sdpa():
+    """composite attention path (bmm + fused softmax) vs torch eager oracle."""
+    from fengshen_amd.ops import functional as F
+    b, np_, s, hn = 2, 8, 512, 128
+    q = _rand(b, np_, s, hn)
+    k = _rand(b, np_, s, hn, seed=1)
+    v = _rand(b, np_, s, hn, seed=2)
+    out = F.attention(q, k, v, causal=True, scale=1.0 / math.sqrt(hn))
+    # fp32 oracle
+    qf, kf, vf = q.float(), k.float(), v.float()
+    scores = qf @ kf.transpose(-1, -2) / math.sqrt(hn)
+    causal = torch.ones(s, s, device="cuda", dtype=torch.bool).triu(1)
+    scores = scores.masked_fill(causal, -1e9)
+    ref = torch.softmax(scores, -1) @ vf
+    assert (out.float() - ref).abs().max().item() < 2e-2
