@@ -94,18 +94,26 @@ def main():
     is_bert = args.model == "erlangshen-1.3b"
     b, s = args.micro_batch, args.seq_len
     g = torch.Generator(device="cpu").manual_seed(42 + rank)
-    ids = torch.randint(3, vocab, (b, s), generator=g).to(device)
-    if is_bert:
-        labels = ids.clone()
-        mask_pos = torch.rand(b, s, generator=g) < 0.15
-        labels[~mask_pos.to(device)] = -100
-        sop = torch.randint(0, 2, (b,), generator=g).to(device)
-        batch = dict(input_ids=ids, labels=labels, next_sentence_label=sop,
-                     attention_mask=torch.ones_like(ids))
-    else:
-        batch = dict(input_ids=ids, labels=ids)
+    # several distinct synthetic batches (avoids single-batch memorization)
+    batches = []
+    for _ in range(4):
+        ids = torch.randint(3, vocab, (b, s), generator=g).to(device)
+        if is_bert:
+            labels = ids.clone()
+            mask_pos = torch.rand(b, s, generator=g) < 0.15
+            labels[~mask_pos.to(device)] = -100
+            sop = torch.randint(0, 2, (b,), generator=g).to(device)
+            batches.append(dict(input_ids=ids, labels=labels,
+                                next_sentence_label=sop,
+                                attention_mask=torch.ones_like(ids)))
+        else:
+            batches.append(dict(input_ids=ids, labels=ids))
+
+    step_idx = [0]
 
     def step():
+        batch = batches[step_idx[0] % len(batches)]
+        step_idx[0] += 1
         out = model(**batch)
         opt.zero_grad()
         out.loss.backward()

@@ -34,7 +34,8 @@ void fs_layer_norm_bwd(const void*, const void*, const void*, const float*,
                        const float*, void*, float*, float*, int, int, int,
                        hipStream_t);
 void fs_scaled_softmax_fwd(const void*, const unsigned char*, void*, float,
-                           int, int, int, int, int, int, int, hipStream_t);
+                           int, int, int, int, int, int, int, int,
+                           hipStream_t);
 void fs_scaled_softmax_bwd(const void*, const void*, void*, float, int, int,
                            int, hipStream_t);
 void fs_rope(const void*, void*, const float*, const float*, long, int, int,
@@ -118,19 +119,21 @@ static at::Tensor scaled_masked_softmax_fwd(at::Tensor x,
   auto out = at::empty_like(x);
   const unsigned char* mp = nullptr;
   at::Tensor m8;
-  int mb = 1, mode = 0;
+  int mb = 1, mode = 0, mask_sq = 1;
   if (mask.has_value()) {
     m8 = mask->to(at::kByte).contiguous();
-    TORCH_CHECK(m8.dim() == 4 && m8.size(3) == sk && m8.size(2) == sq,
-                "mask must be [mb,1,sq,sk]");
+    TORCH_CHECK(m8.dim() == 4 && m8.size(3) == sk &&
+                    (m8.size(2) == sq || m8.size(2) == 1),
+                "mask must be [mb,1,sq|1,sk]");
     mb = m8.size(0);
+    mask_sq = m8.size(2);
     TORCH_CHECK(mb == 1 || mb == b, "mask batch must be 1 or b");
     mp = m8.data_ptr<unsigned char>();
     mode = 1;
   }
   fs_scaled_softmax_fwd(x.data_ptr(), mp, out.data_ptr(), (float)scale,
-                        b * np * sq, sk, sq, np, mb, mode, fs_dtype(x),
-                        cur_stream());
+                        b * np * sq, sk, sq, np, mb, mask_sq, mode,
+                        fs_dtype(x), cur_stream());
   return out;
 }
 
@@ -141,7 +144,8 @@ static at::Tensor scaled_causal_softmax_fwd(at::Tensor x, double scale) {
   TORCH_CHECK(sq == sk, "causal softmax needs sq == sk");
   auto out = at::empty_like(x);
   fs_scaled_softmax_fwd(x.data_ptr(), nullptr, out.data_ptr(), (float)scale,
-                        ab * sq, sk, sq, 1, 1, 2, fs_dtype(x), cur_stream());
+                        ab * sq, sk, sq, 1, 1, sq, 2, fs_dtype(x),
+                        cur_stream());
   return out;
 }
 

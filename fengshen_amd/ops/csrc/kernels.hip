@@ -220,7 +220,8 @@ __global__ void scaled_softmax_fwd_kernel(const T* __restrict__ x,
                                           const unsigned char* __restrict__ mask,
                                           T* __restrict__ out, float scale,
                                           int rows, int sk, int sq,
-                                          int np, int mask_batches) {
+                                          int np, int mask_batches,
+                                          int mask_sq) {
   extern __shared__ float lds[];  // [sk]
   __shared__ float red[32];
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
@@ -231,7 +232,8 @@ __global__ void scaled_softmax_fwd_kernel(const T* __restrict__ x,
     if (MODE == 1) {
       const int bi = row / (np * sq);
       const int mb = (mask_batches == 1) ? 0 : bi;
-      mr = mask + ((long)mb * sq + si) * sk;
+      const int msi = (mask_sq == 1) ? 0 : si;  // broadcast [b,1,1,sk] masks
+      mr = mask + ((long)mb * mask_sq + msi) * sk;
     }
     const int limit = (MODE == 2) ? (si + 1) : sk;
 
@@ -556,7 +558,7 @@ void fs_layer_norm_bwd(const void* gy, const void* x, const void* w,
 
 void fs_scaled_softmax_fwd(const void* x, const unsigned char* mask, void* out,
                            float scale, int rows, int sk, int sq, int np,
-                           int mask_batches, int mode, int dtype,
+                           int mask_batches, int mask_sq, int mode, int dtype,
                            hipStream_t s) {
   int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
   size_t lds = (size_t)sk * 4;
@@ -564,15 +566,15 @@ void fs_scaled_softmax_fwd(const void* x, const unsigned char* mask, void* out,
     if (mode == 0)
       hipLaunchKernelGGL((scaled_softmax_fwd_kernel<T, 0>), dim3(grid),
                          dim3(256), lds, s, (const T*)x, mask, (T*)out, scale,
-                         rows, sk, sq, np, mask_batches);
+                         rows, sk, sq, np, mask_batches, mask_sq);
     else if (mode == 1)
       hipLaunchKernelGGL((scaled_softmax_fwd_kernel<T, 1>), dim3(grid),
                          dim3(256), lds, s, (const T*)x, mask, (T*)out, scale,
-                         rows, sk, sq, np, mask_batches);
+                         rows, sk, sq, np, mask_batches, mask_sq);
     else
       hipLaunchKernelGGL((scaled_softmax_fwd_kernel<T, 2>), dim3(grid),
                          dim3(256), lds, s, (const T*)x, mask, (T*)out, scale,
-                         rows, sk, sq, np, mask_batches);
+                         rows, sk, sq, np, mask_batches, mask_sq);
   });
 }
 

@@ -280,15 +280,9 @@ class _BiasDropoutAdd(torch.autograd.Function):
             ctx.has_bias = bias is not None
             return y + residual
         keep = 1.0 - p
-        if use_hip(x):
-            out, mask = get_ext().bias_dropout_add_fwd(
-                x, bias if bias is not None else torch.tensor([]), residual,
-                p, torch.cuda.default_generators[x.device.index].seed()
-                if False else int(torch.randint(0, 2**31 - 1, (1,)).item()))
-        else:
-            mask = (torch.rand_like(x, dtype=torch.float32) < keep)
-            y = x if bias is None else x + bias
-            out = y * mask.to(x.dtype) / keep + residual
+        mask = (torch.rand_like(x, dtype=torch.float32) < keep)
+        y = x if bias is None else x + bias
+        out = y * mask.to(x.dtype) / keep + residual
         ctx.p = p
         ctx.has_bias = bias is not None
         ctx.save_for_backward(mask)
@@ -346,7 +340,8 @@ class _ApplyRotary(torch.autograd.Function):
         ctx.save_for_backward(cos, sin)
         ctx.offset = offset
         if use_hip(q):
-            return tuple(get_ext().rope_fwd(q, k, cos, sin, offset))
+            return tuple(get_ext().rope_fwd(q.contiguous(), k.contiguous(),
+                                            cos, sin, offset))
         return eager_apply_rotary(q, k, cos, sin, offset)
 
     @staticmethod

@@ -26,20 +26,29 @@ def _rand(*shape, dtype=torch.bfloat16, seed=0):
         .to(dtype)
 
 
+def _close(a, b, tol=2e-2):
+    """relative-to-range closeness (bf16 outputs vs fp32 oracle)."""
+    a = a.float()
+    b = b.float()
+    scale = b.abs().max().clamp(min=1.0)
+    err = (a - b).abs().max() / scale
+    assert err.item() < tol, f"rel err {err.item():.4g} (scale {scale.item():.3g})"
+
+
 def test_rms_norm_fwd_bwd():
     from fengshen_amd.ops import functional as F
     x = _rand(4, 33, 1024).requires_grad_(True)
     w = _rand(1024, seed=1).requires_grad_(True)
     y = F.rms_norm(x, w, 1e-6)
     ref = F.eager_rms_norm(x.detach().float(), w.detach().float(), 1e-6)
-    assert (y.float() - ref).abs().max().item() < 2e-2
+    _close(y, ref)
 
     gy = _rand(4, 33, 1024, seed=2)
     y.backward(gy)
     x2 = x.detach().float().requires_grad_(True)
     w2 = w.detach().float().requires_grad_(True)
     F.eager_rms_norm(x2, w2, 1e-6).backward(gy.float())
-    assert (x.grad.float() - x2.grad).abs().max().item() < 2e-2
+    _close(x.grad, x2.grad)
     rel = (w.grad.float() - w2.grad).abs().max() / w2.grad.abs().max()
     assert rel.item() < 2e-2
 
@@ -52,7 +61,7 @@ def test_layer_norm_fwd_bwd():
     y = F.layer_norm(x, w, b, 1e-5)
     ref = torch.nn.functional.layer_norm(
         x.detach().float(), (768,), w.detach().float(), b.detach().float(), 1e-5)
-    assert (y.float() - ref).abs().max().item() < 2e-2
+    _close(y, ref)
 
     gy = _rand(6, 17, 768, seed=3)
     y.backward(gy)
@@ -60,7 +69,7 @@ def test_layer_norm_fwd_bwd():
     w2 = w.detach().float().requires_grad_(True)
     b2 = b.detach().float().requires_grad_(True)
     torch.nn.functional.layer_norm(x2, (768,), w2, b2, 1e-5).backward(gy.float())
-    assert (x.grad.float() - x2.grad).abs().max().item() < 2e-2
+    _close(x.grad, x2.grad)
     assert ((w.grad.float() - w2.grad).abs().max()
             / w2.grad.abs().max()).item() < 2e-2
     assert ((b.grad.float() - b2.grad).abs().max()
@@ -77,13 +86,13 @@ def test_scaled_masked_softmax(sk):
     scale = 0.35
     y = F.scaled_masked_softmax(x, mask, scale)
     ref = F.eager_scaled_masked_softmax(x.detach().float(), mask, scale)
-    assert (y.float() - ref).abs().max().item() < 1e-2
+    _close(y, ref, 1e-2)
 
     gy = _rand(b, np_, sq, sk, seed=3)
     y.backward(gy)
     x2 = x.detach().float().requires_grad_(True)
     F.eager_scaled_masked_softmax(x2, mask, scale).backward(gy.float())
-    assert (x.grad.float() - x2.grad).abs().max().item() < 1e-2
+    _close(x.grad, x2.grad, 1e-2)
 
 
 @pytest.mark.parametrize("s", [64, 1024, 2048])
@@ -94,7 +103,7 @@ def test_scaled_causal_softmax(s):
     scale = 1.0 / math.sqrt(128)
     y = F.scaled_causal_softmax(x, scale)
     ref = F.eager_scaled_causal_softmax(x.detach().float(), scale)
-    assert (y.float() - ref).abs().max().item() < 1e-2
+    _close(y, ref, 1e-2)
     # strictly zero above diagonal
     assert y.float().triu(1).abs().max().item() == 0.0
 
@@ -102,7 +111,7 @@ def test_scaled_causal_softmax(s):
     y.backward(gy)
     x2 = x.detach().float().requires_grad_(True)
     F.eager_scaled_causal_softmax(x2, scale).backward(gy.float())
-    assert (x.grad.float() - x2.grad).abs().max().item() < 1e-2
+    _close(x.grad, x2.grad, 1e-2)
 
 
 def test_rope_fwd_bwd():
@@ -115,14 +124,14 @@ def test_rope_fwd_bwd():
     q2 = q.detach().float().requires_grad_(True)
     k2 = k.detach().float().requires_grad_(True)
     rq, rk = F.eager_apply_rotary(q2, k2, cos, sin, offset=7)
-    assert (qo.float() - rq).abs().max().item() < 2e-2
-    assert (ko.float() - rk).abs().max().item() < 2e-2
+    _close(qo, rq)
+    _close(ko, rk)
 
     gq = _rand(b, np_, s, hn, seed=2)
     gk = _rand(b, np_, s, hn, seed=3)
     (qo.float() * gq.float()).sum().backward()
     (rq * gq.float()).sum().backward()
-    assert (q.grad.float() - q2.grad).abs().max().item() < 2e-2
+    _close(q.grad, q2.grad)
 
 
 def test_swiglu_fwd_bwd():
@@ -131,14 +140,14 @@ def test_swiglu_fwd_bwd():
     y = F.swiglu(x)
     g, u = x.detach().float().chunk(2, -1)
     ref = torch.nn.functional.silu(g) * u
-    assert (y.float() - ref).abs().max().item() < 2e-2
+    _close(y, ref)
 
     gy = _rand(64, 1408, seed=5)
     y.backward(gy)
     x2 = x.detach().float().requires_grad_(True)
     g2, u2 = x2.chunk(2, -1)
     (torch.nn.functional.silu(g2) * u2).backward(gy.float())
-    assert (x.grad.float() - x2.grad).abs().max().item() < 2e-2
+    _close(x.grad, x2.grad)
 
 
 def test_bias_gelu_fwd_bwd():
@@ -147,14 +156,14 @@ def test_bias_gelu_fwd_bwd():
     b = _rand(3072, seed=1).requires_grad_(True)
     y = F.bias_gelu(x, b)
     ref = F.eager_gelu(x.detach().float() + b.detach().float())
-    assert (y.float() - ref).abs().max().item() < 2e-2
+    _close(y, ref)
 
     gy = _rand(128, 3072, seed=2)
     y.backward(gy)
     x2 = x.detach().float().requires_grad_(True)
     b2 = b.detach().float().requires_grad_(True)
     F.eager_gelu(x2 + b2).backward(gy.float())
-    assert (x.grad.float() - x2.grad).abs().max().item() < 3e-2
+    _close(x.grad, x2.grad, 3e-2)
     assert ((b.grad.float() - b2.grad).abs().max()
             / b2.grad.abs().max()).item() < 3e-2
 
@@ -179,7 +188,8 @@ def test_fused_adamw_matches_torch():
         ref.grad = grad.float()
         opt.step()
     assert (master - ref.detach()).abs().max().item() < 1e-5
-    assert (out.float() - master).abs().max().item() < 1e-2  # bf16 roundoff
+    rel = (out.float() - master).abs() / master.abs().clamp(min=1.0)
+    assert rel.max().item() < 1e-2  # bf16 roundoff
 
 
 def test_attention_vs_sdpa():
@@ -196,4 +206,4 @@ def test_attention_vs_sdpa():
     causal = torch.ones(s, s, device="cuda", dtype=torch.bool).triu(1)
     scores = scores.masked_fill(causal, -1e9)
     ref = torch.softmax(scores, -1) @ vf
-    assert (out.float() - ref).abs().max().item() < 2e-2
+    _close(out, ref)
