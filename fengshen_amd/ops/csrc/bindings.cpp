@@ -47,6 +47,8 @@ void fs_bias_gelu(const void*, const void*, const void*, void*, long, int, int,
                   int, hipStream_t);
 void fs_fused_adamw(float*, const void*, float*, float*, void*, long, float,
                     float, float, float, float, int, int, int, hipStream_t);
+void fs_flash_attn_fwd(const void*, const void*, const void*, void*, float*,
+                       int, int, int, float, hipStream_t);
 }
 
 // ---------------------------------------------------------------------------
@@ -258,7 +260,24 @@ static void fused_adamw(at::Tensor master, at::Tensor grad, at::Tensor m,
                  outp ? fs_dtype(out_param) : FS_F32, cur_stream());
 }
 
+// q,k,v: [b, h, s, 128] bf16 contiguous, causal, seq %% 64 == 0
+static std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
+                                              at::Tensor v, double scale) {
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "flash_attn: bf16 only");
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == 128, "flash_attn: head_dim 128");
+  const int b = q.size(0), h = q.size(1), s = q.size(2);
+  TORCH_CHECK(s % 64 == 0, "flash_attn: seq must be divisible by 64");
+  auto o = at::empty_like(q);
+  auto lse = at::empty({b, h, s}, q.options().dtype(at::kFloat));
+  fs_flash_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                    lse.data_ptr<float>(), b, h, s, (float)scale,
+                    cur_stream());
+  return {o, lse};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("flash_attn_fwd", &flash_attn_fwd);
   mod.def("rms_norm_fwd", &rms_norm_fwd);
   mod.def("rms_norm_bwd", &rms_norm_bwd);
   mod.def("layer_norm_fwd", &layer_norm_fwd);

@@ -1,0 +1,242 @@
+// Flash-style fused causal attention FORWARD for MI355X (gfx950).
+//
+// Replaces the reference's flash_attn_cuda import (flash_attention.py:7) with
+// a native CDNA4 kernel: per Q-tile, stream K/V tiles through LDS, QK^T and
+// PV on MFMA (v_mfma_f32_16x16x32_bf16), online softmax in fp32 registers.
+// No S×S materialization (removes the reference's fused-softmax sk<=2048 cap
+// and the 3x HBM round trip of the bmm path).
+//
+// v2 structure:
+//   block = 4 waves; wave w owns 16 q-rows  => QBLK = 64
+//   KVBLK = 64; K staged [64][D] with XOR swizzle (guide §6 G4: row-major
+//   [*][128] bf16 is a 16-way ds_read_b128 bank conflict; byte ^= (row&7)<<4
+//   fixes it); V staged transposed [D][64+pad] so PV B-fragments read
+//   16B-contiguous kv runs.  D = 128, bf16, causal, seq % 64 == 0.
+// Outputs: O [b,h,s,D] and LSE [b,h,s] (softmax log-sum-exp, for backward).
+
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define FA_D 128
+#define FA_QBLK 64
+#define FA_KVBLK 64
+#define FA_WAVES 4
+#define FA_VPAD 8
+
+// XOR swizzle for k_lds rows (applies to 16B-aligned byte offsets)
+__device__ __forceinline__ int kswz(int row, int byte_in_row) {
+  return byte_in_row ^ ((row & 7) << 4);
+}
+
+__global__ __launch_bounds__(FA_WAVES * 64)
+void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
+                           const bf16_t* __restrict__ K,
+                           const bf16_t* __restrict__ V,
+                           bf16_t* __restrict__ O,
+                           float* __restrict__ LSE,
+                           int b, int h, int s, float scale) {
+  // raw bf16 bits in short storage (short=bf16_t assignment would convert)
+  __shared__ short k_lds[FA_KVBLK][FA_D];              // swizzled rows
+  __shared__ short vt_lds[FA_D][FA_KVBLK + FA_VPAD];   // V transposed
+  __shared__ short p_lds[FA_WAVES][16][FA_KVBLK + FA_VPAD];
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int qb = blockIdx.x;
+  const int head = blockIdx.y;
+  const int batch = blockIdx.z;
+
+  const long bh_off = ((long)batch * h + head) * s * FA_D;
+  const bf16_t* Qp = Q + bh_off;
+  const bf16_t* Kp = K + bh_off;
+  const bf16_t* Vp = V + bh_off;
+  bf16_t* Op = O + bh_off;
+
+  const int q0 = qb * FA_QBLK + wave * 16;
+
+  // ---- Q tile -> A-fragments (layout LA0: row=l%16, k=(l/16)*8+j;
+  // verified on HW by scripts/mfma_probe), pre-scaled -----------------------
+  bf16x8 q_frag[4];
+  {
+    const int row = lane & 15;
+    const int k0 = (lane >> 4) * 8;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 raw = *reinterpret_cast<const bf16x8*>(
+          Qp + (long)(q0 + row) * FA_D + c * 32 + k0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        unsigned short u = (unsigned short)raw[j];
+        float f = __uint_as_float(((unsigned int)u) << 16) * scale;
+        raw[j] = (short)__hip_bfloat16_raw(__float2bfloat16(f)).x;
+      }
+      q_frag[c] = raw;
+    }
+  }
+
+  float m_run[4], l_run[4];
+  f32x4 o_acc[8];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -INFINITY;
+    l_run[r] = 0.f;
+  }
+#pragma unroll
+  for (int t = 0; t < 8; ++t) o_acc[t] = f32x4{0, 0, 0, 0};
+
+  const int q_hi = qb * FA_QBLK + FA_QBLK - 1;
+  const int n_kv_tiles = (q_hi / FA_KVBLK) + 1;
+
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int k_base = kt * FA_KVBLK;
+    __syncthreads();
+    {
+      // stage K (swizzled) + V^T: 64*128 = 8192 elems each; 16B per thread
+      // per sweep => 4 sweeps
+      const int tid = threadIdx.x;
+      for (int i = tid * 8; i < FA_KVBLK * FA_D; i += 256 * 8) {
+        const int kr = i / FA_D;
+        const int kc = i % FA_D;
+        bf16x8 kk = *reinterpret_cast<const bf16x8*>(
+            Kp + (long)(k_base + kr) * FA_D + kc);
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(&k_lds[kr][0]) + kswz(kr, kc * 2)) = kk;
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
+            Vp + (long)(k_base + kr) * FA_D + kc);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vt_lds[kc + j][kr] = vv[j];
+      }
+    }
+    __syncthreads();
+
+    // ---- S = (sQ) @ K^T : four 16x16 n-tiles ------------------------------
+    f32x4 s_acc[4];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) s_acc[nt] = f32x4{0, 0, 0, 0};
+    {
+      const int col = lane & 15;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const int krow = nt * 16 + col;
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          const int d0 = c * 32 + (lane >> 4) * 8;
+          bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<char*>(&k_lds[krow][0]) + kswz(krow, d0 * 2));
+          s_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              q_frag[c], bfrag, s_acc[nt], 0, 0, 0);
+        }
+      }
+    }
+
+    // ---- causal mask + online softmax -------------------------------------
+    const int col = lane & 15;
+    float p[4][4];
+    float tile_max[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = q0 + (lane >> 4) * 4 + r;
+      float tm = -INFINITY;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const int kcol = k_base + nt * 16 + col;
+        float v = s_acc[nt][r];
+        if (kcol > qrow) v = -INFINITY;
+        p[nt][r] = v;
+        tm = fmaxf(tm, v);
+      }
+      tile_max[r] = tm;
+    }
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        tile_max[r] = fmaxf(tile_max[r], __shfl_xor(tile_max[r], off, 64));
+    }
+    float alpha[4], rowsum[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float m_new = fmaxf(m_run[r], tile_max[r]);
+      alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_new);
+      m_run[r] = m_new;
+      float ps = 0.f;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        float e = (p[nt][r] == -INFINITY) ? 0.f : __expf(p[nt][r] - m_new);
+        p[nt][r] = e;
+        ps += e;
+      }
+      rowsum[r] = ps;
+    }
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        rowsum[r] += __shfl_xor(rowsum[r], off, 64);
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      l_run[r] = l_run[r] * alpha[r] + rowsum[r];
+#pragma unroll
+    for (int t = 0; t < 8; ++t) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_acc[t][r] *= alpha[r];
+    }
+
+    // ---- P -> per-wave LDS, then PV ---------------------------------------
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = (lane >> 4) * 4 + r;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        p_lds[wave][row][nt * 16 + col] =
+            (short)__hip_bfloat16_raw(__float2bfloat16(p[nt][r])).x;
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);  // wave-local LDS visibility
+
+    // A-frags over kv (two K=32 chunks), B-frags from V^T
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      const int row = lane & 15;
+      const int kv0 = kc * 32 + (lane >> 4) * 8;
+      bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wave][row][kv0]);
+#pragma unroll
+      for (int t = 0; t < 8; ++t) {
+        const int dcol = t * 16 + (lane & 15);
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(&vt_lds[dcol][kv0]);
+        o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bfrag, o_acc[t],
+                                                           0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue -----------------------------------------------------------
+  const int col = lane & 15;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + (lane >> 4) * 4 + r;
+    const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+#pragma unroll
+    for (int t = 0; t < 8; ++t) {
+      Op[(long)qrow * FA_D + t * 16 + col] =
+          __float2bfloat16(o_acc[t][r] * inv_l);
+    }
+    if (col == 0 && LSE) {
+      LSE[((long)batch * h + head) * s + qrow] =
+          m_run[r] + logf(fmaxf(l_run[r], 1e-30f));
+    }
+  }
+}
+
+extern "C" void fs_flash_attn_fwd(const void* q, const void* k, const void* v,
+                                  void* o, float* lse, int b, int h, int s,
+                                  float scale, hipStream_t stream) {
+  dim3 grid(s / FA_QBLK, h, b);
+  dim3 block(FA_WAVES * 64);
+  hipLaunchKernelGGL(flash_attn_fwd_kernel, grid, block, 0, stream,
+                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
+                     (bf16_t*)o, lse, b, h, s, scale);
+}

@@ -66,10 +66,13 @@ __global__ void rms_norm_bwd_kernel(const T* __restrict__ gy,
                                     T* __restrict__ gx,
                                     float* __restrict__ gw_f32,
                                     int rows, int H) {
-  extern __shared__ float lds[];  // [2*H]: xhat row, gyw row
+  extern __shared__ float lds[];  // [3*H]: xhat row, gyw row, gw accumulator
   __shared__ float red[32];
   float* xhat_s = lds;
   float* gyw_s = lds + H;
+  float* gw_s = lds + 2 * H;
+  for (int c = threadIdx.x; c < H; c += blockDim.x) gw_s[c] = 0.f;
+  __syncthreads();
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* xr = x + (long)row * H;
     const T* gr = gy + (long)row * H;
@@ -88,7 +91,7 @@ __global__ void rms_norm_bwd_kernel(const T* __restrict__ gy,
         xhat_s[c + i] = xh;
         gyw_s[c + i] = gw_;
         dot += gw_ * xh;
-        atomicAdd(&gw_f32[c + i], gv[i] * xh);
+        gw_s[c + i] += gv[i] * xh;  // block-local; flushed once at the end
       }
     }
     dot = block_reduce_sum(dot, red) / H;
@@ -101,6 +104,8 @@ __global__ void rms_norm_bwd_kernel(const T* __restrict__ gy,
     }
     __syncthreads();
   }
+  for (int c = threadIdx.x; c < H; c += blockDim.x)
+    atomicAdd(&gw_f32[c], gw_s[c]);
 }
 
 // ===========================================================================
@@ -168,10 +173,17 @@ __global__ void layer_norm_bwd_kernel(const T* __restrict__ gy,
                                       float* __restrict__ gw_f32,
                                       float* __restrict__ gb_f32,
                                       int rows, int H) {
-  extern __shared__ float lds[];  // [2*H]
+  extern __shared__ float lds[];  // [4*H]: xhat, gyw, gw acc, gb acc
   __shared__ float red[32];
   float* xhat_s = lds;
   float* gyw_s = lds + H;
+  float* gw_s = lds + 2 * H;
+  float* gb_s = lds + 3 * H;
+  for (int c = threadIdx.x; c < H; c += blockDim.x) {
+    gw_s[c] = 0.f;
+    gb_s[c] = 0.f;
+  }
+  __syncthreads();
   for (int row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* xr = x + (long)row * H;
     const T* gr = gy + (long)row * H;
@@ -192,8 +204,8 @@ __global__ void layer_norm_bwd_kernel(const T* __restrict__ gy,
         gyw_s[c + i] = gw_;
         dot += gw_ * xh;
         gsum += gw_;
-        atomicAdd(&gw_f32[c + i], gv[i] * xh);
-        if (gb_f32) atomicAdd(&gb_f32[c + i], gv[i]);
+        gw_s[c + i] += gv[i] * xh;
+        gb_s[c + i] += gv[i];
       }
     }
     dot = block_reduce_sum(dot, red) / H;
@@ -206,6 +218,10 @@ __global__ void layer_norm_bwd_kernel(const T* __restrict__ gy,
       store8<T>(gxr + c, o);
     }
     __syncthreads();
+  }
+  for (int c = threadIdx.x; c < H; c += blockDim.x) {
+    atomicAdd(&gw_f32[c], gw_s[c]);
+    if (gb_f32) atomicAdd(&gb_f32[c], gb_s[c]);
   }
 }
 
@@ -473,24 +489,47 @@ __global__ void fused_adamw_kernel(float* __restrict__ master,
                                    OT* __restrict__ out_param,
                                    long n, float lr, float beta1, float beta2,
                                    float eps, float wd, float bc1, float bc2) {
+  // vectorized: 4 fp32 = 16B per lane for master/m/v (ZeRO shards are
+  // 128-element aligned so n % 4 == 0 in practice; tail handled scalar)
+  const long n4 = n & ~3L;
   const long stride = (long)gridDim.x * blockDim.x * 4;
-  for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 4; i < n;
+  for (long i = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 4; i < n4;
        i += stride) {
-    const int k = (int)min((long)4, n - i);
+    float4_t p = *reinterpret_cast<float4_t*>(master + i);
+    float4_t mj = *reinterpret_cast<float4_t*>(m + i);
+    float4_t vj = *reinterpret_cast<float4_t*>(v + i);
+    float g[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) g[j] = to_f32<GT>(grad[i + j]);
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      if (j >= k) break;
-      const long t = i + j;
-      float g = to_f32<GT>(grad[t]);
-      float p = master[t];
-      p *= (1.f - lr * wd);
-      float mj = m[t] = beta1 * m[t] + (1.f - beta1) * g;
-      float vj = v[t] = beta2 * v[t] + (1.f - beta2) * g * g;
-      const float denom = sqrtf(vj / bc2) + eps;
-      p -= lr / bc1 * (mj / denom);
-      master[t] = p;
-      if (out_param) out_param[t] = from_f32<OT>(p);
+      float pj = p[j] * (1.f - lr * wd);
+      float mm = beta1 * mj[j] + (1.f - beta1) * g[j];
+      float vv = beta2 * vj[j] + (1.f - beta2) * g[j] * g[j];
+      const float denom = sqrtf(vv / bc2) + eps;
+      pj -= lr / bc1 * (mm / denom);
+      p[j] = pj;
+      mj[j] = mm;
+      vj[j] = vv;
     }
+    *reinterpret_cast<float4_t*>(master + i) = p;
+    *reinterpret_cast<float4_t*>(m + i) = mj;
+    *reinterpret_cast<float4_t*>(v + i) = vj;
+    if (out_param) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) out_param[i + j] = from_f32<OT>(p[j]);
+    }
+  }
+  // scalar tail
+  for (long t = n4 + blockIdx.x * blockDim.x + threadIdx.x; t < n;
+       t += (long)gridDim.x * blockDim.x) {
+    float g = to_f32<GT>(grad[t]);
+    float pj = master[t] * (1.f - lr * wd);
+    float mm = m[t] = beta1 * m[t] + (1.f - beta1) * g;
+    float vv = v[t] = beta2 * v[t] + (1.f - beta2) * g * g;
+    pj -= lr / bc1 * (mm / (sqrtf(vv / bc2) + eps));
+    master[t] = pj;
+    if (out_param) out_param[t] = from_f32<OT>(pj);
   }
 }
 
@@ -526,7 +565,7 @@ void fs_rms_norm_bwd(const void* gy, const void* x, const void* w,
                      const float* invrms, void* gx, float* gw_f32, int rows,
                      int H, int dtype, hipStream_t s) {
   int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
-  size_t lds = (size_t)H * 8;
+  size_t lds = (size_t)H * 12;
   DISPATCH_DTYPE(dtype, T,
     hipLaunchKernelGGL((rms_norm_bwd_kernel<T>), dim3(grid), dim3(256), lds, s,
                        (const T*)gy, (const T*)x, (const T*)w, invrms, (T*)gx,
@@ -549,7 +588,7 @@ void fs_layer_norm_bwd(const void* gy, const void* x, const void* w,
                        float* gw_f32, float* gb_f32, int rows, int H, int dtype,
                        hipStream_t s) {
   int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
-  size_t lds = (size_t)H * 8;
+  size_t lds = (size_t)H * 16;
   DISPATCH_DTYPE(dtype, T,
     hipLaunchKernelGGL((layer_norm_bwd_kernel<T>), dim3(grid), dim3(256), lds, s,
                        (const T*)gy, (const T*)x, (const T*)w, mean, invstd,

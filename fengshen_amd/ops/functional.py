@@ -382,10 +382,16 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     if scale is None:
         scale = 1.0 / math.sqrt(hn)
     ext = get_ext() if use_hip(q) else None
-    if (ext is not None and hasattr(ext, "flash_attn_fwd") and causal
-            and mask is None and sq == sk and dropout_p == 0.0):
-        from fengshen_amd.ops.flash import flash_attention
-        return flash_attention(q, k, v, scale)
+    if ext is not None and hasattr(ext, "flash_attn_fwd"):
+        from fengshen_amd.ops.flash import flash_attention, flash_attn_supported
+        if flash_attn_supported(q, k, v, causal, mask, dropout_p):
+            # Under activation checkpointing the no-grad first pass takes the
+            # fused flash kernel (no SxS materialization); the grad-enabled
+            # recompute takes the prob-saving bmm path below so backward
+            # reuses P instead of rebuilding it.  Once the fused flash
+            # BACKWARD kernel lands, flash handles both.
+            if not (torch.is_grad_enabled() and q.requires_grad):
+                return flash_attention(q, k, v, scale)
     # bmm path: scores in [b*np, sq, sk]
     q2 = q.reshape(b * np_, sq, hn)
     k2 = k.reshape(b * np_, sk, hn)

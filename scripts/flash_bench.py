@@ -1,0 +1,28 @@
+import sys, os; sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import time, math, torch
+from fengshen_amd.ops.flash import flash_attention
+
+b, h, s, d = 8, 40, 2048, 128
+g = torch.Generator(device="cuda").manual_seed(0)
+q = torch.randn(b,h,s,d, generator=g, device="cuda").to(torch.bfloat16)
+k = torch.randn(b,h,s,d, generator=g, device="cuda").to(torch.bfloat16)
+v = torch.randn(b,h,s,d, generator=g, device="cuda").to(torch.bfloat16)
+scale = 1/math.sqrt(d)
+# correctness spot-check at s=512 slice
+qs, ks, vs = q[:1,:4,:512], k[:1,:4,:512], v[:1,:4,:512]
+out = flash_attention(qs.contiguous(), ks.contiguous(), vs.contiguous(), scale)
+sc = (qs.float() @ ks.float().transpose(-1,-2)) * scale
+causal = torch.ones(512,512,device="cuda",dtype=torch.bool).triu(1)
+ref = torch.softmax(sc.masked_fill(causal, float("-inf")), -1) @ vs.float()
+err = (out.float()-ref).abs().max()/ref.abs().max()
+print("refcheck rel err:", err.item())
+for _ in range(3):
+    flash_attention(q, k, v, scale)
+torch.cuda.synchronize()
+t0 = time.perf_counter(); N = 10
+for _ in range(N):
+    flash_attention(q, k, v, scale)
+torch.cuda.synchronize()
+dt = (time.perf_counter()-t0)/N
+flops = 2*2*b*h*s*s*d*0.5
+print(f"flash fwd: {dt*1000:.2f} ms, {flops/dt/1e12:.1f} TF")

@@ -207,3 +207,45 @@ def test_attention_vs_sdpa():
     scores = scores.masked_fill(causal, -1e9)
     ref = torch.softmax(scores, -1) @ vf
     _close(out, ref)
+
+
+@pytest.mark.parametrize("s", [64, 512, 2048])
+def test_flash_attention_fwd(s):
+    from fengshen_amd.ops.flash import flash_attention
+    from fengshen_amd.ops import functional as F
+    b, h, d = 2, 4, 128
+    q = _rand(b, h, s, d)
+    k = _rand(b, h, s, d, seed=1)
+    v = _rand(b, h, s, d, seed=2)
+    scale = 1.0 / math.sqrt(d)
+    out = flash_attention(q, k, v, scale)
+    qf, kf, vf = q.float(), k.float(), v.float()
+    scores = qf @ kf.transpose(-1, -2) * scale
+    causal = torch.ones(s, s, device="cuda", dtype=torch.bool).triu(1)
+    scores = scores.masked_fill(causal, float("-inf"))
+    ref = torch.softmax(scores, -1) @ vf
+    _close(out, ref)
+
+
+def test_flash_attention_bwd():
+    from fengshen_amd.ops.flash import flash_attention
+    b, h, s, d = 2, 4, 256, 128
+    q = _rand(b, h, s, d).requires_grad_(True)
+    k = _rand(b, h, s, d, seed=1).requires_grad_(True)
+    v = _rand(b, h, s, d, seed=2).requires_grad_(True)
+    scale = 1.0 / math.sqrt(d)
+    out = flash_attention(q, k, v, scale)
+    gy = _rand(b, h, s, d, seed=3)
+    out.backward(gy)
+
+    q2 = q.detach().float().requires_grad_(True)
+    k2 = k.detach().float().requires_grad_(True)
+    v2 = v.detach().float().requires_grad_(True)
+    scores = q2 @ k2.transpose(-1, -2) * scale
+    causal = torch.ones(s, s, device="cuda", dtype=torch.bool).triu(1)
+    scores = scores.masked_fill(causal, float("-inf"))
+    ref = torch.softmax(scores, -1) @ v2
+    ref.backward(gy.float())
+    _close(q.grad, q2.grad, 3e-2)
+    _close(k.grad, k2.grad, 3e-2)
+    _close(v.grad, v2.grad, 3e-2)
