@@ -27,7 +27,7 @@ def parse_args():
                    choices=["ziya-llama-13b", "wenzhong-gpt2-3.5b",
                             "erlangshen-1.3b", "llama-tiny"])
     p.add_argument("--seq_len", type=int, default=2048)
-    p.add_argument("--micro_batch", type=int, default=8)
+    p.add_argument("--micro_batch", type=int, default=16)
     p.add_argument("--zero_stage", type=int, default=2)
     p.add_argument("--lr", type=float, default=1e-5)
     return p.parse_args()

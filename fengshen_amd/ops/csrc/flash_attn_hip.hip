@@ -106,8 +106,11 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
             reinterpret_cast<char*>(&k_lds[kr][0]) + kswz(kr, kc * 2)) = kk;
         bf16x8 vv = *reinterpret_cast<const bf16x8*>(
             Vp + (long)(k_base + kr) * FA_D + kc);
+        // kv index XOR-swizzled by d (8-granular) so the transpose scatter
+        // spreads banks (unswizzled: 16 lanes of one K-row hit ONE bank)
 #pragma unroll
-        for (int j = 0; j < 8; ++j) vt_lds[kc + j][kr] = vv[j];
+        for (int j = 0; j < 8; ++j)
+          vt_lds[kc + j][kr ^ ((kc + j) & 0x38)] = vv[j];  // XOR bits 3-5 of d
       }
     }
     __syncthreads();
@@ -207,7 +210,8 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
       for (int t = 0; t < 8; ++t) {
         const int dcol = t * 16 + (lane & 15);
-        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(&vt_lds[dcol][kv0]);
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+            &vt_lds[dcol][kv0 ^ (dcol & 0x38)]);
         o_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bfrag, o_acc[t],
                                                            0, 0, 0);
       }
