@@ -88,8 +88,14 @@ def main():
         model.gradient_checkpointing_enable()
     model.train()
 
-    opt = ZeroOptimizer(model.parameters(), stage=args.zero_stage, lr=args.lr,
-                        betas=(0.9, 0.95), eps=1e-8, weight_decay=0.1)
+    if args.zero_stage == 3:
+        from fengshen_amd.parallel.zero3 import Zero3Engine
+        opt = Zero3Engine(model, lr=args.lr, betas=(0.9, 0.95), eps=1e-8,
+                          weight_decay=0.1)
+    else:
+        opt = ZeroOptimizer(model.parameters(), stage=args.zero_stage,
+                            lr=args.lr, betas=(0.9, 0.95), eps=1e-8,
+                            weight_decay=0.1)
 
     is_bert = args.model == "erlangshen-1.3b"
     b, s = args.micro_batch, args.seq_len

@@ -71,7 +71,14 @@ def configure_optimizers(pl_model, model_params=None):
     betas = (getattr(args, "adam_beta1", 0.9), getattr(args, "adam_beta2", 0.999))
     eps = getattr(args, "adam_epsilon", 1e-8)
 
-    if strategy is not None and strategy.kind == "zero" and strategy.stage >= 1:
+    if strategy is not None and strategy.kind == "zero" and strategy.stage == 3:
+        from fengshen_amd.parallel.zero3 import Zero3Engine
+        from fengshen_amd.parallel import groups as pgroups
+        optimizer = Zero3Engine(
+            pl_model, lr=args.learning_rate, betas=betas, eps=eps,
+            weight_decay=args.weight_decay,
+            process_group=pgroups.get_data_parallel_group())
+    elif strategy is not None and strategy.kind == "zero" and strategy.stage >= 1:
         from fengshen_amd.parallel.zero import ZeroOptimizer
         from fengshen_amd.parallel import groups as pgroups
         optimizer = ZeroOptimizer(

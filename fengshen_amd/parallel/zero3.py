@@ -1,0 +1,399 @@
+"""Native ZeRO-3: parameter + gradient + optimizer-state sharding.
+
+Replaces DeepSpeed ZeRO stage 3 (the reference's zero_optimization stage 3
+path, strategies/megatron_deepspeed.py:120-173).  MI355X-first design:
+
+  * The model is partitioned into *units* (each transformer layer + one
+    catch-all for embeddings/norms/head).  Each unit's params live as ONE
+    flat bf16 shard per DP rank (1/world of the unit).
+  * forward_pre_hook all-gathers the unit's flat buffer over RCCL/xGMI and
+    points the params into it; after forward (or after backward when grads
+    are flowing) the full buffer is freed — peak param memory is
+    params/world + a couple of in-flight units.
+  * In the no-grad forward pass the NEXT unit's all-gather is prefetched
+    (async) so gathers hide under compute.
+  * Gradients: post-accumulate hooks add param.grad into a flat staging
+    buffer; on the sync micro-step it is reduce-scattered and only this
+    rank's grad shard is kept.
+  * Optimizer states (fp32 master + Adam moments) exist only for the local
+    shard; the step is the fused AdamW HIP kernel on the shard — NO param
+    all-gather in the optimizer (params re-gather lazily at next forward).
+  * Weight decay: each unit's params are ordered [decay | no-decay] so the
+    shard splits into two fused-adamw calls (wd and 0).
+
+TP-compatible: shards over the DATA-parallel group only.
+"""
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from fengshen_amd.ops.adamw import fused_adamw_flat_
+from fengshen_amd.parallel.comm import all_gather_flat, reduce_scatter_flat
+
+logger = logging.getLogger(__name__)
+
+_ALIGN = 128
+
+
+def _pad(n: int, mult: int) -> int:
+    return ((n + mult - 1) // mult) * mult
+
+
+def _is_no_decay(name: str, p: torch.nn.Parameter) -> bool:
+    return p.dim() < 2 or "bias" in name or "norm" in name.lower() \
+        or "ln_" in name
+
+
+class _Unit:
+    def __init__(self, name: str, named_params: List[Tuple[str, nn.Parameter]],
+                 world: int, rank: int):
+        self.name = name
+        # order [decay..., no_decay...] so wd applies to a contiguous prefix
+        decay = [(n, p) for n, p in named_params if not _is_no_decay(n, p)]
+        nodecay = [(n, p) for n, p in named_params if _is_no_decay(n, p)]
+        self.params: List[nn.Parameter] = [p for _, p in decay + nodecay]
+        self.param_names = [n for n, _ in decay + nodecay]
+        self.decay_numel = sum(p.numel() for _, p in decay)
+        self.numel = sum(p.numel() for p in self.params)
+        self.numel_padded = _pad(self.numel, _ALIGN * world)
+        self.shard_numel = self.numel_padded // world
+        self.world = world
+        self.rank = rank
+        self.offsets: List[int] = []
+        off = 0
+        for p in self.params:
+            self.offsets.append(off)
+            off += p.numel()
+        self.shapes = [p.shape for p in self.params]
+        self.numels = [p.numel() for p in self.params]
+        self.dtype = self.params[0].dtype
+        self.device = self.params[0].device
+
+        # flatten current values, keep only this rank's shard
+        flat = torch.zeros(self.numel_padded, dtype=self.dtype, device=self.device)
+        for p, o in zip(self.params, self.offsets):
+            flat[o:o + p.numel()].copy_(p.data.reshape(-1))
+        start = rank * self.shard_numel
+        self.shard = flat[start:start + self.shard_numel].clone()
+        del flat
+
+        # optimizer states on the shard
+        self.master = self.shard.detach().float().clone()
+        self.exp_avg = torch.zeros_like(self.master)
+        self.exp_avg_sq = torch.zeros_like(self.master)
+
+        # release full params
+        self._stub = torch.empty(0, dtype=self.dtype, device=self.device)
+        for p in self.params:
+            p.data = self._stub
+        self.full: Optional[torch.Tensor] = None
+        self.gather_work = None
+        self.grad_stage: Optional[torch.Tensor] = None
+        self.grad_shard: Optional[torch.Tensor] = None
+        self.rs_work = None
+        self.grad_ready = 0
+        self.gathered = False
+
+    # ------------------------------------------------------------------
+    def launch_gather(self, group, async_op=True):
+        if self.gathered or self.gather_work is not None:
+            return
+        if self.world == 1:
+            self.full = self.shard
+            self._attach()
+            return
+        self.full = torch.empty(self.numel_padded, dtype=self.dtype,
+                                device=self.device)
+        self.gather_work = all_gather_flat(self.full, self.shard,
+                                           group, self.world, async_op=async_op)
+
+    def ensure_gathered(self, group):
+        if self.gathered:
+            return
+        if self.gather_work is None and self.full is None:
+            self.launch_gather(group, async_op=False)
+        if self.gather_work is not None:
+            self.gather_work.wait()
+            self.gather_work = None
+        self._attach()
+
+    def _attach(self):
+        for p, o, shp, n in zip(self.params, self.offsets, self.shapes,
+                                self.numels):
+            p.data = self.full[o:o + n].view(shp)
+        self.gathered = True
+
+    def release(self):
+        if not self.gathered:
+            return
+        for p in self.params:
+            p.data = self._stub
+        if self.world > 1:
+            self.full = None  # caching allocator reclaims
+        self.gathered = False
+
+    # ------------------------------------------------------------------
+    def add_grad(self, p: nn.Parameter):
+        if self.grad_stage is None:
+            self.grad_stage = torch.zeros(self.numel_padded, dtype=self.dtype,
+                                          device=self.device)
+        i = next(j for j, q in enumerate(self.params) if q is p)
+        o = self.offsets[i]
+        self.grad_stage[o:o + self.numels[i]].add_(p.grad.reshape(-1))
+        p.grad = None
+
+    def launch_reduce_scatter(self, group):
+        if self.rs_work is not None or self.grad_stage is None:
+            return
+        if self.world == 1:
+            self.grad_shard = self.grad_stage
+            return
+        self.grad_stage.div_(self.world)
+        if self.grad_shard is None:
+            self.grad_shard = torch.empty(self.shard_numel, dtype=self.dtype,
+                                          device=self.device)
+        self.rs_work = reduce_scatter_flat(self.grad_stage, self.grad_shard,
+                                           group, self.rank, async_op=True)
+
+    def finish_reduce_scatter(self):
+        if self.rs_work is not None:
+            self.rs_work.wait()
+            self.rs_work = None
+        if self.world > 1:
+            self.grad_stage = None
+
+
+class Zero3Engine(torch.optim.Optimizer):
+    """Optimizer-compatible engine owning sharded params + step.
+
+    Subclasses torch.optim.Optimizer so LR schedulers accept it; the base
+    param_groups are replaced with a single group carrying the hyperparams
+    (no live params inside — states live in the units).
+    """
+
+    def __init__(self, module: nn.Module, lr: float = 1e-4,
+                 betas=(0.9, 0.999), eps: float = 1e-8,
+                 weight_decay: float = 0.01, process_group=None,
+                 unit_classes: Optional[Sequence[type]] = None,
+                 prefetch: bool = True, clip_grad: float = 0.0):
+        # satisfy Optimizer.__init__ with a dummy param, then take over
+        dummy = nn.Parameter(torch.zeros(1))
+        super().__init__([dummy], dict(lr=lr, betas=betas, eps=eps,
+                                       weight_decay=weight_decay))
+        self.module = module
+        self.group = process_group
+        if self.group is None and dist.is_available() and dist.is_initialized():
+            from fengshen_amd.parallel import groups as pg
+            self.group = pg.get_data_parallel_group()
+        self.world = dist.get_world_size(self.group) if (
+            dist.is_available() and dist.is_initialized()) else 1
+        self.rank = dist.get_rank(self.group) if (
+            dist.is_available() and dist.is_initialized()) else 0
+        self.param_groups = [
+            dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay,
+                 params=[])]
+        self.clip_grad = clip_grad
+        self.prefetch = prefetch
+        self._sync = True
+        self._step_count = 0
+        self._last_grad_norm = None
+
+        if unit_classes is None:
+            from fengshen_amd.models.layers import ParallelTransformerLayer
+            unit_classes = (ParallelTransformerLayer,)
+
+        # ---- partition into units ----------------------------------------
+        unit_modules: List[Tuple[str, nn.Module]] = []
+        claimed = set()
+        for name, m in module.named_modules():
+            if isinstance(m, tuple(unit_classes)):
+                unit_modules.append((name, m))
+                for p in m.parameters():
+                    claimed.add(id(p))
+        rest = [(n, p) for n, p in module.named_parameters()
+                if id(p) not in claimed and p.requires_grad]
+
+        self.units: List[_Unit] = []
+        self._unit_of_module: Dict[int, _Unit] = {}
+        self._unit_of_param: Dict[int, _Unit] = {}
+        for name, m in unit_modules:
+            named = [(f"{name}.{pn}", p) for pn, p in m.named_parameters()
+                     if p.requires_grad]
+            if not named:
+                continue
+            u = _Unit(name, named, self.world, self.rank)
+            self.units.append(u)
+            self._unit_of_module[id(m)] = u
+            for p in u.params:
+                self._unit_of_param[id(p)] = u
+            self._register_module_hooks(m, u)
+        if rest:
+            u = _Unit("(rest)", rest, self.world, self.rank)
+            self.units.append(u)
+            self._unit_of_module[id(module)] = u
+            for p in u.params:
+                self._unit_of_param[id(p)] = u
+            self._register_module_hooks(module, u, is_root=True)
+
+        # grad hooks
+        for u in self.units:
+            for p in u.params:
+                p.register_post_accumulate_grad_hook(self._grad_hook(u))
+
+        logger.info("Zero3Engine: %d units, world %d, %.2f GB shard/rank",
+                    len(self.units),
+                    self.world,
+                    sum(u.shard_numel for u in self.units) * 2 / 2 ** 30)
+
+    # ------------------------------------------------------------------
+    def _register_module_hooks(self, m: nn.Module, u: _Unit, is_root=False):
+        def pre(_m, _inp):
+            u.ensure_gathered(self.group)
+            if self.prefetch and not torch.is_grad_enabled():
+                i = self.units.index(u)
+                if i + 1 < len(self.units):
+                    self.units[i + 1].launch_gather(self.group)
+            return None
+
+        def post(_m, _inp, _out):
+            if not torch.is_grad_enabled():
+                u.release()
+            return None
+
+        m.register_forward_pre_hook(pre)
+        if not is_root:
+            # grad-enabled passes release in the grad hook instead (releasing
+            # any earlier races with wgrad AccumulateGrad shape checks); the
+            # root unit (embeddings early, head late) stays gathered all pass
+            m.register_forward_hook(post)
+
+    def _grad_hook(self, u: _Unit):
+        def hook(p):
+            u.add_grad(p)
+            u.grad_ready += 1
+            if u.grad_ready == len(u.params):
+                u.grad_ready = 0
+                if u.name != "(rest)":
+                    u.release()  # all wgrads in: params no longer needed
+                if self._sync:
+                    u.launch_reduce_scatter(self.group)
+        return hook
+
+    # ------------------------------------------------------------------
+    def set_sync(self, flag: bool):
+        self._sync = flag
+
+    @torch.no_grad()
+    def _grad_norm(self) -> torch.Tensor:
+        dev = self.units[0].device
+        sq = torch.zeros((), dtype=torch.float32, device=dev)
+        for u in self.units:
+            if u.grad_shard is not None:
+                g = u.grad_shard.float()
+                sq += (g * g).sum()
+        if self.world > 1:
+            dist.all_reduce(sq, group=self.group)
+        return sq.sqrt()
+
+    @torch.no_grad()
+    def step(self):
+        # flush any units whose RS didn't launch (unused params etc.)
+        for u in self.units:
+            if u.grad_stage is not None and u.rs_work is None and self.world > 1:
+                u.launch_reduce_scatter(self.group)
+        for u in self.units:
+            u.finish_reduce_scatter()
+        self._step_count += 1
+
+        grad_norm = None
+        clip_coef = None
+        if self.clip_grad and self.clip_grad > 0:
+            grad_norm = self._grad_norm()
+            coef = self.clip_grad / (grad_norm + 1e-6)
+            if float(coef) < 1.0:
+                clip_coef = coef
+        g = self.param_groups[0]
+        beta1, beta2 = g["betas"]
+        for u in self.units:
+            if u.grad_shard is None:
+                continue
+            if clip_coef is not None:
+                u.grad_shard.mul_(clip_coef)
+            # shard range [lo, hi); decay prefix is [0, decay_numel)
+            lo = self.rank * u.shard_numel
+            split = min(max(u.decay_numel - lo, 0), u.shard_numel)
+            if split > 0:
+                fused_adamw_flat_(
+                    u.master[:split], u.grad_shard[:split],
+                    u.exp_avg[:split], u.exp_avg_sq[:split], u.shard[:split],
+                    lr=g["lr"], beta1=beta1, beta2=beta2, eps=g["eps"],
+                    weight_decay=g["weight_decay"], step=self._step_count)
+            if split < u.shard_numel:
+                fused_adamw_flat_(
+                    u.master[split:], u.grad_shard[split:],
+                    u.exp_avg[split:], u.exp_avg_sq[split:], u.shard[split:],
+                    lr=g["lr"], beta1=beta1, beta2=beta2, eps=g["eps"],
+                    weight_decay=0.0, step=self._step_count)
+        # any still-gathered unit holds pre-step values: force re-gather
+        for u in self.units:
+            u.release()
+        self._last_grad_norm = grad_norm
+
+    @torch.no_grad()
+    def zero_grad(self, set_to_none: bool = False):
+        for u in self.units:
+            if self.world == 1 and u.grad_stage is not None:
+                u.grad_stage.zero_()
+            else:
+                u.grad_stage = None
+            u.grad_ready = 0
+            u.rs_work = None
+
+    # ------------------------------------------------------------------
+    class _GatherAll:
+        def __init__(self, engine):
+            self.engine = engine
+
+        def __enter__(self):
+            for u in self.engine.units:
+                u.ensure_gathered(self.engine.group)
+            return self.engine
+
+        def __exit__(self, *a):
+            for u in self.engine.units:
+                u.release()
+
+    def gathered_params(self):
+        """Context manager: all params materialized (for state_dict etc.)."""
+        return Zero3Engine._GatherAll(self)
+
+    def state_dict(self) -> dict:
+        return {
+            "stage": 3,
+            "step": self._step_count,
+            "world_size": self.world,
+            "rank": self.rank,
+            "param_groups": [
+                {k: v for k, v in g.items() if k != "params"}
+                for g in self.param_groups],
+            "units": [
+                {"master": u.master, "exp_avg": u.exp_avg,
+                 "exp_avg_sq": u.exp_avg_sq, "shard": u.shard}
+                for u in self.units],
+        }
+
+    def load_state_dict(self, sd: dict):
+        assert sd["world_size"] == self.world, "ZeRO-3 resume needs same world"
+        self._step_count = sd["step"]
+        for g, gs in zip(self.param_groups, sd["param_groups"]):
+            g.update(gs)
+        for u, us in zip(self.units, sd["units"]):
+            u.master.copy_(us["master"])
+            u.exp_avg.copy_(us["exp_avg"])
+            u.exp_avg_sq.copy_(us["exp_avg_sq"])
+            u.shard.copy_(us["shard"])

@@ -20,6 +20,7 @@ from fengshen_amd.parallel import groups as pgroups
 from fengshen_amd.parallel.ddp import GradReducer
 from fengshen_amd.parallel.random import model_parallel_manual_seed
 from fengshen_amd.parallel.zero import ZeroOptimizer
+from fengshen_amd.parallel.zero3 import Zero3Engine
 from fengshen_amd.ops.adamw import FusedAdamW
 
 logger = logging.getLogger(__name__)
@@ -92,9 +93,20 @@ class Strategy:
     def setup_optimizers(self, model, optimizer, scheduler_cfg):
         """Wrap/convert the module's optimizer for this strategy."""
         dp_world = self.data_parallel_world_size
-        if isinstance(optimizer, ZeroOptimizer):
+        if isinstance(optimizer, (ZeroOptimizer, Zero3Engine)):
             self.zero_optimizer = optimizer
             return optimizer, scheduler_cfg
+        if self.kind == "zero" and self.stage == 3 \
+                and isinstance(optimizer, _ADAM_FAMILY):
+            d = optimizer.defaults
+            eng = Zero3Engine(
+                model, lr=d.get("lr", 1e-3), betas=d.get("betas", (0.9, 0.999)),
+                eps=d.get("eps", 1e-8), weight_decay=d.get("weight_decay", 0.0),
+                process_group=pgroups.get_data_parallel_group())
+            self.zero_optimizer = eng
+            if scheduler_cfg is not None:
+                scheduler_cfg["scheduler"].optimizer = eng
+            return eng, scheduler_cfg
         use_zero = self.kind == "zero" or (self.kind == "ddp" and dp_world > 1)
         if use_zero and isinstance(optimizer, _ADAM_FAMILY):
             d = optimizer.defaults
@@ -111,7 +123,7 @@ class Strategy:
             if scheduler_cfg is not None:
                 scheduler_cfg["scheduler"].optimizer = zopt
             return zopt, scheduler_cfg
-        if self.kind == "zero" and self.stage >= 1:
+        if self.kind == "zero" and 1 <= self.stage <= 2:
             raise ValueError(
                 f"ZeRO stage {self.stage} requires an Adam-family optimizer, "
                 f"got {type(optimizer).__name__}")

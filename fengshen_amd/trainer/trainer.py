@@ -390,8 +390,9 @@ class Trainer:
 
     def _clip_and_step(self, model):
         from fengshen_amd.parallel.zero import ZeroOptimizer
+        from fengshen_amd.parallel.zero3 import Zero3Engine
 
-        if isinstance(self.optimizer, ZeroOptimizer):
+        if isinstance(self.optimizer, (ZeroOptimizer, Zero3Engine)):
             self.optimizer.clip_grad = self.gradient_clip_val or 0.0
             self.optimizer.step()
             if getattr(self.optimizer, "_last_grad_norm", None) is not None:
