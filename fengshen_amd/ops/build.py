@@ -65,7 +65,35 @@ def build(force: bool = False, verbose: bool = True) -> str:
     shutil.copy2(built, _TARGET)
     if verbose:
         print(f"[fengshen_amd.ops.build] built {_TARGET}")
+    build_data_helpers(force=force, verbose=verbose)
     return _TARGET
+
+
+_DATA_DIR = os.path.normpath(os.path.join(_OPS_DIR, "..", "data"))
+_HELPERS_SRC = os.path.join(_DATA_DIR, "csrc", "helpers.cpp")
+_HELPERS_TARGET = os.path.join(_DATA_DIR, "_helpers.so")
+
+
+def build_data_helpers(force: bool = False, verbose: bool = True) -> str:
+    """CPU pybind11 index-builder extension (ref helpers.cpp equivalent)."""
+    if not force and os.path.exists(_HELPERS_TARGET) and \
+            os.path.getmtime(_HELPERS_TARGET) > os.path.getmtime(_HELPERS_SRC):
+        return _HELPERS_TARGET
+    import subprocess
+    import sysconfig
+
+    import pybind11
+
+    cmd = [
+        "c++", "-O3", "-std=c++17", "-shared", "-fPIC",
+        f"-I{pybind11.get_include()}",
+        f"-I{sysconfig.get_paths()['include']}",
+        _HELPERS_SRC, "-o", _HELPERS_TARGET,
+    ]
+    subprocess.check_call(cmd)
+    if verbose:
+        print(f"[fengshen_amd.ops.build] built {_HELPERS_TARGET}")
+    return _HELPERS_TARGET
 
 
 if __name__ == "__main__":

@@ -1,0 +1,186 @@
+"""Data layer tests: mmap dataset roundtrip, GPT sample windows, blending,
+samplers, masking utils, collators."""
+import numpy as np
+import pytest
+import torch
+
+from fengshen_amd.data.indexed_dataset import (
+    MMapIndexedDataset,
+    MMapIndexedDatasetBuilder,
+    best_fitting_dtype,
+)
+
+
+class FakeTokenizer:
+    """Minimal HF-ish tokenizer over characters (for collator tests)."""
+
+    def __init__(self):
+        chars = [chr(c) for c in range(0x4E00, 0x4E00 + 200)] + \
+            list("abcdefghijklmnopqrstuvwxyz0123456789.,!? :\n<>_")
+        self._vocab = {"[PAD]": 0, "[CLS]": 1, "[SEP]": 2, "[MASK]": 3,
+                       "[UNK]": 4, "<s>": 5, "</s>": 6}
+        for c in chars:
+            self._vocab.setdefault(c, len(self._vocab))
+        self.pad_token_id = 0
+        self.cls_token_id = 1
+        self.sep_token_id = 2
+        self.mask_token_id = 3
+        self.bos_token_id = 5
+        self.eos_token_id = 6
+
+    def get_vocab(self):
+        return self._vocab
+
+    def encode(self, text, add_special_tokens=True):
+        ids = [self._vocab.get(c, 4) for c in text]
+        if add_special_tokens:
+            return [self.cls_token_id] + ids + [self.sep_token_id]
+        return ids
+
+
+def test_mmap_dataset_roundtrip(tmp_path):
+    prefix = str(tmp_path / "corpus")
+    builder = MMapIndexedDatasetBuilder(prefix + ".bin",
+                                        dtype=best_fitting_dtype(30000))
+    docs = [[list(range(10, 20)), list(range(30, 35))],
+            [list(range(100, 140))]]
+    for doc in docs:
+        for sent in doc:
+            builder.add_item(np.array(sent))
+        builder.end_document()
+    builder.finalize(prefix + ".idx")
+
+    ds = MMapIndexedDataset(prefix)
+    assert len(ds) == 3
+    assert ds.get(0).tolist() == list(range(10, 20))
+    assert ds.get(1).tolist() == list(range(30, 35))
+    assert ds.get(2).tolist() == list(range(100, 140))
+    assert ds.get(2, offset=5, length=3).tolist() == [105, 106, 107]
+    assert ds.doc_idx.tolist() == [0, 2, 3]
+    assert ds.dtype == np.uint16
+
+
+def test_gpt_dataset_windows(tmp_path):
+    from fengshen_amd.data.gpt_dataset import GPTDataset
+    prefix = str(tmp_path / "gpt")
+    builder = MMapIndexedDatasetBuilder(prefix + ".bin", dtype=np.int32)
+    rng = np.random.RandomState(0)
+    total = 0
+    for _ in range(7):
+        n = rng.randint(5, 40)
+        builder.add_item(rng.randint(0, 1000, size=n))
+        builder.end_document()
+        total += n
+    builder.finalize(prefix + ".idx")
+    ds = MMapIndexedDataset(prefix)
+    g = GPTDataset(ds, seq_length=16, num_epochs=2, seed=1)
+    assert len(g) == (2 * total - 1) // 16
+    for i in range(len(g)):
+        item = g[i]
+        assert item["input_ids"].shape == (16,)
+        assert item["labels"].shape == (16,)
+    # shifted-by-one relationship
+    it = g[0]
+    full = torch.cat([it["input_ids"][:1], it["labels"]])
+    assert torch.equal(it["input_ids"], full[:-1])
+
+
+def test_blendable_dataset_weights():
+    from fengshen_amd.data.gpt_dataset import BlendableDataset
+
+    class Const(torch.utils.data.Dataset):
+        def __init__(self, v, n=100):
+            self.v, self.n = v, n
+
+        def __len__(self):
+            return self.n
+
+        def __getitem__(self, i):
+            return self.v
+
+    b = BlendableDataset([Const(0), Const(1)], [0.75, 0.25], size=400)
+    vals = [b[i] for i in range(400)]
+    assert abs(sum(vals) / 400 - 0.25) < 0.02
+
+
+def test_mlm_sop_collator():
+    from fengshen_amd.data.collators import MlmSopCollator
+    tk = FakeTokenizer()
+    coll = MlmSopCollator(tk, max_seq_length=64)
+    texts = [{"text": "今天天气不错。我们出去玩吧！明天继续工作。"},
+             {"text": "他说：好的。然后就走了。再也没有回来。"}]
+    out = coll(texts)
+    assert out["input_ids"].shape == (2, 64)
+    assert out["labels"].shape == (2, 64)
+    assert out["next_sentence_label"].shape == (2,)
+    # some positions masked, labels set only there
+    masked = (out["labels"] != -100)
+    assert masked.any()
+    # every masked label is a valid token id
+    assert (out["labels"][masked] >= 0).all()
+    # CLS at position 0, unmasked rows start with cls
+    assert (out["input_ids"][:, 0] == tk.cls_token_id).all()
+
+
+def test_sft_collator_prompt_masking():
+    from fengshen_amd.data.collators import SftCollator
+    tk = FakeTokenizer()
+    coll = SftCollator(tk, max_seq_length=128)
+    out = coll([{"query": "1+1?", "answer": "2"},
+                {"query": ["hi", "more"], "answer": ["yo", "ok"]}])
+    assert out["input_ids"].shape == out["labels"].shape
+    labels = out["labels"]
+    # prompt region must be -100; answers must appear in labels
+    assert (labels[0] == -100).sum() > 2
+    assert (labels[0] != -100).sum() >= 2  # answer + eos
+
+
+def test_t5_span_collator():
+    from fengshen_amd.data.collators import T5SpanCollator
+    tk = FakeTokenizer()
+    coll = T5SpanCollator(tk, max_seq_length=64)
+    out = coll([{"text": "abcdefghij" * 10}, {"text": "0123456789" * 8}])
+    assert out["input_ids"].dim() == 2
+    assert out["labels"].dim() == 2
+    # corrupted input shorter than raw
+    assert out["input_ids"].shape[1] <= 64
+
+
+def test_causal_collator():
+    from fengshen_amd.data.collators import CausalCollator
+    tk = FakeTokenizer()
+    coll = CausalCollator(tk, max_seq_length=32)
+    out = coll([{"text": "hello world"}, {"text": "a"}])
+    assert out["input_ids"].shape == out["labels"].shape
+    assert (out["labels"][1] == -100).sum() > 0  # padding masked
+
+
+def test_sampler_exact_resume():
+    from fengshen_amd.data.universal_sampler import PretrainingRandomSampler
+    full = PretrainingRandomSampler(
+        total_samples=100, consumed_samples=0, micro_batch_size=4,
+        data_parallel_rank=0, data_parallel_size=2, epoch=0, seed=7)
+    batches = list(full)
+    resumed = PretrainingRandomSampler(
+        total_samples=100, consumed_samples=24, micro_batch_size=4,
+        data_parallel_rank=0, data_parallel_size=2, epoch=0, seed=7)
+    resumed_batches = list(resumed)
+    assert batches[3:] == resumed_batches
+
+
+def test_masking_utils():
+    from fengshen_amd.data.data_utils import (
+        ChineseSentenceSplitter, create_masked_lm_predictions)
+    sp = ChineseSentenceSplitter()
+    s = sp.tokenize("你好。今天怎么样？很好！")
+    assert len(s) == 3
+    rng = np.random.RandomState(0)
+    tokens = list(range(10, 40))
+    vocab = list(range(5, 100))
+    id2tok = {i: f"t{i}" for i in vocab + tokens}
+    out, pos, lab = create_masked_lm_predictions(
+        tokens, vocab, id2tok, 0.3, cls_id=1, sep_id=2, mask_id=3,
+        max_predictions_per_seq=10, np_rng=rng)
+    assert len(pos) == len(lab) > 0
+    for p, l in zip(pos, lab):
+        assert tokens[p] == l
