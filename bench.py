@@ -28,7 +28,7 @@ def parse_args():
                             "erlangshen-1.3b", "llama-tiny"])
     p.add_argument("--seq_len", type=int, default=2048)
     p.add_argument("--micro_batch", type=int, default=16)
-    p.add_argument("--zero_stage", type=int, default=2)
+    p.add_argument("--zero_stage", type=int, default=3)
     p.add_argument("--lr", type=float, default=1e-5)
     return p.parse_args()
 
