@@ -1,0 +1,181 @@
+"""Model-zoo batch 2: T5, BART, RoFormer, CLIP, UBERT, UniMC, tagging heads,
+CRF, metrics."""
+import torch
+
+from tests.test_data import FakeTokenizer
+
+
+def test_t5_train_and_generate():
+    from fengshen_amd.models.t5.modeling_t5 import (
+        T5ForConditionalGeneration, t5_tiny_config)
+    torch.manual_seed(0)
+    m = T5ForConditionalGeneration(t5_tiny_config())
+    src = torch.randint(3, 256, (2, 20))
+    lab = torch.randint(3, 256, (2, 12))
+    mask = torch.ones_like(src)
+    mask[:, -4:] = 0
+    out = m(input_ids=src, attention_mask=mask, labels=lab)
+    assert out.loss.isfinite()
+    out.loss.backward()
+    m.eval()
+    gen = m.generate(src, attention_mask=mask, max_new_tokens=6,
+                     do_sample=False)
+    assert gen.shape[0] == 2
+
+
+def test_bart_train_and_generate():
+    from fengshen_amd.models.bart.modeling_bart import (
+        BartForConditionalGeneration, bart_tiny_config)
+    torch.manual_seed(0)
+    m = BartForConditionalGeneration(bart_tiny_config())
+    src = torch.randint(3, 256, (2, 18))
+    lab = torch.randint(3, 256, (2, 10))
+    out = m(input_ids=src, labels=lab)
+    assert out.loss.isfinite()
+    out.loss.backward()
+    m.eval()
+    gen = m.generate(src, max_new_tokens=5, do_sample=False)
+    assert gen.shape[0] == 2
+
+
+def test_roformer_mlm():
+    from fengshen_amd.models.roformer.modeling_roformer import (
+        RoFormerForMaskedLM, roformer_tiny_config)
+    torch.manual_seed(0)
+    m = RoFormerForMaskedLM(roformer_tiny_config())
+    ids = torch.randint(3, 256, (2, 16))
+    labels = ids.clone()
+    labels[:, ::2] = -100
+    out = m(ids, labels=labels)
+    assert out.loss.isfinite()
+    out.loss.backward()
+
+
+def test_taiyi_clip_contrastive():
+    from fengshen_amd.models.clip.modeling_taiyi_clip import (
+        TaiyiCLIPModel, taiyi_clip_tiny_config)
+    torch.manual_seed(0)
+    m = TaiyiCLIPModel(taiyi_clip_tiny_config())
+    ids = torch.randint(3, 256, (4, 12))
+    pix = torch.randn(4, 3, 32, 32)
+    out = m(input_ids=ids, pixel_values=pix, return_loss=True)
+    assert out.loss.isfinite()
+    assert out.logits_per_image.shape == (4, 4)
+    out.loss.backward()
+
+
+def test_ubert_span_extraction():
+    from fengshen_amd.models.ubert.modeling_ubert import UbertModel
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    torch.manual_seed(0)
+    m = UbertModel(bert_tiny_config())
+    b, nl, s = 2, 3, 12
+    ids = torch.randint(3, 256, (b, nl, s))
+    labels = torch.zeros(b, nl, s, s)
+    labels[:, :, 2, 4] = 1
+    mask = torch.ones(b, nl, s, s)
+    out = m(ids, span_labels=labels, span_mask=mask)
+    assert out.loss.isfinite()
+    out.loss.backward()
+    res = m.extract(ids, threshold=0.9)
+    assert len(res) == b and len(res[0]) == nl
+
+
+def test_unimc_option_choice():
+    from fengshen_amd.models.unimc.modeling_unimc import UniMCModel
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    torch.manual_seed(0)
+    m = UniMCModel(bert_tiny_config(), yes_token_id=5)
+    ids = torch.randint(3, 256, (2, 20))
+    opt_pos = torch.tensor([[1, 5, 9], [2, 6, 10]])
+    labels = torch.tensor([0, 2])
+    out = m(ids, option_positions=opt_pos, labels=labels)
+    assert out.loss.isfinite()
+    out.loss.backward()
+    pred = m.predict(ids, None, None, opt_pos)
+    assert pred.shape == (2,)
+
+
+def test_crf_learns_and_decodes():
+    from fengshen_amd.models.tagging_models.crf import CRF
+    torch.manual_seed(0)
+    crf = CRF(num_tags=4, batch_first=True)
+    emissions = torch.randn(3, 7, 4)
+    tags = torch.randint(0, 4, (3, 7))
+    mask = torch.ones(3, 7, dtype=torch.bool)
+    mask[1, 5:] = False
+    nll = crf(emissions, tags, mask=mask)
+    assert nll.isfinite()
+    nll.backward()
+    paths = crf.decode(emissions, mask=mask)
+    assert len(paths) == 3
+    assert len(paths[1]) == 5  # masked length
+
+    # overfit sanity: the decoded path converges to the target tags
+    crf2 = CRF(num_tags=3)
+    em = torch.zeros(1, 5, 3)
+    target = torch.tensor([[0, 1, 2, 1, 0]])
+    em.requires_grad_(True)
+    opt = torch.optim.Adam([em] + list(crf2.parameters()), lr=0.1)
+    for _ in range(100):
+        loss = crf2(em, target)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+    assert crf2.decode(em.detach()) == [[0, 1, 2, 1, 0]]
+
+
+def test_tagging_heads():
+    from fengshen_amd.models.tagging_models.bert_for_tagging import (
+        BertLinear, BertCrf, BertSpan, BertBiaffine)
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    torch.manual_seed(0)
+    cfg = bert_tiny_config()
+    ids = torch.randint(3, 256, (2, 10))
+    mask = torch.ones_like(ids)
+    labels = torch.randint(0, 5, (2, 10))
+
+    m = BertLinear(cfg, num_labels=5, loss_type="focal")
+    out = m(ids, attention_mask=mask, labels=labels)
+    assert out.loss.isfinite()
+    out.loss.backward()
+
+    m = BertCrf(cfg, num_labels=5)
+    out = m(ids, attention_mask=mask, labels=labels, decode=True)
+    assert out.loss.isfinite() and len(out.predictions) == 2
+    out.loss.backward()
+
+    m = BertSpan(cfg, num_labels=5)
+    out = m(ids, attention_mask=mask, start_positions=labels,
+            end_positions=labels)
+    assert out.loss.isfinite()
+    out.loss.backward()
+
+    m = BertBiaffine(cfg, num_labels=5)
+    span_labels = torch.randint(0, 5, (2, 10, 10))
+    out = m(ids, attention_mask=mask, span_labels=span_labels)
+    assert out.loss.isfinite()
+    out.loss.backward()
+
+
+def test_ner_metrics():
+    from fengshen_amd.metric.metric import SeqEntityScore, metrics_mlm_acc
+    id2label = {0: "O", 1: "B-PER", 2: "I-PER", 3: "B-LOC", 4: "I-LOC"}
+    scorer = SeqEntityScore(id2label, markup="bio")
+    truth = [[0, 1, 2, 0, 3, 4]]
+    pred = [[0, 1, 2, 0, 3, 0]]
+    scorer.update(truth, pred)
+    overall, per_class = scorer.result()
+    assert overall["acc"] == 0.5 and overall["recall"] == 0.5
+    assert "PER" in per_class
+
+    logits = torch.zeros(2, 4, 10)
+    logits[..., 3] = 1.0
+    labels = torch.full((2, 4), -100)
+    labels[0, 1] = 3
+    labels[1, 2] = 5
+    acc = metrics_mlm_acc(logits, labels)
+    assert abs(acc.item() - 0.5) < 1e-6
