@@ -1,0 +1,55 @@
+"""Information extraction pipeline (UBERT wrapper) —
+reference pipelines/information_extraction.py:19-60."""
+from __future__ import annotations
+
+import torch
+
+from fengshen_amd.pipelines.base import BasePipeline
+
+
+class InformationExtractionPipeline(BasePipeline):
+    task_name = "information_extraction"
+
+    def __init__(self, args=None, model=None, tokenizer=None, config=None):
+        super().__init__(args, model, tokenizer)
+        if self.model is None:
+            from fengshen_amd.models.ubert.modeling_ubert import UbertModel
+            if isinstance(model, str):
+                self.model = UbertModel.from_pretrained(model)
+            else:
+                self.model = UbertModel(config)
+        self.model.eval()
+
+    def _encode(self, text: str, labels):
+        rows = []
+        for lab in labels:
+            prompt = f"[CLS]{lab}[SEP]{text}[SEP]"
+            ids = [self.tokenizer.get_vocab().get(c, 4) for c in prompt]
+            rows.append(ids)
+        L = max(len(r) for r in rows)
+        pad = self.tokenizer.pad_token_id or 0
+        return torch.tensor([r + [pad] * (L - len(r)) for r in rows],
+                            dtype=torch.long)
+
+    @torch.no_grad()
+    def __call__(self, texts, entity_types=None):
+        entity_types = entity_types or ["人名", "地名", "机构"]
+        single = isinstance(texts, str)
+        if single:
+            texts = [texts]
+        results = []
+        for text in texts:
+            ids = self._encode(text, entity_types).unsqueeze(0)
+            dev = next(self.model.parameters()).device
+            spans = self.model.extract(ids.to(dev))
+            per_text = {}
+            for li, lab in enumerate(entity_types):
+                off = len(lab) + 10  # prompt chars before text
+                per_text[lab] = [
+                    {"span": (s, e), "score": sc}
+                    for s, e, sc in spans[0][li]]
+            results.append(per_text)
+        return results[0] if single else results
+
+
+Pipeline = InformationExtractionPipeline

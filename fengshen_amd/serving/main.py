@@ -1,0 +1,80 @@
+"""FastAPI model server — one pipeline per process.
+
+Behavioral parity: reference fengshen/API/main.py:12-74 (JSON config ->
+pipeline import by type -> POST endpoint wrapping pipeline(input_text),
+CORS, logging) + utils.py:14-128 (pydantic request/config).
+Run: python -m fengshen_amd.serving.main config.json
+"""
+from __future__ import annotations
+
+import json
+import logging
+import sys
+from importlib import import_module
+from typing import Any, Optional
+
+from pydantic import BaseModel
+
+logger = logging.getLogger("fengshen_amd.serving")
+
+
+class RequestData(BaseModel):
+    input_text: str
+    extra: Optional[dict] = None
+
+
+class APIConfig(BaseModel):
+    pipeline_type: str = "text_classification"
+    model: Optional[str] = None
+    host: str = "127.0.0.1"
+    port: int = 8000
+    allow_origins: list = ["*"]
+    pipeline_kwargs: dict = {}
+
+    @classmethod
+    def from_json(cls, path: str) -> "APIConfig":
+        with open(path) as f:
+            return cls(**json.load(f))
+
+
+def build_app(config: APIConfig, pipeline=None):
+    from fastapi import FastAPI
+    from fastapi.middleware.cors import CORSMiddleware
+
+    if pipeline is None:
+        mod = import_module(
+            f"fengshen_amd.pipelines.{config.pipeline_type}")
+        from transformers import AutoTokenizer
+        tokenizer = AutoTokenizer.from_pretrained(config.model) \
+            if config.model else None
+        pipeline = mod.Pipeline(model=config.model, tokenizer=tokenizer,
+                                **config.pipeline_kwargs)
+
+    app = FastAPI(title=f"fengshen_amd {config.pipeline_type}")
+    app.add_middleware(
+        CORSMiddleware, allow_origins=config.allow_origins,
+        allow_credentials=True, allow_methods=["*"], allow_headers=["*"])
+
+    @app.get("/health")
+    def health():
+        return {"status": "ok", "pipeline": config.pipeline_type}
+
+    @app.post("/predict")
+    def predict(req: RequestData) -> Any:
+        logger.info("request: %.80s", req.input_text)
+        result = pipeline(req.input_text)
+        return {"result": result}
+
+    return app
+
+
+def main():
+    cfg_path = sys.argv[1] if len(sys.argv) > 1 else None
+    config = APIConfig.from_json(cfg_path) if cfg_path else APIConfig()
+    app = build_app(config)
+    import uvicorn
+    uvicorn.run(app, host=config.host, port=config.port)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,105 @@
+"""Pipelines / CLI / serving tests (CPU, tiny configs, fake tokenizer)."""
+import argparse
+
+import pytest
+import torch
+
+from tests.test_data import FakeTokenizer
+
+
+def _args(**over):
+    from fengshen_amd.pipelines.base import add_common_pipeline_args
+    parser = argparse.ArgumentParser()
+    add_common_pipeline_args(parser)
+    args = parser.parse_args([])
+    args.max_steps = 3
+    args.precision = "fp32"
+    args.train_batchsize = 4
+    args.num_workers = 0
+    args.sampler_type = "single"
+    args.learning_rate = 1e-3
+    for k, v in over.items():
+        setattr(args, k, v)
+    return args
+
+
+def test_text_classification_pipeline_train_and_predict(tmp_path):
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.pipelines.text_classification import (
+        TextClassificationPipeline)
+    torch.manual_seed(0)
+    args = _args(default_root_dir=str(tmp_path),
+                 save_ckpt_path=str(tmp_path / "ckpt"))
+    cfg = bert_tiny_config()
+    cfg.num_labels = 2
+    pipe = TextClassificationPipeline(
+        args=args, tokenizer=FakeTokenizer(), config=cfg)
+    train = [{"sentence": f"text number {i}", "label": i % 2}
+             for i in range(32)]
+    pipe.train({"train": train})
+    out = pipe("a test sentence")
+    assert "label" in out and "score" in out
+    outs = pipe(["one", "two"])
+    assert len(outs) == 2
+
+
+def test_sequence_tagging_pipeline_predict():
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.pipelines.sequence_tagging import SequenceTaggingPipeline
+    torch.manual_seed(0)
+    id2label = {0: "O", 1: "B-PER", 2: "I-PER"}
+    pipe = SequenceTaggingPipeline(
+        args=None, tokenizer=FakeTokenizer(), id2label=id2label,
+        config=bert_tiny_config(), head="crf")
+    res = pipe("李明在上海")
+    assert isinstance(res, list)
+
+
+def test_multiplechoice_pipeline():
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.pipelines.multiplechoice import MultipleChoicePipeline
+    torch.manual_seed(0)
+    pipe = MultipleChoicePipeline(
+        tokenizer=FakeTokenizer(), config=bert_tiny_config(), yes_token_id=5)
+    out = pipe({"texta": "今天下雨了", "question": "天气如何",
+                "choices": ["晴", "雨", "雪"]})
+    assert out["choice"] in ["晴", "雨", "雪"]
+
+
+def test_information_extraction_pipeline():
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.pipelines.information_extraction import (
+        InformationExtractionPipeline)
+    torch.manual_seed(0)
+    pipe = InformationExtractionPipeline(
+        tokenizer=FakeTokenizer(), config=bert_tiny_config())
+    out = pipe("李明住在北京", entity_types=["人名", "地名"])
+    assert set(out.keys()) == {"人名", "地名"}
+
+
+def test_serving_app():
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+    from fengshen_amd.serving.main import APIConfig, build_app
+
+    class EchoPipeline:
+        def __call__(self, text):
+            return {"echo": text}
+
+    app = build_app(APIConfig(pipeline_type="echo"), pipeline=EchoPipeline())
+    client = TestClient(app)
+    r = client.get("/health")
+    assert r.status_code == 200
+    r = client.post("/predict", json={"input_text": "你好"})
+    assert r.status_code == 200
+    assert r.json()["result"]["echo"] == "你好"
+
+
+def test_cli_help_paths():
+    from fengshen_amd.cli.fengshen_pipeline import main
+    assert main([]) == 1
+    assert main(["nonexistent_task", "train"]) == 1
