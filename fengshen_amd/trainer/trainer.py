@@ -490,8 +490,16 @@ class Trainer:
         }
         self.module.on_save_checkpoint(checkpoint)
         self._call("on_save_checkpoint", checkpoint)
-        # model shard: saved once per TP rank by DP rank 0
-        if pg.get_data_parallel_rank() == 0:
+        # model shard: saved once per TP rank by DP rank 0; ZeRO-3 params
+        # must be materialized for state_dict
+        from fengshen_amd.parallel.zero3 import Zero3Engine
+        if isinstance(self.optimizer, Zero3Engine):
+            with self.optimizer.gathered_params():
+                if pg.get_data_parallel_rank() == 0:
+                    sd = {k: v.detach().clone()
+                          for k, v in self.module.state_dict().items()}
+                    torch.save(sd, paths["model"])
+        elif pg.get_data_parallel_rank() == 0:
             torch.save(self.module.state_dict(), paths["model"])
         if not weights_only and self.optimizer is not None:
             payload = {"optimizer": self.optimizer.state_dict()}

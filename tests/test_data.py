@@ -11,31 +11,7 @@ from fengshen_amd.data.indexed_dataset import (
 )
 
 
-class FakeTokenizer:
-    """Minimal HF-ish tokenizer over characters (for collator tests)."""
-
-    def __init__(self):
-        chars = [chr(c) for c in range(0x4E00, 0x4E00 + 200)] + \
-            list("abcdefghijklmnopqrstuvwxyz0123456789.,!? :\n<>_")
-        self._vocab = {"[PAD]": 0, "[CLS]": 1, "[SEP]": 2, "[MASK]": 3,
-                       "[UNK]": 4, "<s>": 5, "</s>": 6}
-        for c in chars:
-            self._vocab.setdefault(c, len(self._vocab))
-        self.pad_token_id = 0
-        self.cls_token_id = 1
-        self.sep_token_id = 2
-        self.mask_token_id = 3
-        self.bos_token_id = 5
-        self.eos_token_id = 6
-
-    def get_vocab(self):
-        return self._vocab
-
-    def encode(self, text, add_special_tokens=True):
-        ids = [self._vocab.get(c, 4) for c in text]
-        if add_special_tokens:
-            return [self.cls_token_id] + ids + [self.sep_token_id]
-        return ids
+from fengshen_amd.tokenizer import SimpleCharTokenizer as FakeTokenizer  # noqa
 
 
 def test_mmap_dataset_roundtrip(tmp_path):
