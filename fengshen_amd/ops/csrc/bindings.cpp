@@ -49,6 +49,9 @@ void fs_fused_adamw(float*, const void*, float*, float*, void*, long, float,
                     float, float, float, float, int, int, int, hipStream_t);
 void fs_flash_attn_fwd(const void*, const void*, const void*, void*, float*,
                        int, int, int, float, hipStream_t);
+void fs_flash_attn_bwd(const void*, const void*, const void*, const void*,
+                       const void*, const float*, void*, void*, void*, float*,
+                       int, int, int, float, hipStream_t);
 }
 
 // ---------------------------------------------------------------------------
@@ -276,8 +279,26 @@ static std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
   return {o, lse};
 }
 
+static std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
+                                              at::Tensor v, at::Tensor o,
+                                              at::Tensor dout, at::Tensor lse,
+                                              double scale) {
+  const int b = q.size(0), h = q.size(1), s = q.size(2);
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto delta = at::empty({b, h, s}, q.options().dtype(at::kFloat));
+  fs_flash_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                    dout.contiguous().data_ptr(), lse.data_ptr<float>(),
+                    dq.data_ptr(), dk.data_ptr(), dv.data_ptr(),
+                    delta.data_ptr<float>(), b, h, s, (float)scale,
+                    cur_stream());
+  return {dq, dk, dv};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("flash_attn_fwd", &flash_attn_fwd);
+  mod.def("flash_attn_bwd", &flash_attn_bwd);
   mod.def("rms_norm_fwd", &rms_norm_fwd);
   mod.def("rms_norm_bwd", &rms_norm_bwd);
   mod.def("layer_norm_fwd", &layer_norm_fwd);

@@ -245,3 +245,416 @@ extern "C" void fs_flash_attn_fwd(const void* q, const void* k, const void* v,
                      (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
                      (bf16_t*)o, lse, b, h, s, scale);
 }
+
+// ===========================================================================
+// BACKWARD
+// ===========================================================================
+// FlashAttention-2-style: delta = rowsum(dO*O); then
+//   dQ kernel (q-major):   S=(sQ)K^T; P=exp(S-LSE); dP=dO V^T;
+//                          dS=P*(dP-delta)*s; dQ += dS K
+//   dKdV kernel (kv-major): St=(sK)Q^T; Pt=exp(St-LSE[col]); dPt=V dO^T;
+//                          dV += Pt dO; dSt=Pt*(dPt-delta[col])*s; dK += dSt Q
+// Same fragment layouts / swizzles as forward (HW-verified via mfma_probe).
+
+__global__ __launch_bounds__(256)
+void flash_delta_kernel(const bf16_t* __restrict__ dO,
+                        const bf16_t* __restrict__ O,
+                        float* __restrict__ delta, long rows) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const long row0 = (long)blockIdx.x * 4 + wid;
+  const long stride = (long)gridDim.x * 4;
+  for (long row = row0; row < rows; row += stride) {
+    const bf16_t* dop = dO + row * FA_D + lane * 2;
+    const bf16_t* op = O + row * FA_D + lane * 2;
+    float s = 0.f;
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      s += __bfloat162float(dop[j]) * __bfloat162float(op[j]);
+    }
+    s = wave_reduce_sum(s);
+    if (lane == 0) delta[row] = s;
+  }
+}
+
+__global__ __launch_bounds__(FA_WAVES * 64)
+void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
+                              const bf16_t* __restrict__ K,
+                              const bf16_t* __restrict__ V,
+                              const bf16_t* __restrict__ dO,
+                              const float* __restrict__ LSE,
+                              const float* __restrict__ Delta,
+                              bf16_t* __restrict__ dQ,
+                              int b, int h, int s, float scale) {
+  __shared__ short k_lds[FA_KVBLK][FA_D];                 // K rows, swizzled
+  __shared__ short v_lds[FA_KVBLK][FA_D];                 // V rows, swizzled
+  __shared__ short kt_lds[FA_D][FA_KVBLK + FA_VPAD];      // K^T, kv-swizzled
+  __shared__ short p_lds[FA_WAVES][16][FA_KVBLK + FA_VPAD];
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const long bh = (long)blockIdx.z * h + blockIdx.y;
+  const bf16_t* Qp = Q + bh * s * FA_D;
+  const bf16_t* Kp = K + bh * s * FA_D;
+  const bf16_t* Vp = V + bh * s * FA_D;
+  const bf16_t* dOp = dO + bh * s * FA_D;
+  bf16_t* dQp = dQ + bh * s * FA_D;
+  const float* lse = LSE + bh * s;
+  const float* dlt = Delta + bh * s;
+
+  const int q0 = blockIdx.x * FA_QBLK + wave * 16;
+
+  // Q (pre-scaled) and dO A-fragments in registers
+  bf16x8 q_frag[4], do_frag[4];
+  {
+    const int row = lane & 15;
+    const int k0 = (lane >> 4) * 8;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 raw = *reinterpret_cast<const bf16x8*>(
+          Qp + (long)(q0 + row) * FA_D + c * 32 + k0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        unsigned short u = (unsigned short)raw[j];
+        float f = __uint_as_float(((unsigned int)u) << 16) * scale;
+        raw[j] = (short)__hip_bfloat16_raw(__float2bfloat16(f)).x;
+      }
+      q_frag[c] = raw;
+      do_frag[c] = *reinterpret_cast<const bf16x8*>(
+          dOp + (long)(q0 + row) * FA_D + c * 32 + k0);
+    }
+  }
+  // per-row LSE/delta (D-layout rows)
+  float lse_r[4], dlt_r[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + (lane >> 4) * 4 + r;
+    lse_r[r] = lse[qrow];
+    dlt_r[r] = dlt[qrow];
+  }
+
+  f32x4 dq_acc[8];
+#pragma unroll
+  for (int t = 0; t < 8; ++t) dq_acc[t] = f32x4{0, 0, 0, 0};
+
+  const int q_hi = blockIdx.x * FA_QBLK + FA_QBLK - 1;
+  const int n_kv_tiles = (q_hi / FA_KVBLK) + 1;
+
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int k_base = kt * FA_KVBLK;
+    __syncthreads();
+    {
+      const int tid = threadIdx.x;
+      for (int i = tid * 8; i < FA_KVBLK * FA_D; i += 256 * 8) {
+        const int kr = i / FA_D;
+        const int kc = i % FA_D;
+        bf16x8 kk = *reinterpret_cast<const bf16x8*>(
+            Kp + (long)(k_base + kr) * FA_D + kc);
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(&k_lds[kr][0]) + kswz(kr, kc * 2)) = kk;
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
+            Vp + (long)(k_base + kr) * FA_D + kc);
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(&v_lds[kr][0]) + kswz(kr, kc * 2)) = vv;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          kt_lds[kc + j][kr ^ ((kc + j) & 0x38)] = kk[j];
+      }
+    }
+    __syncthreads();
+
+    // S and dP tiles (both [q, kv] D-layout)
+    f32x4 s_acc[4], dp_acc[4];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      s_acc[nt] = f32x4{0, 0, 0, 0};
+      dp_acc[nt] = f32x4{0, 0, 0, 0};
+    }
+    {
+      const int col = lane & 15;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const int krow = nt * 16 + col;
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          const int d0 = c * 32 + (lane >> 4) * 8;
+          bf16x8 kb = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<char*>(&k_lds[krow][0]) + kswz(krow, d0 * 2));
+          s_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              q_frag[c], kb, s_acc[nt], 0, 0, 0);
+          bf16x8 vb = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<char*>(&v_lds[krow][0]) + kswz(krow, d0 * 2));
+          dp_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              do_frag[c], vb, dp_acc[nt], 0, 0, 0);
+        }
+      }
+    }
+
+    // dS = P * (dP - delta) * scale   (P = exp(S - LSE))
+    const int col = lane & 15;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = q0 + (lane >> 4) * 4 + r;
+      const int row = (lane >> 4) * 4 + r;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const int kcol = k_base + nt * 16 + col;
+        float pv = (kcol > qrow) ? 0.f : __expf(s_acc[nt][r] - lse_r[r]);
+        float ds = pv * (dp_acc[nt][r] - dlt_r[r]) * scale;
+        p_lds[wave][row][nt * 16 + col] =
+            (short)__hip_bfloat16_raw(__float2bfloat16(ds)).x;
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+
+    // dQ += dS @ K  (B from kt_lds)
+#pragma unroll
+    for (int kc = 0; kc < 2; ++kc) {
+      const int row = lane & 15;
+      const int kv0 = kc * 32 + (lane >> 4) * 8;
+      bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wave][row][kv0]);
+#pragma unroll
+      for (int t = 0; t < 8; ++t) {
+        const int dcol = t * 16 + (lane & 15);
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+            &kt_lds[dcol][kv0 ^ (dcol & 0x38)]);
+        dq_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pa, bfrag, dq_acc[t], 0, 0, 0);
+      }
+    }
+  }
+
+  const int col = lane & 15;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + (lane >> 4) * 4 + r;
+#pragma unroll
+    for (int t = 0; t < 8; ++t) {
+      dQp[(long)qrow * FA_D + t * 16 + col] =
+          __float2bfloat16(dq_acc[t][r]);
+    }
+  }
+}
+
+__global__ __launch_bounds__(FA_WAVES * 64)
+void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
+                               const bf16_t* __restrict__ K,
+                               const bf16_t* __restrict__ V,
+                               const bf16_t* __restrict__ dO,
+                               const float* __restrict__ LSE,
+                               const float* __restrict__ Delta,
+                               bf16_t* __restrict__ dK,
+                               bf16_t* __restrict__ dV,
+                               int b, int h, int s, float scale) {
+  __shared__ short q_lds[FA_QBLK][FA_D];                  // Q rows, swizzled
+  __shared__ short do_lds[FA_QBLK][FA_D];                 // dO rows, swizzled
+  __shared__ short qt_lds[FA_D][FA_QBLK + FA_VPAD];       // Q^T
+  __shared__ short dot_lds[FA_D][FA_QBLK + FA_VPAD];      // dO^T
+  __shared__ short p_lds[FA_WAVES][16][FA_QBLK + FA_VPAD];
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const long bh = (long)blockIdx.z * h + blockIdx.y;
+  const bf16_t* Qp = Q + bh * s * FA_D;
+  const bf16_t* Kp = K + bh * s * FA_D;
+  const bf16_t* Vp = V + bh * s * FA_D;
+  const bf16_t* dOp = dO + bh * s * FA_D;
+  bf16_t* dKp = dK + bh * s * FA_D;
+  bf16_t* dVp = dV + bh * s * FA_D;
+  const float* lse = LSE + bh * s;
+  const float* dlt = Delta + bh * s;
+
+  const int kv0_blk = blockIdx.x * FA_KVBLK;
+  const int kv0_wave = kv0_blk + wave * 16;  // this wave's 16 kv rows
+
+  // K (pre-scaled) and V A-fragments in registers
+  bf16x8 k_frag[4], v_frag[4];
+  {
+    const int row = lane & 15;
+    const int c0 = (lane >> 4) * 8;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 raw = *reinterpret_cast<const bf16x8*>(
+          Kp + (long)(kv0_wave + row) * FA_D + c * 32 + c0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        unsigned short u = (unsigned short)raw[j];
+        float f = __uint_as_float(((unsigned int)u) << 16) * scale;
+        raw[j] = (short)__hip_bfloat16_raw(__float2bfloat16(f)).x;
+      }
+      k_frag[c] = raw;
+      v_frag[c] = *reinterpret_cast<const bf16x8*>(
+          Vp + (long)(kv0_wave + row) * FA_D + c * 32 + c0);
+    }
+  }
+
+  f32x4 dv_acc[8], dk_acc[8];
+#pragma unroll
+  for (int t = 0; t < 8; ++t) {
+    dv_acc[t] = f32x4{0, 0, 0, 0};
+    dk_acc[t] = f32x4{0, 0, 0, 0};
+  }
+
+  const int first_qt = kv0_blk / FA_QBLK;  // causal: q tiles >= diag
+  const int n_q_tiles = s / FA_QBLK;
+
+  for (int qt = first_qt; qt < n_q_tiles; ++qt) {
+    const int q_base = qt * FA_QBLK;
+    __syncthreads();
+    {
+      const int tid = threadIdx.x;
+      for (int i = tid * 8; i < FA_QBLK * FA_D; i += 256 * 8) {
+        const int qr = i / FA_D;
+        const int qc = i % FA_D;
+        bf16x8 qq = *reinterpret_cast<const bf16x8*>(
+            Qp + (long)(q_base + qr) * FA_D + qc);
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(&q_lds[qr][0]) + kswz(qr, qc * 2)) = qq;
+        bf16x8 dd = *reinterpret_cast<const bf16x8*>(
+            dOp + (long)(q_base + qr) * FA_D + qc);
+        *reinterpret_cast<bf16x8*>(
+            reinterpret_cast<char*>(&do_lds[qr][0]) + kswz(qr, qc * 2)) = dd;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          qt_lds[qc + j][qr ^ ((qc + j) & 0x38)] = qq[j];
+          dot_lds[qc + j][qr ^ ((qc + j) & 0x38)] = dd[j];
+        }
+      }
+    }
+    __syncthreads();
+
+    // St = (sK) @ Q^T ; dPt = V @ dO^T  (both [kv, q] D-layout)
+    f32x4 st_acc[4], dpt_acc[4];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      st_acc[nt] = f32x4{0, 0, 0, 0};
+      dpt_acc[nt] = f32x4{0, 0, 0, 0};
+    }
+    {
+      const int col = lane & 15;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const int qrow = nt * 16 + col;
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          const int d0 = c * 32 + (lane >> 4) * 8;
+          bf16x8 qb = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<char*>(&q_lds[qrow][0]) + kswz(qrow, d0 * 2));
+          st_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              k_frag[c], qb, st_acc[nt], 0, 0, 0);
+          bf16x8 db = *reinterpret_cast<const bf16x8*>(
+              reinterpret_cast<char*>(&do_lds[qrow][0]) + kswz(qrow, d0 * 2));
+          dpt_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              v_frag[c], db, dpt_acc[nt], 0, 0, 0);
+        }
+      }
+    }
+
+    // Pt and dSt
+    const int col = lane & 15;
+    float pt[4][4];
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt) {
+      const int qcol = q_base + nt * 16 + col;
+      const float lse_c = lse[qcol];
+      const float dlt_c = dlt[qcol];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kvrow = kv0_wave + (lane >> 4) * 4 + r;
+        float pv = (qcol < kvrow) ? 0.f : __expf(st_acc[nt][r] - lse_c);
+        pt[nt][r] = pv;
+        // stash dSt in dpt_acc (reuse)
+        dpt_acc[nt][r] = pv * (dpt_acc[nt][r] - dlt_c) * scale;
+      }
+    }
+
+    // Pt -> p_lds, dV += Pt @ dO
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = (lane >> 4) * 4 + r;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt)
+        p_lds[wave][row][nt * 16 + col] =
+            (short)__hip_bfloat16_raw(__float2bfloat16(pt[nt][r])).x;
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+    for (int qc = 0; qc < 2; ++qc) {
+      const int row = lane & 15;
+      const int q0f = qc * 32 + (lane >> 4) * 8;
+      bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wave][row][q0f]);
+#pragma unroll
+      for (int t = 0; t < 8; ++t) {
+        const int dcol = t * 16 + (lane & 15);
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+            &dot_lds[dcol][q0f ^ (dcol & 0x38)]);
+        dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pa, bfrag, dv_acc[t], 0, 0, 0);
+      }
+    }
+
+    // dSt -> p_lds, dK += dSt @ Q
+    __builtin_amdgcn_s_waitcnt(0);  // all waves done reading Pt
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = (lane >> 4) * 4 + r;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt)
+        p_lds[wave][row][nt * 16 + col] =
+            (short)__hip_bfloat16_raw(__float2bfloat16(dpt_acc[nt][r])).x;
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+    for (int qc = 0; qc < 2; ++qc) {
+      const int row = lane & 15;
+      const int q0f = qc * 32 + (lane >> 4) * 8;
+      bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wave][row][q0f]);
+#pragma unroll
+      for (int t = 0; t < 8; ++t) {
+        const int dcol = t * 16 + (lane & 15);
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+            &qt_lds[dcol][q0f ^ (dcol & 0x38)]);
+        dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pa, bfrag, dk_acc[t], 0, 0, 0);
+      }
+    }
+  }
+
+  const int col = lane & 15;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int kvrow = kv0_wave + (lane >> 4) * 4 + r;
+#pragma unroll
+    for (int t = 0; t < 8; ++t) {
+      dKp[(long)kvrow * FA_D + t * 16 + col] =
+          __float2bfloat16(dk_acc[t][r]);
+      dVp[(long)kvrow * FA_D + t * 16 + col] =
+          __float2bfloat16(dv_acc[t][r]);
+    }
+  }
+}
+
+extern "C" void fs_flash_attn_bwd(const void* q, const void* k, const void* v,
+                                  const void* o, const void* dout,
+                                  const float* lse, void* dq, void* dk,
+                                  void* dv, float* delta_ws, int b, int h,
+                                  int s, float scale, hipStream_t stream) {
+  const long rows = (long)b * h * s;
+  {
+    long blocks = (rows + 3) / 4;
+    if (blocks > 4096) blocks = 4096;
+    hipLaunchKernelGGL(flash_delta_kernel, dim3((unsigned)blocks), dim3(256),
+                       0, stream, (const bf16_t*)dout, (const bf16_t*)o,
+                       delta_ws, rows);
+  }
+  dim3 grid(s / FA_QBLK, h, b);
+  dim3 block(FA_WAVES * 64);
+  hipLaunchKernelGGL(flash_attn_bwd_dq_kernel, grid, block, 0, stream,
+                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
+                     (const bf16_t*)dout, lse, delta_ws, (bf16_t*)dq, b, h, s,
+                     scale);
+  hipLaunchKernelGGL(flash_attn_bwd_dkv_kernel, grid, block, 0, stream,
+                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
+                     (const bf16_t*)dout, lse, delta_ws, (bf16_t*)dk,
+                     (bf16_t*)dv, b, h, s, scale);
+}

@@ -19,14 +19,18 @@ class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale):
         o, lse = get_ext().flash_attn_fwd(q, k, v, scale)
-        ctx.save_for_backward(q, k, v, lse)
+        ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale
         return o
 
     @staticmethod
     def backward(ctx, do):
-        q, k, v, lse = ctx.saved_tensors
+        q, k, v, o, lse = ctx.saved_tensors
         scale = ctx.scale
+        if hasattr(get_ext(), "flash_attn_bwd"):
+            dq, dk, dv = get_ext().flash_attn_bwd(q, k, v, o, do.contiguous(),
+                                                  lse, scale)
+            return dq, dk, dv, None
         b, h, s, d = q.shape
         qf = q.reshape(b * h, s, d)
         kf = k.reshape(b * h, s, d)

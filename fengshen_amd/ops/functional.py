@@ -385,13 +385,7 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     if ext is not None and hasattr(ext, "flash_attn_fwd"):
         from fengshen_amd.ops.flash import flash_attention, flash_attn_supported
         if flash_attn_supported(q, k, v, causal, mask, dropout_p):
-            # Under activation checkpointing the no-grad first pass takes the
-            # fused flash kernel (no SxS materialization); the grad-enabled
-            # recompute takes the prob-saving bmm path below so backward
-            # reuses P instead of rebuilding it.  Once the fused flash
-            # BACKWARD kernel lands, flash handles both.
-            if not (torch.is_grad_enabled() and q.requires_grad):
-                return flash_attention(q, k, v, scale)
+            return flash_attention(q, k, v, scale)
     # bmm path: scores in [b*np, sq, sk]
     q2 = q.reshape(b * np_, sq, hn)
     k2 = k.reshape(b * np_, sk, hn)

@@ -26,3 +26,17 @@ torch.cuda.synchronize()
 dt = (time.perf_counter()-t0)/N
 flops = 2*2*b*h*s*s*d*0.5
 print(f"flash fwd: {dt*1000:.2f} ms, {flops/dt/1e12:.1f} TF")
+
+# bwd timing
+q.requires_grad_(True); k.requires_grad_(True); v.requires_grad_(True)
+out = flash_attention(q, k, v, scale)
+gy = torch.randn_like(out)
+torch.cuda.synchronize()
+t0 = time.perf_counter(); N2 = 5
+for _ in range(N2):
+    out = flash_attention(q, k, v, scale)
+    torch.autograd.backward(out, gy)
+torch.cuda.synchronize()
+dt = (time.perf_counter()-t0)/N2
+fb_flops = flops * 3.5  # fwd + bwd(2.5x)
+print(f"flash fwd+bwd: {dt*1000:.2f} ms, {fb_flops/dt/1e12:.1f} TF-equiv")
