@@ -254,6 +254,10 @@ extern "C" void fs_flash_attn_fwd(const void* q, const void* k, const void* v,
 //   dKdV kernel (kv-major): St=(sK)Q^T; Pt=exp(St-LSE[col]); dPt=V dO^T;
 //                          dV += Pt dO; dSt=Pt*(dPt-delta[col])*s; dK += dSt Q
 // Same fragment layouts / swizzles as forward (HW-verified via mfma_probe).
+// Streaming tiles are 32-wide (FB_T) so LDS stays ~40KB -> 3-4 blocks/CU
+// (the 64-wide first cut sat at 1 block/CU and was latency-bound).
+
+#define FB_T 32
 
 __global__ __launch_bounds__(256)
 void flash_delta_kernel(const bf16_t* __restrict__ dO,
@@ -266,16 +270,16 @@ void flash_delta_kernel(const bf16_t* __restrict__ dO,
   for (long row = row0; row < rows; row += stride) {
     const bf16_t* dop = dO + row * FA_D + lane * 2;
     const bf16_t* op = O + row * FA_D + lane * 2;
-    float s = 0.f;
+    float sacc = 0.f;
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      s += __bfloat162float(dop[j]) * __bfloat162float(op[j]);
-    }
-    s = wave_reduce_sum(s);
-    if (lane == 0) delta[row] = s;
+    for (int j = 0; j < 2; ++j)
+      sacc += __bfloat162float(dop[j]) * __bfloat162float(op[j]);
+    sacc = wave_reduce_sum(sacc);
+    if (lane == 0) delta[row] = sacc;
   }
 }
 
+// Q-major: each block owns 64 q rows (wave -> 16), streams KV in 32-tiles.
 __global__ __launch_bounds__(FA_WAVES * 64)
 void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
                               const bf16_t* __restrict__ K,
@@ -285,10 +289,10 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
                               const float* __restrict__ Delta,
                               bf16_t* __restrict__ dQ,
                               int b, int h, int s, float scale) {
-  __shared__ short k_lds[FA_KVBLK][FA_D];                 // K rows, swizzled
-  __shared__ short v_lds[FA_KVBLK][FA_D];                 // V rows, swizzled
-  __shared__ short kt_lds[FA_D][FA_KVBLK + FA_VPAD];      // K^T, kv-swizzled
-  __shared__ short p_lds[FA_WAVES][16][FA_KVBLK + FA_VPAD];
+  __shared__ short k_lds[FB_T][FA_D];                 // K rows, swizzled
+  __shared__ short v_lds[FB_T][FA_D];                 // V rows, swizzled
+  __shared__ short kt_lds[FA_D][FB_T + FA_VPAD];      // K^T, kv-swizzled
+  __shared__ short p_lds[FA_WAVES][16][FB_T + FA_VPAD];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -303,7 +307,6 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
 
   const int q0 = blockIdx.x * FA_QBLK + wave * 16;
 
-  // Q (pre-scaled) and dO A-fragments in registers
   bf16x8 q_frag[4], do_frag[4];
   {
     const int row = lane & 15;
@@ -323,7 +326,6 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
           dOp + (long)(q0 + row) * FA_D + c * 32 + k0);
     }
   }
-  // per-row LSE/delta (D-layout rows)
   float lse_r[4], dlt_r[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -337,14 +339,14 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
   for (int t = 0; t < 8; ++t) dq_acc[t] = f32x4{0, 0, 0, 0};
 
   const int q_hi = blockIdx.x * FA_QBLK + FA_QBLK - 1;
-  const int n_kv_tiles = (q_hi / FA_KVBLK) + 1;
+  const int n_kv_tiles = (q_hi / FB_T) + 1;
 
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
-    const int k_base = kt * FA_KVBLK;
+    const int k_base = kt * FB_T;
     __syncthreads();
     {
       const int tid = threadIdx.x;
-      for (int i = tid * 8; i < FA_KVBLK * FA_D; i += 256 * 8) {
+      for (int i = tid * 8; i < FB_T * FA_D; i += 256 * 8) {
         const int kr = i / FA_D;
         const int kc = i % FA_D;
         bf16x8 kk = *reinterpret_cast<const bf16x8*>(
@@ -357,22 +359,21 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
             reinterpret_cast<char*>(&v_lds[kr][0]) + kswz(kr, kc * 2)) = vv;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          kt_lds[kc + j][kr ^ ((kc + j) & 0x38)] = kk[j];
+          kt_lds[kc + j][kr ^ ((kc + j) & 0x18)] = kk[j];
       }
     }
     __syncthreads();
 
-    // S and dP tiles (both [q, kv] D-layout)
-    f32x4 s_acc[4], dp_acc[4];
+    f32x4 s_acc[2], dp_acc[2];
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) {
+    for (int nt = 0; nt < 2; ++nt) {
       s_acc[nt] = f32x4{0, 0, 0, 0};
       dp_acc[nt] = f32x4{0, 0, 0, 0};
     }
     {
       const int col = lane & 15;
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < 2; ++nt) {
         const int krow = nt * 16 + col;
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
@@ -389,14 +390,13 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
       }
     }
 
-    // dS = P * (dP - delta) * scale   (P = exp(S - LSE))
     const int col = lane & 15;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow = q0 + (lane >> 4) * 4 + r;
       const int row = (lane >> 4) * 4 + r;
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < 2; ++nt) {
         const int kcol = k_base + nt * 16 + col;
         float pv = (kcol > qrow) ? 0.f : __expf(s_acc[nt][r] - lse_r[r]);
         float ds = pv * (dp_acc[nt][r] - dlt_r[r]) * scale;
@@ -406,17 +406,15 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
     }
     __builtin_amdgcn_s_waitcnt(0);
 
-    // dQ += dS @ K  (B from kt_lds)
-#pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
+    {
       const int row = lane & 15;
-      const int kv0 = kc * 32 + (lane >> 4) * 8;
+      const int kv0 = (lane >> 4) * 8;
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wave][row][kv0]);
 #pragma unroll
       for (int t = 0; t < 8; ++t) {
         const int dcol = t * 16 + (lane & 15);
         bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-            &kt_lds[dcol][kv0 ^ (dcol & 0x38)]);
+            &kt_lds[dcol][kv0 ^ (dcol & 0x18)]);
         dq_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             pa, bfrag, dq_acc[t], 0, 0, 0);
       }
@@ -428,13 +426,12 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
   for (int r = 0; r < 4; ++r) {
     const int qrow = q0 + (lane >> 4) * 4 + r;
 #pragma unroll
-    for (int t = 0; t < 8; ++t) {
-      dQp[(long)qrow * FA_D + t * 16 + col] =
-          __float2bfloat16(dq_acc[t][r]);
-    }
+    for (int t = 0; t < 8; ++t)
+      dQp[(long)qrow * FA_D + t * 16 + col] = __float2bfloat16(dq_acc[t][r]);
   }
 }
 
+// KV-major: each block owns 64 kv rows (wave -> 16), streams Q in 32-tiles.
 __global__ __launch_bounds__(FA_WAVES * 64)
 void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
                                const bf16_t* __restrict__ K,
@@ -445,11 +442,11 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
                                bf16_t* __restrict__ dK,
                                bf16_t* __restrict__ dV,
                                int b, int h, int s, float scale) {
-  __shared__ short q_lds[FA_QBLK][FA_D];                  // Q rows, swizzled
-  __shared__ short do_lds[FA_QBLK][FA_D];                 // dO rows, swizzled
-  __shared__ short qt_lds[FA_D][FA_QBLK + FA_VPAD];       // Q^T
-  __shared__ short dot_lds[FA_D][FA_QBLK + FA_VPAD];      // dO^T
-  __shared__ short p_lds[FA_WAVES][16][FA_QBLK + FA_VPAD];
+  __shared__ short q_lds[FB_T][FA_D];
+  __shared__ short do_lds[FB_T][FA_D];
+  __shared__ short qt_lds[FA_D][FB_T + FA_VPAD];
+  __shared__ short dot_lds[FA_D][FB_T + FA_VPAD];
+  __shared__ short p_lds[FA_WAVES][16][FB_T + FA_VPAD];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -464,9 +461,8 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
   const float* dlt = Delta + bh * s;
 
   const int kv0_blk = blockIdx.x * FA_KVBLK;
-  const int kv0_wave = kv0_blk + wave * 16;  // this wave's 16 kv rows
+  const int kv0_wave = kv0_blk + wave * 16;
 
-  // K (pre-scaled) and V A-fragments in registers
   bf16x8 k_frag[4], v_frag[4];
   {
     const int row = lane & 15;
@@ -494,15 +490,15 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
     dk_acc[t] = f32x4{0, 0, 0, 0};
   }
 
-  const int first_qt = kv0_blk / FA_QBLK;  // causal: q tiles >= diag
-  const int n_q_tiles = s / FA_QBLK;
+  const int first_qt = kv0_blk / FB_T;
+  const int n_q_tiles = s / FB_T;
 
   for (int qt = first_qt; qt < n_q_tiles; ++qt) {
-    const int q_base = qt * FA_QBLK;
+    const int q_base = qt * FB_T;
     __syncthreads();
     {
       const int tid = threadIdx.x;
-      for (int i = tid * 8; i < FA_QBLK * FA_D; i += 256 * 8) {
+      for (int i = tid * 8; i < FB_T * FA_D; i += 256 * 8) {
         const int qr = i / FA_D;
         const int qc = i % FA_D;
         bf16x8 qq = *reinterpret_cast<const bf16x8*>(
@@ -515,24 +511,23 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
             reinterpret_cast<char*>(&do_lds[qr][0]) + kswz(qr, qc * 2)) = dd;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          qt_lds[qc + j][qr ^ ((qc + j) & 0x38)] = qq[j];
-          dot_lds[qc + j][qr ^ ((qc + j) & 0x38)] = dd[j];
+          qt_lds[qc + j][qr ^ ((qc + j) & 0x18)] = qq[j];
+          dot_lds[qc + j][qr ^ ((qc + j) & 0x18)] = dd[j];
         }
       }
     }
     __syncthreads();
 
-    // St = (sK) @ Q^T ; dPt = V @ dO^T  (both [kv, q] D-layout)
-    f32x4 st_acc[4], dpt_acc[4];
+    f32x4 st_acc[2], dpt_acc[2];
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) {
+    for (int nt = 0; nt < 2; ++nt) {
       st_acc[nt] = f32x4{0, 0, 0, 0};
       dpt_acc[nt] = f32x4{0, 0, 0, 0};
     }
     {
       const int col = lane & 15;
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt) {
+      for (int nt = 0; nt < 2; ++nt) {
         const int qrow = nt * 16 + col;
 #pragma unroll
         for (int c = 0; c < 4; ++c) {
@@ -549,11 +544,10 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
       }
     }
 
-    // Pt and dSt
     const int col = lane & 15;
-    float pt[4][4];
+    float pt[2][4];
 #pragma unroll
-    for (int nt = 0; nt < 4; ++nt) {
+    for (int nt = 0; nt < 2; ++nt) {
       const int qcol = q_base + nt * 16 + col;
       const float lse_c = lse[qcol];
       const float dlt_c = dlt[qcol];
@@ -562,57 +556,52 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
         const int kvrow = kv0_wave + (lane >> 4) * 4 + r;
         float pv = (qcol < kvrow) ? 0.f : __expf(st_acc[nt][r] - lse_c);
         pt[nt][r] = pv;
-        // stash dSt in dpt_acc (reuse)
-        dpt_acc[nt][r] = pv * (dpt_acc[nt][r] - dlt_c) * scale;
+        dpt_acc[nt][r] = pv * (dpt_acc[nt][r] - dlt_c) * scale;  // = dSt
       }
     }
 
-    // Pt -> p_lds, dV += Pt @ dO
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = (lane >> 4) * 4 + r;
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt)
+      for (int nt = 0; nt < 2; ++nt)
         p_lds[wave][row][nt * 16 + col] =
             (short)__hip_bfloat16_raw(__float2bfloat16(pt[nt][r])).x;
     }
     __builtin_amdgcn_s_waitcnt(0);
-#pragma unroll
-    for (int qc = 0; qc < 2; ++qc) {
+    {
       const int row = lane & 15;
-      const int q0f = qc * 32 + (lane >> 4) * 8;
+      const int q0f = (lane >> 4) * 8;
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wave][row][q0f]);
 #pragma unroll
       for (int t = 0; t < 8; ++t) {
         const int dcol = t * 16 + (lane & 15);
         bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-            &dot_lds[dcol][q0f ^ (dcol & 0x38)]);
+            &dot_lds[dcol][q0f ^ (dcol & 0x18)]);
         dv_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             pa, bfrag, dv_acc[t], 0, 0, 0);
       }
     }
 
-    // dSt -> p_lds, dK += dSt @ Q
-    __builtin_amdgcn_s_waitcnt(0);  // all waves done reading Pt
+    __builtin_amdgcn_s_waitcnt(0);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int row = (lane >> 4) * 4 + r;
 #pragma unroll
-      for (int nt = 0; nt < 4; ++nt)
+      for (int nt = 0; nt < 2; ++nt)
         p_lds[wave][row][nt * 16 + col] =
             (short)__hip_bfloat16_raw(__float2bfloat16(dpt_acc[nt][r])).x;
     }
     __builtin_amdgcn_s_waitcnt(0);
-#pragma unroll
-    for (int qc = 0; qc < 2; ++qc) {
+    {
       const int row = lane & 15;
-      const int q0f = qc * 32 + (lane >> 4) * 8;
+      const int q0f = (lane >> 4) * 8;
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wave][row][q0f]);
 #pragma unroll
       for (int t = 0; t < 8; ++t) {
         const int dcol = t * 16 + (lane & 15);
         bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-            &qt_lds[dcol][q0f ^ (dcol & 0x38)]);
+            &qt_lds[dcol][q0f ^ (dcol & 0x18)]);
         dk_acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             pa, bfrag, dk_acc[t], 0, 0, 0);
       }
@@ -625,10 +614,8 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
     const int kvrow = kv0_wave + (lane >> 4) * 4 + r;
 #pragma unroll
     for (int t = 0; t < 8; ++t) {
-      dKp[(long)kvrow * FA_D + t * 16 + col] =
-          __float2bfloat16(dk_acc[t][r]);
-      dVp[(long)kvrow * FA_D + t * 16 + col] =
-          __float2bfloat16(dv_acc[t][r]);
+      dKp[(long)kvrow * FA_D + t * 16 + col] = __float2bfloat16(dk_acc[t][r]);
+      dVp[(long)kvrow * FA_D + t * 16 + col] = __float2bfloat16(dv_acc[t][r]);
     }
   }
 }
