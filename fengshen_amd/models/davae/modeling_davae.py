@@ -1,0 +1,132 @@
+"""Randeng-DAVAE: BERT latent encoder + GPT2-for-latent decoder.
+
+Behavioral parity: reference models/DAVAE/DAVAEModel.py:35 +
+GPT2ModelForLatent.py:581 — posterior q(z|x) from a BERT encoder's CLS,
+reparameterized z injected into a GPT2 decoder (prefix conditioning),
+ELBO = reconstruction CE + beta * KL.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.nn as nn
+from transformers import PretrainedConfig, PreTrainedModel
+from transformers.utils import ModelOutput
+
+from fengshen_amd.models.gpt2.configuration_gpt2 import GPT2Config
+from fengshen_amd.models.gpt2.modeling_gpt2 import GPT2Model
+from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+    MegatronBertConfig,
+)
+from fengshen_amd.models.megatron_bert.modeling_megatron_bert import (
+    MegatronBertModel,
+)
+from fengshen_amd.models.layers import parallel_lm_logits
+from fengshen_amd.parallel.cross_entropy import vocab_parallel_cross_entropy
+
+
+class DAVAEConfig(PretrainedConfig):
+    model_type = "fengshen_davae"
+
+    def __init__(self, latent_dim: int = 32, beta_kl: float = 1.0,
+                 encoder_config: Optional[dict] = None,
+                 decoder_config: Optional[dict] = None,
+                 torch_dtype="bfloat16", **kw):
+        self.latent_dim = latent_dim
+        self.beta_kl = beta_kl
+        self.encoder_config = encoder_config or {}
+        self.decoder_config = decoder_config or {}
+        super().__init__(torch_dtype=torch_dtype, **kw)
+
+
+def davae_tiny_config(**over):
+    enc = dict(vocab_size=256, hidden_size=64, num_hidden_layers=2,
+               num_attention_heads=4, intermediate_size=128,
+               max_position_embeddings=64)
+    dec = dict(vocab_size=256, hidden_size=64, num_hidden_layers=2,
+               num_attention_heads=4, max_position_embeddings=64)
+    cfg = dict(latent_dim=16, encoder_config=enc, decoder_config=dec)
+    cfg.update(over)
+    return DAVAEConfig(**cfg)
+
+
+@dataclass
+class DAVAEOutput(ModelOutput):
+    loss: Optional[torch.Tensor] = None
+    rec_loss: Optional[torch.Tensor] = None
+    kl_loss: Optional[torch.Tensor] = None
+    logits: Optional[torch.Tensor] = None
+
+
+class DAVAEModel(PreTrainedModel):
+    config_class = DAVAEConfig
+
+    def _init_weights(self, module):
+        pass
+
+    def __init__(self, config: DAVAEConfig):
+        super().__init__(config)
+        enc_cfg = MegatronBertConfig(**config.encoder_config)
+        dec_cfg = GPT2Config(**config.decoder_config)
+        self.encoder = MegatronBertModel(enc_cfg, add_pooling_layer=False)
+        self.decoder = GPT2Model(dec_cfg)
+        self.to_mu = nn.Linear(enc_cfg.hidden_size, config.latent_dim)
+        self.to_logvar = nn.Linear(enc_cfg.hidden_size, config.latent_dim)
+        self.latent_to_prefix = nn.Linear(config.latent_dim,
+                                          dec_cfg.hidden_size)
+        self.post_init()
+
+    def encode(self, input_ids, attention_mask=None):
+        h = self.encoder(input_ids, attention_mask).last_hidden_state[:, 0]
+        return self.to_mu(h), self.to_logvar(h)
+
+    def reparameterize(self, mu, logvar):
+        std = torch.exp(0.5 * logvar)
+        return mu + std * torch.randn_like(std)
+
+    def decode(self, z, decoder_input_ids):
+        prefix = self.latent_to_prefix(z).unsqueeze(1)  # [b,1,h]
+        emb = self.decoder.wte(decoder_input_ids)
+        pos = torch.arange(decoder_input_ids.shape[1] + 1,
+                           device=decoder_input_ids.device).unsqueeze(0)
+        h = torch.cat([prefix, emb], dim=1) + self.decoder.wpe(pos)
+        h = self.decoder.drop(h)
+        for layer in self.decoder.h:
+            h = layer(h)
+        h = self.decoder.ln_f(h)
+        return h[:, 1:]  # drop prefix position
+
+    def forward(self, input_ids, attention_mask=None, labels=None, **_kw):
+        mu, logvar = self.encode(input_ids, attention_mask)
+        z = self.reparameterize(mu, logvar) if self.training else mu
+        h = self.decode(z, input_ids)
+        logits = parallel_lm_logits(h, self.decoder.wte.weight,
+                                    parallel_output=True)
+        loss = rec = kl = None
+        if labels is not None:
+            shift_logits = logits[:, :-1].contiguous()
+            shift_labels = labels[:, 1:].contiguous()
+            per_token = vocab_parallel_cross_entropy(
+                shift_logits, shift_labels.clamp(min=0))
+            valid = (shift_labels != -100)
+            rec = (per_token * valid).sum() / valid.sum().clamp(min=1)
+            kl = -0.5 * (1 + logvar - mu.pow(2) - logvar.exp()).sum(-1).mean()
+            loss = rec + self.config.beta_kl * kl
+        return DAVAEOutput(loss=loss, rec_loss=rec, kl_loss=kl, logits=logits)
+
+    @torch.no_grad()
+    def sample(self, n: int, seq_len: int, device=None, bos_id: int = 5):
+        """unconditional generation from the prior."""
+        device = device or next(self.parameters()).device
+        z = torch.randn(n, self.config.latent_dim, device=device,
+                        dtype=next(self.parameters()).dtype)
+        ids = torch.full((n, 1), bos_id, dtype=torch.long, device=device)
+        for _ in range(seq_len - 1):
+            h = self.decode(z, ids)
+            logits = parallel_lm_logits(h[:, -1:], self.decoder.wte.weight,
+                                        parallel_output=False)
+            ids = torch.cat([ids, logits[:, -1].argmax(-1, keepdim=True)],
+                            dim=1)
+        return ids

@@ -1,0 +1,239 @@
+"""Erlangshen-DeBERTa-v2: disentangled attention.
+
+Behavioral parity: reference models/deberta_v2/modeling_deberta_v2.py
+(DisentangledSelfAttention :590 — content<->position cross terms with
+bucketed relative positions).
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+from transformers import PretrainedConfig, PreTrainedModel
+from transformers.modeling_outputs import (
+    BaseModelOutput,
+    MaskedLMOutput,
+    SequenceClassifierOutput,
+)
+
+from fengshen_amd.models.layers import LayerNorm, ParallelMLP, init_normal, scaled_init_normal
+from fengshen_amd.ops import functional as F_ops
+
+
+class DebertaV2Config(PretrainedConfig):
+    model_type = "fengshen_deberta_v2"
+
+    def __init__(self, vocab_size: int = 21128, hidden_size: int = 768,
+                 num_hidden_layers: int = 12, num_attention_heads: int = 12,
+                 intermediate_size: int = 3072,
+                 max_position_embeddings: int = 512,
+                 position_buckets: int = 256, max_relative_positions: int = 512,
+                 type_vocab_size: int = 2, layer_norm_eps: float = 1e-7,
+                 initializer_range: float = 0.02, hidden_dropout: float = 0.1,
+                 attention_dropout: float = 0.1, pad_token_id: int = 0,
+                 torch_dtype="bfloat16", **kw):
+        self.vocab_size = vocab_size
+        self.hidden_size = hidden_size
+        self.num_hidden_layers = num_hidden_layers
+        self.num_attention_heads = num_attention_heads
+        self.intermediate_size = intermediate_size
+        self.max_position_embeddings = max_position_embeddings
+        self.position_buckets = position_buckets
+        self.max_relative_positions = max_relative_positions
+        self.type_vocab_size = type_vocab_size
+        self.layer_norm_eps = layer_norm_eps
+        self.initializer_range = initializer_range
+        self.hidden_dropout = hidden_dropout
+        self.attention_dropout = attention_dropout
+        super().__init__(pad_token_id=pad_token_id, torch_dtype=torch_dtype, **kw)
+
+
+def deberta_tiny_config(**over):
+    cfg = dict(vocab_size=256, hidden_size=64, num_hidden_layers=2,
+               num_attention_heads=4, intermediate_size=128,
+               max_position_embeddings=128, position_buckets=32)
+    cfg.update(over)
+    return DebertaV2Config(**cfg)
+
+
+def build_relative_position(q_len: int, k_len: int, bucket_size: int,
+                            max_position: int, device) -> torch.Tensor:
+    """log-bucketed relative positions (HF deberta-v2 semantics)."""
+    q_ids = torch.arange(q_len, device=device)
+    k_ids = torch.arange(k_len, device=device)
+    rel = q_ids[:, None] - k_ids[None, :]
+    sign = torch.sign(rel)
+    mid = bucket_size // 2
+    abs_pos = torch.where((rel < mid) & (rel > -mid),
+                          torch.full_like(rel, mid - 1), rel.abs())
+    log_pos = (torch.ceil(
+        torch.log(abs_pos.float() / mid)
+        / math.log((max_position - 1) / mid) * (mid - 1)) + mid).long()
+    bucket = torch.where(abs_pos <= mid, rel, (log_pos * sign).long())
+    return bucket + bucket_size  # shift to [0, 2*bucket_size)
+
+
+class DisentangledSelfAttention(nn.Module):
+    """c2c + c2p + p2c attention (ref modeling_deberta_v2.py:590)."""
+
+    def __init__(self, config):
+        super().__init__()
+        self.num_heads = config.num_attention_heads
+        self.head_dim = config.hidden_size // config.num_attention_heads
+        self.qkv = nn.Linear(config.hidden_size, 3 * config.hidden_size)
+        self.pos_key = nn.Linear(config.hidden_size, config.hidden_size)
+        self.pos_query = nn.Linear(config.hidden_size, config.hidden_size)
+        self.out = nn.Linear(config.hidden_size, config.hidden_size)
+        self.dropout = nn.Dropout(config.attention_dropout)
+        self.position_buckets = config.position_buckets
+        self.max_relative_positions = config.max_relative_positions
+
+    def forward(self, x, rel_embeddings, attention_mask=None):
+        b, s, hdim = x.shape
+        np_, hn = self.num_heads, self.head_dim
+        q, k, v = self.qkv(x).chunk(3, dim=-1)
+        q = q.view(b, s, np_, hn).transpose(1, 2)
+        k = k.view(b, s, np_, hn).transpose(1, 2)
+        v = v.view(b, s, np_, hn).transpose(1, 2)
+        # 3 score components -> scale by 1/sqrt(3d)
+        scale = 1.0 / math.sqrt(hn * 3)
+
+        c2c = q.float() @ k.float().transpose(-1, -2)
+
+        rel_idx = build_relative_position(
+            s, s, self.position_buckets, self.max_relative_positions, x.device)
+        n_buckets = 2 * self.position_buckets
+        pe = rel_embeddings[:n_buckets]  # [2B, h]
+        pk = self.pos_key(pe).view(n_buckets, np_, hn).transpose(0, 1)  # [np,2B,hn]
+        pq = self.pos_query(pe).view(n_buckets, np_, hn).transpose(0, 1)
+
+        # c2p: score[i,j] += q_i . pos_key[bucket(i,j)]
+        c2p_all = torch.einsum("bnih,nrh->bnir", q.float(), pk.float())
+        c2p = torch.gather(
+            c2p_all, -1,
+            rel_idx[None, None, :, :].expand(b, np_, s, s))
+        # p2c: score[i,j] += k_j . pos_query[bucket(j,i)]
+        p2c_all = torch.einsum("bnjh,nrh->bnjr", k.float(), pq.float())
+        p2c = torch.gather(
+            p2c_all, -1,
+            rel_idx.t()[None, None, :, :].expand(b, np_, s, s)
+        ).transpose(-1, -2)
+
+        scores = (c2c + c2p + p2c) * scale
+        if attention_mask is not None:
+            scores = scores.masked_fill(
+                (attention_mask == 0)[:, None, None, :], float("-inf"))
+        probs = torch.softmax(scores, dim=-1).to(x.dtype)
+        probs = self.dropout(probs)
+        ctx = probs @ v
+        ctx = ctx.transpose(1, 2).reshape(b, s, hdim)
+        return self.out(ctx)
+
+
+class DebertaV2Layer(nn.Module):
+    def __init__(self, config):
+        super().__init__()
+        self.attn = DisentangledSelfAttention(config)
+        self.ln1 = LayerNorm(config.hidden_size, eps=config.layer_norm_eps)
+        im = init_normal(config.initializer_range)
+        om = scaled_init_normal(config.initializer_range,
+                                config.num_hidden_layers)
+        self.mlp = ParallelMLP(config.hidden_size, config.intermediate_size,
+                               init_method=im, output_init_method=om)
+        self.ln2 = LayerNorm(config.hidden_size, eps=config.layer_norm_eps)
+        self.dropout = nn.Dropout(config.hidden_dropout)
+
+    def forward(self, x, rel_embeddings, attention_mask=None):
+        h = self.attn(x, rel_embeddings, attention_mask)
+        x = self.ln1(x + self.dropout(h))  # post-LN like deberta
+        h = self.mlp(x)
+        return self.ln2(x + self.dropout(h))
+
+
+class DebertaV2PreTrainedModel(PreTrainedModel):
+    config_class = DebertaV2Config
+    base_model_prefix = "deberta"
+
+    def _init_weights(self, module):
+        if isinstance(module, (nn.Linear, nn.Embedding)):
+            module.weight.data.normal_(0.0, self.config.initializer_range)
+            if isinstance(module, nn.Linear) and module.bias is not None:
+                module.bias.data.zero_()
+
+
+class DebertaV2Model(DebertaV2PreTrainedModel):
+    def __init__(self, config):
+        super().__init__(config)
+        self.word_embeddings = nn.Embedding(config.vocab_size,
+                                            config.hidden_size)
+        self.token_type_embeddings = nn.Embedding(config.type_vocab_size,
+                                                  config.hidden_size)
+        self.emb_ln = LayerNorm(config.hidden_size, eps=config.layer_norm_eps)
+        self.dropout = nn.Dropout(config.hidden_dropout)
+        self.rel_embeddings = nn.Embedding(2 * config.position_buckets,
+                                           config.hidden_size)
+        self.layers = nn.ModuleList(
+            [DebertaV2Layer(config) for _ in range(config.num_hidden_layers)])
+        self.post_init()
+
+    def get_input_embeddings(self):
+        return self.word_embeddings
+
+    def forward(self, input_ids, attention_mask=None, token_type_ids=None,
+                **_kw):
+        if token_type_ids is None:
+            token_type_ids = torch.zeros_like(input_ids)
+        h = self.word_embeddings(input_ids) \
+            + self.token_type_embeddings(token_type_ids)
+        h = self.dropout(self.emb_ln(h))
+        rel = self.rel_embeddings.weight
+        for layer in self.layers:
+            h = layer(h, rel, attention_mask)
+        return BaseModelOutput(last_hidden_state=h)
+
+
+class DebertaV2ForMaskedLM(DebertaV2PreTrainedModel):
+    def __init__(self, config):
+        super().__init__(config)
+        self.deberta = DebertaV2Model(config)
+        self.dense = nn.Linear(config.hidden_size, config.hidden_size)
+        self.ln = LayerNorm(config.hidden_size, eps=config.layer_norm_eps)
+        self.decoder = nn.Linear(config.hidden_size, config.vocab_size)
+        self.post_init()
+
+    def forward(self, input_ids, attention_mask=None, token_type_ids=None,
+                labels=None, **_kw):
+        h = self.deberta(input_ids, attention_mask,
+                         token_type_ids).last_hidden_state
+        h = self.ln(F_ops.eager_gelu(self.dense(h).float()).to(h.dtype))
+        logits = self.decoder(h)
+        loss = None
+        if labels is not None:
+            loss = nn.functional.cross_entropy(
+                logits.float().view(-1, logits.shape[-1]), labels.view(-1),
+                ignore_index=-100)
+        return MaskedLMOutput(loss=loss, logits=logits)
+
+
+class DebertaV2ForSequenceClassification(DebertaV2PreTrainedModel):
+    def __init__(self, config):
+        super().__init__(config)
+        self.num_labels = getattr(config, "num_labels", 2)
+        self.deberta = DebertaV2Model(config)
+        self.pooler = nn.Linear(config.hidden_size, config.hidden_size)
+        self.classifier = nn.Linear(config.hidden_size, self.num_labels)
+        self.post_init()
+
+    def forward(self, input_ids, attention_mask=None, token_type_ids=None,
+                labels=None, **_kw):
+        h = self.deberta(input_ids, attention_mask,
+                         token_type_ids).last_hidden_state
+        pooled = torch.tanh(self.pooler(h[:, 0]))
+        logits = self.classifier(pooled)
+        loss = None
+        if labels is not None:
+            loss = nn.functional.cross_entropy(
+                logits.float().view(-1, self.num_labels), labels.view(-1))
+        return SequenceClassifierOutput(loss=loss, logits=logits)
