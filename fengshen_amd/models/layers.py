@@ -307,3 +307,105 @@ def parallel_lm_logits(hidden: torch.Tensor, word_embeddings_weight: torch.Tenso
     if parallel_output:
         return logits_parallel
     return gather_from_tensor_model_parallel_region(logits_parallel)
+
+
+class SinusoidalPositionalEmbedding(nn.Module):
+    """ref layers/positional_embeddings.py:19."""
+
+    def __init__(self, dim: int, base: float = 10000.0):
+        super().__init__()
+        self.dim = dim
+        self.base = base
+
+    def forward(self, positions: torch.Tensor) -> torch.Tensor:
+        half = self.dim // 2
+        inv = 1.0 / (self.base ** (torch.arange(
+            half, device=positions.device, dtype=torch.float32) / half))
+        ang = positions.float()[..., None] * inv
+        return torch.cat([ang.sin(), ang.cos()], dim=-1)
+
+
+class AliBi(nn.Module):
+    """Cached ALiBi bias matrix (ref positional_embeddings.py:90-173);
+    TP-aware: each rank holds its heads' slopes."""
+
+    def __init__(self, num_heads: int, mp_size: int = 1, mp_rank: int = 0):
+        super().__init__()
+        self.num_heads = num_heads
+        slopes = torch.tensor(self._slopes(num_heads))
+        per = num_heads // mp_size
+        self.register_buffer("slopes", slopes[mp_rank * per:(mp_rank + 1) * per],
+                             persistent=False)
+        self._cache = None
+
+    @staticmethod
+    def _slopes(n: int):
+        def pow2(n):
+            start = 2.0 ** (-(2.0 ** -(math.log2(n) - 3)))
+            return [start * (start ** i) for i in range(n)]
+        if math.log2(n).is_integer():
+            return pow2(n)
+        closest = 2 ** math.floor(math.log2(n))
+        return pow2(closest) + AliBi._slopes(2 * closest)[0::2][: n - closest]
+
+    def forward(self, sq: int, sk: int, device, dtype):
+        if (self._cache is None or self._cache.shape[-1] < sk
+                or self._cache.device != device):
+            pos = torch.arange(sk, device=device, dtype=torch.float32)
+            self._cache = self.slopes.to(device)[:, None, None] * \
+                (pos[None, None, :] - pos[None, :, None])
+        bias = self._cache[:, :sq, :sk] if self._cache.shape[1] >= sq else \
+            self._cache[:, -sq:, :sk]
+        return bias.to(dtype)
+
+
+class SoftEmbedding(nn.Module):
+    """Prompt-tuning soft tokens prepended to the input embedding
+    (ref word_embeddings.py:157)."""
+
+    def __init__(self, wte: nn.Module, n_tokens: int = 10,
+                 init_range: float = 0.5):
+        super().__init__()
+        self.wte = wte
+        self.n_tokens = n_tokens
+        dim = wte.weight.shape[1]
+        self.soft_prompt = nn.Parameter(
+            torch.empty(n_tokens, dim).uniform_(-init_range, init_range))
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        emb = self.wte(input_ids)
+        b = emb.shape[0]
+        prompt = self.soft_prompt.unsqueeze(0).expand(b, -1, -1).to(emb.dtype)
+        return torch.cat([prompt, emb], dim=1)
+
+
+class SpatialGatingUnit(nn.Module):
+    """gMLP SGU (ref layers/gmlp.py:28-141)."""
+
+    def __init__(self, dim_ff: int, seq_len: int):
+        super().__init__()
+        self.norm = LayerNorm(dim_ff // 2)
+        self.proj = nn.Linear(seq_len, seq_len)
+        nn.init.zeros_(self.proj.weight)
+        nn.init.ones_(self.proj.bias)
+
+    def forward(self, x):
+        res, gate = x.chunk(2, dim=-1)
+        gate = self.norm(gate)
+        gate = self.proj(gate.transpose(1, 2)).transpose(1, 2)
+        return res * gate
+
+
+class GMLPBlock(nn.Module):
+    def __init__(self, dim: int, dim_ff: int, seq_len: int):
+        super().__init__()
+        self.norm = LayerNorm(dim)
+        self.fc_in = nn.Linear(dim, dim_ff)
+        self.sgu = SpatialGatingUnit(dim_ff, seq_len)
+        self.fc_out = nn.Linear(dim_ff // 2, dim)
+
+    def forward(self, x):
+        h = self.norm(x)
+        h = F_ops.eager_gelu(self.fc_in(h).float()).to(h.dtype)
+        h = self.sgu(h)
+        return x + self.fc_out(h)

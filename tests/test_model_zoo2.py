@@ -136,3 +136,86 @@ def test_davae_elbo_and_sample():
     m.eval()
     gen = m.sample(2, 8)
     assert gen.shape == (2, 8)
+
+
+def test_tcbert_prompt_classification():
+    from fengshen_amd.models.tcbert.modeling_tcbert import TCBertModel
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    torch.manual_seed(0)
+    m = TCBertModel(bert_tiny_config())
+    ids = torch.randint(3, 256, (2, 20))
+    mask_pos = torch.tensor([[1, 2], [1, 2]])
+    label_tok = torch.randint(3, 256, (4, 2))  # 4 labels x 2 verbalizer toks
+    labels = torch.tensor([0, 3])
+    out = m(ids, mask_positions=mask_pos, label_token_ids=label_tok,
+            labels=labels)
+    assert out.loss.isfinite() and out.label_logits.shape == (2, 4)
+    out.loss.backward()
+
+
+def test_uniex_span_and_type():
+    from fengshen_amd.models.uniex.modeling_uniex import UniEXModel
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    torch.manual_seed(0)
+    m = UniEXModel(bert_tiny_config())
+    b, s = 2, 16
+    ids = torch.randint(3, 256, (b, s))
+    prompts = torch.randint(3, 256, (3, 6))  # 3 types
+    span_labels = torch.zeros(b, s, s)
+    span_labels[:, 2, 4] = 1
+    cand = torch.tensor([[[2, 4], [5, 7]]] * b)
+    type_labels = torch.tensor([[0, 2]] * b)
+    out = m(ids, label_prompt_ids=prompts, span_labels=span_labels,
+            candidate_spans=cand, type_labels=type_labels)
+    assert out.loss.isfinite()
+    out.loss.backward()
+    res = m.extract(ids, prompts, threshold=0.9)
+    assert len(res) == b
+
+
+def test_tcbert_pipeline():
+    from fengshen_amd.pipelines.tcbert import TCBertPipeline
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.tokenizer import SimpleCharTokenizer
+    torch.manual_seed(0)
+    pipe = TCBertPipeline(tokenizer=SimpleCharTokenizer(),
+                          config=bert_tiny_config(),
+                          labels=["体育", "科技", "财经"])
+    out = pipe("这是一段测试文本")
+    assert out["label_name"] in ["体育", "科技", "财经"]
+
+
+def test_llama_convert_roundtrip():
+    import torch as t
+    from fengshen_amd.utils.llama_convert import (
+        hf_to_fs_llama, fs_to_hf_llama, make_delta, apply_delta)
+    from fengshen_amd.models.llama.configuration_llama import llama_tiny_config
+    from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+    t.manual_seed(0)
+    m = LlamaForCausalLM(llama_tiny_config())
+    fs_sd = m.state_dict()
+    hf_sd = fs_to_hf_llama(fs_sd, num_layers=2)
+    back = hf_to_fs_llama(hf_sd, num_layers=2)
+    for k, v in fs_sd.items():
+        assert t.equal(v, back[k]), k
+    # delta roundtrip
+    target = {k: v + 0.5 for k, v in hf_sd.items()}
+    delta = make_delta(hf_sd, target)
+    rec = apply_delta(hf_sd, delta)
+    for k in hf_sd:
+        assert t.allclose(rec[k], target[k], atol=1e-6)
+
+
+def test_mmap_index_dataset(tmp_path):
+    from fengshen_amd.data.mmap_dataloader import (
+        MMapIndexDataset, convert_py_to_npy)
+    samples = [{"input_ids": list(range(5))},
+               {"input_ids": list(range(10, 17))}]
+    prefix = str(tmp_path / "simple")
+    convert_py_to_npy(samples, prefix)
+    ds = MMapIndexDataset(prefix)
+    assert len(ds) == 2
+    assert ds[1]["input_ids"].tolist() == list(range(10, 17))
