@@ -219,3 +219,19 @@ def test_mmap_index_dataset(tmp_path):
     ds = MMapIndexDataset(prefix)
     assert len(ds) == 2
     assert ds[1]["input_ids"].tolist() == list(range(10, 17))
+
+
+def test_int8_quantization_close():
+    from fengshen_amd.utils.quantize import quantize_model_int8
+    from fengshen_amd.models.llama.configuration_llama import llama_tiny_config
+    from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(llama_tiny_config()).eval()
+    ids = torch.randint(3, 256, (2, 16))
+    with torch.no_grad():
+        ref = m(ids).logits
+    quantize_model_int8(m)
+    with torch.no_grad():
+        q = m(ids).logits
+    rel = (q - ref).abs().max() / ref.abs().max()
+    assert rel < 0.1, rel.item()
