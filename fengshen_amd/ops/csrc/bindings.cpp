@@ -52,6 +52,8 @@ void fs_flash_attn_fwd(const void*, const void*, const void*, void*, float*,
 void fs_flash_attn_bwd(const void*, const void*, const void*, const void*,
                        const void*, const float*, void*, void*, void*, float*,
                        int, int, int, float, hipStream_t);
+void fs_w8_gemv(const void*, const float*, const void*, void*, int, int, int,
+                hipStream_t);
 }
 
 // ---------------------------------------------------------------------------
@@ -296,7 +298,21 @@ static std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
   return {dq, dk, dv};
 }
 
+// x [b, in] bf16, q8 [out, in] int8, scale [out] fp32 -> y [b, out] bf16
+static at::Tensor w8_gemv(at::Tensor q8, at::Tensor scale, at::Tensor x) {
+  TORCH_CHECK(q8.scalar_type() == at::kChar && x.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(q8.is_contiguous() && x.is_contiguous());
+  const int out = q8.size(0), in = q8.size(1);
+  TORCH_CHECK(in % 16 == 0, "in_features must be divisible by 16");
+  const int b = x.numel() / in;
+  auto y = at::empty({b, out}, x.options());
+  fs_w8_gemv(q8.data_ptr(), scale.data_ptr<float>(), x.data_ptr(),
+             y.data_ptr(), b, in, out, cur_stream());
+  return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("w8_gemv", &w8_gemv);
   mod.def("flash_attn_fwd", &flash_attn_fwd);
   mod.def("flash_attn_bwd", &flash_attn_bwd);
   mod.def("rms_norm_fwd", &rms_norm_fwd);

@@ -30,9 +30,17 @@ class W8Linear(nn.Module):
         self.out_features, self.in_features = weight.shape
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        w = (self.qweight.to(x.dtype)
-             * self.scale.to(x.dtype))
-        y = torch.nn.functional.linear(x, w)
+        from fengshen_amd.ops import use_hip, get_ext
+        rows = x.numel() // x.shape[-1]
+        if (use_hip(x) and x.dtype == torch.bfloat16 and rows <= 16
+                and x.shape[-1] % 16 == 0):
+            # decode path: fused int8 GEMV (weights stay int8 in HBM)
+            y = get_ext().w8_gemv(self.qweight, self.scale.view(-1),
+                                  x.reshape(rows, -1).contiguous())
+            y = y.view(*x.shape[:-1], self.out_features)
+        else:
+            w = (self.qweight.to(x.dtype) * self.scale.to(x.dtype))
+            y = torch.nn.functional.linear(x, w)
         if self.bias:
             y = y + self.bias_buf.to(x.dtype)
         return y

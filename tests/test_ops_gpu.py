@@ -249,3 +249,16 @@ def test_flash_attention_bwd():
     _close(q.grad, q2.grad, 3e-2)
     _close(k.grad, k2.grad, 3e-2)
     _close(v.grad, v2.grad, 3e-2)
+
+
+def test_w8_gemv_matches_dequant():
+    from fengshen_amd.ops import get_ext
+    torch.manual_seed(0)
+    out_f, in_f = 1024, 2048
+    w = torch.randn(out_f, in_f, device="cuda")
+    scale = w.abs().amax(1, keepdim=True) / 127.0
+    q8 = torch.round(w / scale).clamp(-127, 127).to(torch.int8)
+    x = _rand(3, in_f)
+    y = get_ext().w8_gemv(q8, scale.view(-1).float(), x)
+    ref = x.float() @ (q8.float() * scale).t()
+    _close(y, ref)
