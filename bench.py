@@ -82,7 +82,12 @@ def main():
         init_distributed(backend="nccl")
 
     torch.manual_seed(1234)
-    model, vocab, model_name = build_model(args.model, args.seq_len)
+    # construct directly on the GPU in bf16: 8 ranks x fp32-on-CPU would
+    # exhaust host RAM for 13B, and GPU-side init is much faster
+    torch.set_default_dtype(torch.bfloat16)
+    with device:
+        model, vocab, model_name = build_model(args.model, args.seq_len)
+    torch.set_default_dtype(torch.float32)
     model = model.to(torch.bfloat16).to(device)
     if hasattr(model, "gradient_checkpointing_enable"):
         model.gradient_checkpointing_enable()

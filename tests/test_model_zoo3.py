@@ -66,3 +66,20 @@ def test_hubert_dataset():
     batch = ds.collater([ds[0], ds[1]])
     assert batch["source"].shape[0] == 2
     assert batch["padding_mask"].dtype == torch.bool
+
+
+def test_auto_registry_roundtrip(tmp_path):
+    from fengshen_amd.models.auto import register_fengshen_auto_classes
+    register_fengshen_auto_classes()
+    from transformers import AutoConfig, AutoModelForCausalLM
+    from fengshen_amd.models.llama.configuration_llama import llama_tiny_config
+    from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(llama_tiny_config()).eval()
+    m.save_pretrained(tmp_path)
+    cfg = AutoConfig.from_pretrained(tmp_path)
+    assert cfg.model_type == "fengshen_llama"
+    m2 = AutoModelForCausalLM.from_pretrained(tmp_path).eval()
+    ids = torch.randint(3, 256, (1, 8))
+    with torch.no_grad():
+        assert torch.allclose(m(ids).logits, m2(ids).logits, atol=1e-5)

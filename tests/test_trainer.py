@@ -185,3 +185,16 @@ def test_validation_loop(tmp_path):
                       default_root_dir=str(tmp_path))
     trainer.fit(model, train_dataloaders=train, val_dataloaders=val)
     assert "val_loss" in trainer._metrics
+
+
+def test_step_profiler_callback(tmp_path):
+    from fengshen_amd.utils.profiling import StepProfiler
+    args = _make_args()
+    torch.manual_seed(0)
+    model = ToyModule(args)
+    loader = DataLoader(ToyDataset(n=128), batch_size=16)
+    prof = StepProfiler(start_step=2, num_steps=1, out_dir=str(tmp_path / "p"))
+    trainer = Trainer(max_steps=5, precision="fp32", callbacks=[prof],
+                      default_root_dir=str(tmp_path))
+    trainer.fit(model, train_dataloaders=loader)
+    assert (tmp_path / "p" / "step_profile.txt").exists()
