@@ -41,8 +41,7 @@ class TCBertPipeline(BasePipeline):
         vocab = self.tokenizer.get_vocab()
         batch_ids, mask_pos = [], []
         for t in texts:
-            prefix = self.prompt.format("") # masks inserted at the "{}" site
-            cut = self.prompt.index("{}")
+            cut = self.prompt.index("{}")  # masks inserted at the "{}" site
             ids = [self.tokenizer.cls_token_id]
             ids += [vocab.get(c, 4) for c in self.prompt[:cut]]
             mp = list(range(len(ids), len(ids) + self.n_mask))
