@@ -535,6 +535,126 @@ __global__ void fused_adamw_kernel(float* __restrict__ master,
 }
 
 // ===========================================================================
+// Wave-per-row softmax (sk <= 2048, sk % 8 == 0): the whole row lives in
+// registers (<=32 fp32/lane), reductions are wave shuffles — no LDS, no
+// __syncthreads.  4 rows per 256-thread block.  The LDS block-per-row
+// kernel above remains the fallback for larger sk.
+// ===========================================================================
+template <typename T, int MODE>
+__global__ void scaled_softmax_fwd_wave_kernel(
+    const T* __restrict__ x, const unsigned char* __restrict__ mask,
+    T* __restrict__ out, float scale, long rows, int sk, int sq,
+    int np, int mask_batches, int mask_sq) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int nchunks = (sk + 511) / 512;
+  float v[32];
+  for (long row = (long)blockIdx.x * 4 + wid; row < rows;
+       row += (long)gridDim.x * 4) {
+    const T* xr = x + row * sk;
+    T* yr = out + row * sk;
+    const int si = (int)(row % sq);
+    const unsigned char* mr = nullptr;
+    if (MODE == 1) {
+      const long bi = row / ((long)np * sq);
+      const long mb = (mask_batches == 1) ? 0 : bi;
+      const int msi = (mask_sq == 1) ? 0 : si;
+      mr = mask + ((long)mb * mask_sq + msi) * sk;
+    }
+    const int limit = (MODE == 2) ? (si + 1) : sk;
+
+    float mx = -INFINITY;
+#pragma unroll 4
+    for (int c = 0; c < nchunks; ++c) {
+      const int base = c * 512 + lane * 8;
+      if (base < sk) {
+        float t[8];
+        load8<T>(xr + base, t);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float val = t[j] * scale;
+          if (MODE == 1 && mr[base + j]) val = -10000.f;
+          if (MODE == 2 && base + j >= limit) val = -INFINITY;
+          v[c * 8 + j] = val;
+          mx = fmaxf(mx, val);
+        }
+      }
+    }
+    mx = wave_reduce_max(mx);
+    float sum = 0.f;
+#pragma unroll 4
+    for (int c = 0; c < nchunks; ++c) {
+      const int base = c * 512 + lane * 8;
+      if (base < sk) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float e = (v[c * 8 + j] == -INFINITY) ? 0.f
+                                                : __expf(v[c * 8 + j] - mx);
+          v[c * 8 + j] = e;
+          sum += e;
+        }
+      }
+    }
+    sum = wave_reduce_sum(sum);
+    const float inv = 1.f / sum;
+#pragma unroll 4
+    for (int c = 0; c < nchunks; ++c) {
+      const int base = c * 512 + lane * 8;
+      if (base < sk) {
+        float o[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) o[j] = v[c * 8 + j] * inv;
+        store8<T>(yr + base, o);
+      }
+    }
+  }
+}
+
+template <typename T>
+__global__ void scaled_softmax_bwd_wave_kernel(
+    const T* __restrict__ gy, const T* __restrict__ y, T* __restrict__ gx,
+    float scale, long rows, int sk) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int nchunks = (sk + 511) / 512;
+  float yv[32], gv[32];
+  for (long row = (long)blockIdx.x * 4 + wid; row < rows;
+       row += (long)gridDim.x * 4) {
+    const T* yr = y + row * sk;
+    const T* gr = gy + row * sk;
+    T* gxr = gx + row * sk;
+    float dot = 0.f;
+#pragma unroll 4
+    for (int c = 0; c < nchunks; ++c) {
+      const int base = c * 512 + lane * 8;
+      if (base < sk) {
+        float ty[8], tg[8];
+        load8<T>(yr + base, ty);
+        load8<T>(gr + base, tg);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          yv[c * 8 + j] = ty[j];
+          gv[c * 8 + j] = tg[j];
+          dot += ty[j] * tg[j];
+        }
+      }
+    }
+    dot = wave_reduce_sum(dot);
+#pragma unroll 4
+    for (int c = 0; c < nchunks; ++c) {
+      const int base = c * 512 + lane * 8;
+      if (base < sk) {
+        float o[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          o[j] = scale * yv[c * 8 + j] * (gv[c * 8 + j] - dot);
+        store8<T>(gxr + base, o);
+      }
+    }
+  }
+}
+
+// ===========================================================================
 // extern "C" launchers
 // ===========================================================================
 static inline int grid_for(long work_items, int block = 256) {
@@ -600,6 +720,29 @@ void fs_scaled_softmax_fwd(const void* x, const unsigned char* mask, void* out,
                            float scale, int rows, int sk, int sq, int np,
                            int mask_batches, int mask_sq, int mode, int dtype,
                            hipStream_t s) {
+  if (sk <= 2048 && sk % 8 == 0) {
+    // wave-per-row fast path (rows in registers)
+    long wgrid = ((long)rows + 3) / 4;
+    if (wgrid > 4096) wgrid = 4096;
+    DISPATCH_DTYPE(dtype, T, {
+      if (mode == 0)
+        hipLaunchKernelGGL((scaled_softmax_fwd_wave_kernel<T, 0>),
+                           dim3((unsigned)wgrid), dim3(256), 0, s,
+                           (const T*)x, mask, (T*)out, scale, (long)rows, sk,
+                           sq, np, mask_batches, mask_sq);
+      else if (mode == 1)
+        hipLaunchKernelGGL((scaled_softmax_fwd_wave_kernel<T, 1>),
+                           dim3((unsigned)wgrid), dim3(256), 0, s,
+                           (const T*)x, mask, (T*)out, scale, (long)rows, sk,
+                           sq, np, mask_batches, mask_sq);
+      else
+        hipLaunchKernelGGL((scaled_softmax_fwd_wave_kernel<T, 2>),
+                           dim3((unsigned)wgrid), dim3(256), 0, s,
+                           (const T*)x, mask, (T*)out, scale, (long)rows, sk,
+                           sq, np, mask_batches, mask_sq);
+    });
+    return;
+  }
   int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
   size_t lds = (size_t)sk * 4;
   DISPATCH_DTYPE(dtype, T, {
@@ -620,6 +763,16 @@ void fs_scaled_softmax_fwd(const void* x, const unsigned char* mask, void* out,
 
 void fs_scaled_softmax_bwd(const void* gy, const void* y, void* gx, float scale,
                            int rows, int sk, int dtype, hipStream_t s) {
+  if (sk <= 2048 && sk % 8 == 0) {
+    long wgrid = ((long)rows + 3) / 4;
+    if (wgrid > 4096) wgrid = 4096;
+    DISPATCH_DTYPE(dtype, T,
+      hipLaunchKernelGGL((scaled_softmax_bwd_wave_kernel<T>),
+                         dim3((unsigned)wgrid), dim3(256), 0, s,
+                         (const T*)gy, (const T*)y, (T*)gx, scale, (long)rows,
+                         sk));
+    return;
+  }
   int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
   size_t lds = (size_t)sk * 8;
   DISPATCH_DTYPE(dtype, T,
@@ -743,3 +896,4 @@ extern "C" void fs_w8_gemv(const void* q8, const float* scale, const void* x,
                      (const signed char*)q8, scale, (const bf16_t*)x,
                      (bf16_t*)y, batch, in_features, out_features);
 }
+
