@@ -29,6 +29,7 @@ def parse_args():
     p.add_argument("--seq_len", type=int, default=2048)
     p.add_argument("--micro_batch", type=int, default=16)
     p.add_argument("--zero_stage", type=int, default=3)
+    p.add_argument("--tensor_model_parallel_size", "--tp", type=int, default=1)
     p.add_argument("--lr", type=float, default=1e-5)
     return p.parse_args()
 
@@ -78,8 +79,11 @@ def main():
     from fengshen_amd.ops import has_ext
     assert has_ext(), "HIP extension must be built (python -m fengshen_amd.ops.build)"
 
-    if world > 1:
+    if world > 1 or args.tensor_model_parallel_size > 1:
         init_distributed(backend="nccl")
+    if args.tensor_model_parallel_size > 1:
+        from fengshen_amd.parallel.groups import initialize_model_parallel
+        initialize_model_parallel(args.tensor_model_parallel_size)
 
     torch.manual_seed(1234)
     # construct directly on the GPU in bf16: 8 ranks x fp32-on-CPU would
@@ -102,9 +106,13 @@ def main():
                             lr=args.lr, betas=(0.9, 0.95), eps=1e-8,
                             weight_decay=0.1)
 
+    from fengshen_amd.parallel import groups as pgroups
+    dp_rank = pgroups.get_data_parallel_rank()
+    dp_world = pgroups.get_data_parallel_world_size()
     is_bert = args.model == "erlangshen-1.3b"
     b, s = args.micro_batch, args.seq_len
-    g = torch.Generator(device="cpu").manual_seed(42 + rank)
+    # TP ranks must see identical batches: seed by DP rank
+    g = torch.Generator(device="cpu").manual_seed(42 + dp_rank)
     # several distinct synthetic batches (avoids single-batch memorization)
     batches = []
     for _ in range(4):
@@ -151,8 +159,8 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = t.item()
 
-    tokens_per_step = b * s * world
-    samples_per_step = b * world
+    tokens_per_step = b * s * dp_world
+    samples_per_step = b * dp_world
     value = (samples_per_step if is_bert else tokens_per_step) \
         * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
@@ -180,7 +188,10 @@ def main():
                 "global_batch": b * world,
                 "micro_batch": b,
                 "seq_len": s,
-                "parallelism": f"zero{args.zero_stage}_dp{world}",
+                "parallelism": (
+                    f"zero{args.zero_stage}_dp{dp_world}"
+                    + (f"_tp{args.tensor_model_parallel_size}"
+                       if args.tensor_model_parallel_size > 1 else "")),
                 "activation_checkpointing": True,
             },
         }), flush=True)
