@@ -160,3 +160,31 @@ def test_masking_utils():
     assert len(pos) == len(lab) > 0
     for p, l in zip(pos, lab):
         assert tokens[p] == l
+
+
+def test_bert_mmap_dataset(tmp_path):
+    from fengshen_amd.data.bert_dataset import BertMmapDataset
+    from fengshen_amd.data.indexed_dataset import MMapIndexedDatasetBuilder, MMapIndexedDataset
+    rng = np.random.RandomState(0)
+    prefix = str(tmp_path / "bertcorp")
+    builder = MMapIndexedDatasetBuilder(prefix + ".bin", dtype=np.int32)
+    for _doc in range(6):
+        for _sent in range(rng.randint(2, 5)):
+            builder.add_item(rng.randint(10, 200, size=rng.randint(5, 20)))
+        builder.end_document()
+    builder.finalize(prefix + ".idx")
+    ds = MMapIndexedDataset(prefix)
+    vocab = list(range(10, 200))
+    id2tok = {i: f"t{i}" for i in range(210)}
+    bert_ds = BertMmapDataset(ds, vocab, id2tok, cls_id=1, sep_id=2,
+                              mask_id=3, pad_id=0, max_seq_length=64,
+                              num_epochs=2)
+    assert len(bert_ds) > 0
+    item = bert_ds[0]
+    assert item["input_ids"].shape == (64,)
+    assert item["input_ids"][0] == 1  # CLS
+    assert (item["labels"] != -100).sum() > 0
+    assert item["next_sentence_label"].item() in (0, 1)
+    # deterministic per-sample
+    item2 = bert_ds[0]
+    assert torch.equal(item["input_ids"], item2["input_ids"])
