@@ -89,30 +89,52 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
   const int q_hi = qb * FA_QBLK + FA_QBLK - 1;
   const int n_kv_tiles = (q_hi / FA_KVBLK) + 1;
 
+  // async-STAGE split (guide §6 G15): global loads for tile t+1 are issued
+  // DURING tile t's compute (registers k_reg/v_reg), and only the cheap
+  // LDS writes sit between the barriers — HBM latency hides under MFMA.
+  const int tid = threadIdx.x;
+  bf16x8 k_reg[4], v_reg[4];
+#pragma unroll
+  for (int sweep = 0; sweep < 4; ++sweep) {
+    const int i = tid * 8 + sweep * 2048;
+    const int kr = i / FA_D;
+    const int kc = i % FA_D;
+    k_reg[sweep] = *reinterpret_cast<const bf16x8*>(Kp + (long)kr * FA_D + kc);
+    v_reg[sweep] = *reinterpret_cast<const bf16x8*>(Vp + (long)kr * FA_D + kc);
+  }
+
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
     const int k_base = kt * FA_KVBLK;
     __syncthreads();
-    {
-      // stage K (swizzled) + V^T: 64*128 = 8192 elems each; 16B per thread
-      // per sweep => 4 sweeps
-      const int tid = threadIdx.x;
-      for (int i = tid * 8; i < FA_KVBLK * FA_D; i += 256 * 8) {
-        const int kr = i / FA_D;
-        const int kc = i % FA_D;
-        bf16x8 kk = *reinterpret_cast<const bf16x8*>(
-            Kp + (long)(k_base + kr) * FA_D + kc);
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(&k_lds[kr][0]) + kswz(kr, kc * 2)) = kk;
-        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
-            Vp + (long)(k_base + kr) * FA_D + kc);
-        // kv index XOR-swizzled by d (8-granular) so the transpose scatter
-        // spreads banks (unswizzled: 16 lanes of one K-row hit ONE bank)
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          vt_lds[kc + j][kr ^ ((kc + j) & 0x38)] = vv[j];  // XOR bits 3-5 of d
-      }
+    for (int sweep = 0; sweep < 4; ++sweep) {
+      const int i = tid * 8 + sweep * 2048;
+      const int kr = i / FA_D;
+      const int kc = i % FA_D;
+      *reinterpret_cast<bf16x8*>(
+          reinterpret_cast<char*>(&k_lds[kr][0]) + kswz(kr, kc * 2)) =
+          k_reg[sweep];
+      bf16x8 vv = v_reg[sweep];
+      // kv index XOR-swizzled by d bits 3-5 (the transpose scatter would
+      // otherwise put 16 lanes of one K-row into ONE bank)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        vt_lds[kc + j][kr ^ ((kc + j) & 0x38)] = vv[j];
     }
     __syncthreads();
+    if (kt + 1 < n_kv_tiles) {
+      const int nb = (kt + 1) * FA_KVBLK;
+#pragma unroll
+      for (int sweep = 0; sweep < 4; ++sweep) {
+        const int i = tid * 8 + sweep * 2048;
+        const int kr = i / FA_D;
+        const int kc = i % FA_D;
+        k_reg[sweep] = *reinterpret_cast<const bf16x8*>(
+            Kp + (long)(nb + kr) * FA_D + kc);
+        v_reg[sweep] = *reinterpret_cast<const bf16x8*>(
+            Vp + (long)(nb + kr) * FA_D + kc);
+      }
+    }
 
     // ---- S = (sQ) @ K^T : four 16x16 n-tiles ------------------------------
     f32x4 s_acc[4];
