@@ -20,9 +20,9 @@ typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 #define FA_D 128
-#define FA_QBLK 64
+#define FA_QBLK 128          // 8 waves x 16 q rows: 2x compute per barrier
 #define FA_KVBLK 64
-#define FA_WAVES 4
+#define FA_WAVES 8
 #define FA_VPAD 8
 
 // XOR swizzle for k_lds rows (applies to 16B-aligned byte offsets)
@@ -55,6 +55,10 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
   bf16_t* Op = O + bh_off;
 
   const int q0 = qb * FA_QBLK + wave * 16;
+  // partial blocks (s % FA_QBLK != 0): OOB waves compute on clamped rows
+  // (must still hit every __syncthreads) and skip their stores
+  const bool q_active = q0 < s;
+  const int q0c = q_active ? q0 : s - 16;
 
   // ---- Q tile -> A-fragments (layout LA0: row=l%16, k=(l/16)*8+j;
   // verified on HW by scripts/mfma_probe), pre-scaled -----------------------
@@ -65,7 +69,7 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       bf16x8 raw = *reinterpret_cast<const bf16x8*>(
-          Qp + (long)(q0 + row) * FA_D + c * 32 + k0);
+          Qp + (long)(q0c + row) * FA_D + c * 32 + k0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         unsigned short u = (unsigned short)raw[j];
@@ -86,17 +90,17 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
   for (int t = 0; t < 8; ++t) o_acc[t] = f32x4{0, 0, 0, 0};
 
-  const int q_hi = qb * FA_QBLK + FA_QBLK - 1;
+  const int q_hi = min(qb * FA_QBLK + FA_QBLK, s) - 1;
   const int n_kv_tiles = (q_hi / FA_KVBLK) + 1;
 
   // async-STAGE split (guide §6 G15): global loads for tile t+1 are issued
   // DURING tile t's compute (registers k_reg/v_reg), and only the cheap
   // LDS writes sit between the barriers — HBM latency hides under MFMA.
   const int tid = threadIdx.x;
-  bf16x8 k_reg[4], v_reg[4];
+  bf16x8 k_reg[2], v_reg[2];
 #pragma unroll
-  for (int sweep = 0; sweep < 4; ++sweep) {
-    const int i = tid * 8 + sweep * 2048;
+  for (int sweep = 0; sweep < 2; ++sweep) {
+    const int i = tid * 8 + sweep * 4096;
     const int kr = i / FA_D;
     const int kc = i % FA_D;
     k_reg[sweep] = *reinterpret_cast<const bf16x8*>(Kp + (long)kr * FA_D + kc);
@@ -107,8 +111,8 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
     const int k_base = kt * FA_KVBLK;
     __syncthreads();
 #pragma unroll
-    for (int sweep = 0; sweep < 4; ++sweep) {
-      const int i = tid * 8 + sweep * 2048;
+    for (int sweep = 0; sweep < 2; ++sweep) {
+      const int i = tid * 8 + sweep * 4096;
       const int kr = i / FA_D;
       const int kc = i % FA_D;
       *reinterpret_cast<bf16x8*>(
@@ -125,8 +129,8 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
     if (kt + 1 < n_kv_tiles) {
       const int nb = (kt + 1) * FA_KVBLK;
 #pragma unroll
-      for (int sweep = 0; sweep < 4; ++sweep) {
-        const int i = tid * 8 + sweep * 2048;
+      for (int sweep = 0; sweep < 2; ++sweep) {
+        const int i = tid * 8 + sweep * 4096;
         const int kr = i / FA_D;
         const int kc = i % FA_D;
         k_reg[sweep] = *reinterpret_cast<const bf16x8*>(
@@ -162,7 +166,7 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
     float tile_max[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int qrow = q0 + (lane >> 4) * 4 + r;
+      const int qrow = q0c + (lane >> 4) * 4 + r;
       float tm = -INFINITY;
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
@@ -240,6 +244,7 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
   }
 
   // ---- epilogue -----------------------------------------------------------
+  if (!q_active) return;  // after the last barrier: safe to exit
   const int col = lane & 15;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -260,7 +265,7 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
 extern "C" void fs_flash_attn_fwd(const void* q, const void* k, const void* v,
                                   void* o, float* lse, int b, int h, int s,
                                   float scale, hipStream_t stream) {
-  dim3 grid(s / FA_QBLK, h, b);
+  dim3 grid((s + FA_QBLK - 1) / FA_QBLK, h, b);
   dim3 block(FA_WAVES * 64);
   hipLaunchKernelGGL(flash_attn_fwd_kernel, grid, block, 0, stream,
                      (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
@@ -328,6 +333,8 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
   const float* dlt = Delta + bh * s;
 
   const int q0 = blockIdx.x * FA_QBLK + wave * 16;
+  const bool q_active = q0 < s;
+  const int q0c = q_active ? q0 : s - 16;
 
   bf16x8 q_frag[4], do_frag[4];
   {
@@ -336,7 +343,7 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       bf16x8 raw = *reinterpret_cast<const bf16x8*>(
-          Qp + (long)(q0 + row) * FA_D + c * 32 + k0);
+          Qp + (long)(q0c + row) * FA_D + c * 32 + k0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         unsigned short u = (unsigned short)raw[j];
@@ -345,13 +352,13 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
       }
       q_frag[c] = raw;
       do_frag[c] = *reinterpret_cast<const bf16x8*>(
-          dOp + (long)(q0 + row) * FA_D + c * 32 + k0);
+          dOp + (long)(q0c + row) * FA_D + c * 32 + k0);
     }
   }
   float lse_r[4], dlt_r[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    const int qrow = q0 + (lane >> 4) * 4 + r;
+    const int qrow = q0c + (lane >> 4) * 4 + r;
     lse_r[r] = lse[qrow];
     dlt_r[r] = dlt[qrow];
   }
@@ -360,7 +367,7 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
   for (int t = 0; t < 8; ++t) dq_acc[t] = f32x4{0, 0, 0, 0};
 
-  const int q_hi = blockIdx.x * FA_QBLK + FA_QBLK - 1;
+  const int q_hi = min(blockIdx.x * FA_QBLK + FA_QBLK, s) - 1;
   const int n_kv_tiles = (q_hi / FB_T) + 1;
 
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
@@ -415,7 +422,7 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
     const int col = lane & 15;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      const int qrow = q0 + (lane >> 4) * 4 + r;
+      const int qrow = q0c + (lane >> 4) * 4 + r;
       const int row = (lane >> 4) * 4 + r;
 #pragma unroll
       for (int nt = 0; nt < 2; ++nt) {
@@ -443,6 +450,7 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
     }
   }
 
+  if (!q_active) return;
   const int col = lane & 15;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -482,8 +490,12 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
   const float* lse = LSE + bh * s;
   const float* dlt = Delta + bh * s;
 
-  const int kv0_blk = blockIdx.x * FA_KVBLK;
+  // each block owns FA_WAVES*16 kv rows (grid is s / FA_QBLK with
+  // FA_QBLK == FA_WAVES*16)
+  const int kv0_blk = blockIdx.x * (FA_WAVES * 16);
   const int kv0_wave = kv0_blk + wave * 16;
+  const bool kv_active = kv0_wave < s;
+  const int kv0c = kv_active ? kv0_wave : s - 16;
 
   bf16x8 k_frag[4], v_frag[4];
   {
@@ -492,7 +504,7 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       bf16x8 raw = *reinterpret_cast<const bf16x8*>(
-          Kp + (long)(kv0_wave + row) * FA_D + c * 32 + c0);
+          Kp + (long)(kv0c + row) * FA_D + c * 32 + c0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         unsigned short u = (unsigned short)raw[j];
@@ -501,7 +513,7 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
       }
       k_frag[c] = raw;
       v_frag[c] = *reinterpret_cast<const bf16x8*>(
-          Vp + (long)(kv0_wave + row) * FA_D + c * 32 + c0);
+          Vp + (long)(kv0c + row) * FA_D + c * 32 + c0);
     }
   }
 
@@ -575,7 +587,7 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
       const float dlt_c = dlt[qcol];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int kvrow = kv0_wave + (lane >> 4) * 4 + r;
+        const int kvrow = kv0c + (lane >> 4) * 4 + r;
         float pv = (qcol < kvrow) ? 0.f : __expf(st_acc[nt][r] - lse_c);
         pt[nt][r] = pv;
         dpt_acc[nt][r] = pv * (dpt_acc[nt][r] - dlt_c) * scale;  // = dSt
@@ -630,6 +642,7 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
     }
   }
 
+  if (!kv_active) return;
   const int col = lane & 15;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -655,7 +668,7 @@ extern "C" void fs_flash_attn_bwd(const void* q, const void* k, const void* v,
                        0, stream, (const bf16_t*)dout, (const bf16_t*)o,
                        delta_ws, rows);
   }
-  dim3 grid(s / FA_QBLK, h, b);
+  dim3 grid((s + FA_QBLK - 1) / FA_QBLK, h, b);
   dim3 block(FA_WAVES * 64);
   hipLaunchKernelGGL(flash_attn_bwd_dq_kernel, grid, block, 0, stream,
                      (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
