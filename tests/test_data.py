@@ -188,3 +188,30 @@ def test_bert_mmap_dataset(tmp_path):
     # deterministic per-sample
     item2 = bert_ds[0]
     assert torch.equal(item["input_ids"], item2["input_ids"])
+
+
+def test_bert_preprocessing_pipeline(tmp_path):
+    import json
+    from fengshen_amd.data.bert_preprocessing import (
+        split_shards, presplit_sentences, jsonl_to_mmap)
+    from fengshen_amd.data.indexed_dataset import MMapIndexedDataset
+    from fengshen_amd.tokenizer import SimpleCharTokenizer
+    src = tmp_path / "corpus.jsonl"
+    with open(src, "w", encoding="utf-8") as f:
+        for i in range(20):
+            f.write(json.dumps(
+                {"text": f"第{i}篇文章的第一句。第二句内容在这里。最后一句。"},
+                ensure_ascii=False) + "\n")
+    shards = split_shards(str(src), str(tmp_path / "shards"), shard_bytes=500)
+    assert len(shards) > 1
+    n = presplit_sentences(str(src), str(tmp_path / "presplit.jsonl"))
+    assert n == 20
+    with open(tmp_path / "presplit.jsonl", encoding="utf-8") as f:
+        doc = json.loads(f.readline())
+        assert len(doc["sentences"]) == 3
+    tk = SimpleCharTokenizer()
+    docs = jsonl_to_mmap(str(src), str(tmp_path / "corpus"), tk)
+    assert docs == 20
+    ds = MMapIndexedDataset(str(tmp_path / "corpus"))
+    assert len(ds.doc_idx) == 21  # 20 docs + leading 0
+    assert len(ds) == 60  # 3 sentences each
