@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import json
 import os
-from typing import Callable, Iterator, Optional
+from typing import Optional
 
 from fengshen_amd.data.data_utils import ChineseSentenceSplitter
 

@@ -9,7 +9,7 @@ ChineseSentenceSplitter (sentence_split.py:4).
 from __future__ import annotations
 
 import re
-from typing import List, Optional, Sequence, Tuple
+from typing import List, Sequence, Tuple
 
 import numpy as np
 

@@ -6,7 +6,7 @@ blendable_dataset.py:64 (weighted corpus mix).
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import Optional, Sequence
 
 import numpy as np
 import torch

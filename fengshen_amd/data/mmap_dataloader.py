@@ -7,7 +7,6 @@ and utils/convert_py_to_npy.py:20.
 """
 from __future__ import annotations
 
-import os
 from typing import Dict, List, Sequence
 
 import numpy as np

@@ -10,7 +10,6 @@ or passed in directly.
 from __future__ import annotations
 
 import argparse
-import os
 from typing import Optional
 
 from torch.utils.data import DataLoader, DistributedSampler

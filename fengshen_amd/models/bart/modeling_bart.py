@@ -6,8 +6,6 @@ blocks, tied LM head + final bias) on our parallel layer library.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn as nn
 from transformers import PretrainedConfig, PreTrainedModel

@@ -7,8 +7,6 @@ bucketed relative positions).
 from __future__ import annotations
 
 import math
-from typing import Optional
-
 import torch
 import torch.nn as nn
 from transformers import PretrainedConfig, PreTrainedModel

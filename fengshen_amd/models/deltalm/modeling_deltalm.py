@@ -8,8 +8,6 @@ the layer count.  Built on our parallel enc-dec library.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn as nn
 from transformers import PretrainedConfig, PreTrainedModel

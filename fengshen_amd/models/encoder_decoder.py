@@ -16,8 +16,6 @@ from fengshen_amd.models.layers import (
     LayerNorm,
     ParallelAttention,
     ParallelMLP,
-    init_normal,
-    scaled_init_normal,
 )
 from fengshen_amd.ops import functional as F_ops
 from fengshen_amd.parallel.layers import (

@@ -7,8 +7,6 @@ layer library with fused HIP ops on the hot path.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn as nn
 from transformers import PreTrainedModel

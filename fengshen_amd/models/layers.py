@@ -13,7 +13,7 @@ ParallelTransformerLayer :626, parallel_lm_logits :800) — redesigned:
 from __future__ import annotations
 
 import math
-from typing import Callable, List, Optional, Tuple
+from typing import Callable, List, Optional
 
 import torch
 import torch.nn as nn
@@ -31,7 +31,6 @@ from fengshen_amd.parallel.mappings import (
     copy_to_tensor_model_parallel_region,
     gather_from_tensor_model_parallel_region,
 )
-from fengshen_amd.parallel.random import get_rng_tracker
 
 
 def init_normal(std: float) -> Callable:

@@ -11,8 +11,6 @@ from __future__ import annotations
 
 import argparse
 import math
-from typing import Optional
-
 import torch
 from torch.optim.lr_scheduler import LambdaLR
 

@@ -6,8 +6,6 @@ compact KL autoencoder with the standard 0.18215 latent scaling.
 """
 from __future__ import annotations
 
-from typing import Tuple
-
 import torch
 import torch.nn as nn
 from transformers import PretrainedConfig, PreTrainedModel

@@ -7,7 +7,7 @@ logits at the mask positions for each candidate label's tokens.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import List, Optional
+from typing import Optional
 
 import torch
 import torch.nn as nn

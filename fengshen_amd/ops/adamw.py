@@ -8,7 +8,6 @@ csrc/adamw.hip; this file holds the dispatch + the eager oracle.
 """
 from __future__ import annotations
 
-import math
 
 import torch
 

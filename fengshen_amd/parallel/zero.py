@@ -28,8 +28,7 @@ Stage semantics:
 """
 from __future__ import annotations
 
-import math
-from typing import Dict, Iterable, List, Optional
+from typing import Dict, List, Optional
 
 import torch
 import torch.distributed as dist

@@ -8,7 +8,6 @@ __call__ preprocess->forward->postprocess.
 from __future__ import annotations
 
 import argparse
-from typing import Any, Dict, List, Optional
 
 import torch
 

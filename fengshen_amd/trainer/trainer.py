@@ -15,10 +15,8 @@ from __future__ import annotations
 
 import argparse
 import logging
-import math
 import os
 import random
-import time
 from typing import Any, Dict, List, Optional
 
 import numpy as np
