@@ -215,3 +215,36 @@ def test_bert_preprocessing_pipeline(tmp_path):
     ds = MMapIndexedDataset(str(tmp_path / "corpus"))
     assert len(ds.doc_idx) == 21  # 20 docs + leading 0
     assert len(ds) == 60  # 3 sentences each
+
+
+def test_bart_mmap_dataset(tmp_path):
+    from fengshen_amd.data.bart_dataset import BartMmapDataset
+    from fengshen_amd.data.indexed_dataset import (
+        MMapIndexedDataset,
+        MMapIndexedDatasetBuilder,
+    )
+    rng = np.random.RandomState(0)
+    prefix = str(tmp_path / "bartcorp")
+    builder = MMapIndexedDatasetBuilder(prefix + ".bin", dtype=np.int32)
+    for _doc in range(6):
+        for _sent in range(rng.randint(2, 5)):
+            builder.add_item(rng.randint(10, 200, size=rng.randint(5, 20)))
+        builder.end_document()
+    builder.finalize(prefix + ".idx")
+    ds = MMapIndexedDataset(prefix)
+    id2tok = {i: f"t{i}" for i in range(210)}
+    bart_ds = BartMmapDataset(ds, id2tok, cls_id=1, sep_id=2, mask_id=3,
+                              pad_id=0, vocab_size=210, max_seq_length=64,
+                              num_epochs=2)
+    assert len(bart_ds) > 0
+    item = bart_ds[0]
+    assert item["input_ids"].shape == (64,)
+    assert item["input_ids"][0] == 1                      # CLS kept
+    assert (item["input_ids"] == 3).sum() > 0             # masking happened
+    assert (item["labels"] != -100).sum() > 0
+    # labels are the clean shifted stream, never containing [MASK]
+    assert (item["labels"][item["labels"] != -100] != 3).all()
+    # deterministic per-sample
+    assert torch.equal(item["input_ids"], bart_ds[0]["input_ids"])
+    # different samples get different noise
+    assert not torch.equal(bart_ds[0]["input_ids"], bart_ds[1]["input_ids"])
