@@ -1,0 +1,119 @@
+"""Randeng-T5 QA finetune (reading comprehension).
+
+Behavioral parity: reference examples/qa_t5/finetune_t5_cmrc.py —
+"question:{q} context:{c}" -> answer, T5 conditional generation.
+
+Run:
+  torchrun --standalone --nproc-per-node N finetune_t5_qa.py \
+    --strategy zero2 --max_steps 2000 --train_file cmrc.jsonl
+With no --train_file a synthetic dataset is generated (smoke mode).
+"""
+from __future__ import annotations
+
+import os
+import sys
+
+sys.path.insert(0, os.path.abspath(os.path.join(
+    os.path.dirname(__file__), "..", "..")))
+
+
+import argparse
+
+import torch
+
+from fengshen_amd import FengshenModule, Trainer, UniversalDataModule
+from fengshen_amd.models.model_utils import add_module_args, configure_optimizers
+from fengshen_amd.models.t5.modeling_t5 import T5Config, T5ForConditionalGeneration
+from fengshen_amd.trainer.callbacks import LearningRateMonitor, ThroughputMonitor
+from fengshen_amd.utils.universal_checkpoint import UniversalCheckpoint
+
+
+def t5_tiny_config(**over):
+    cfg = dict(vocab_size=256, hidden_size=64, num_layers=2,
+               num_decoder_layers=2, num_attention_heads=4,
+               intermediate_size=128)
+    cfg.update(over)
+    return T5Config(**cfg)
+
+
+class QACollator:
+    def __init__(self, tokenizer, max_src=384, max_tgt=48):
+        self.tk = tokenizer
+        self.max_src, self.max_tgt = max_src, max_tgt
+
+    def __call__(self, samples):
+        src_ids, tgt_ids = [], []
+        for s in samples:
+            src = f"question:{s['question']} context:{s['context']}"
+            src_ids.append(self.tk.encode(src)[:self.max_src])
+            tgt_ids.append(self.tk.encode(s["answer"])[:self.max_tgt])
+        pad = self.tk.pad_token_id or 0
+        Ls = max(len(x) for x in src_ids)
+        Lt = max(len(x) for x in tgt_ids)
+        return {
+            "input_ids": torch.tensor(
+                [x + [pad] * (Ls - len(x)) for x in src_ids]),
+            "attention_mask": torch.tensor(
+                [[1] * len(x) + [0] * (Ls - len(x)) for x in src_ids]),
+            "labels": torch.tensor(
+                [x + [-100] * (Lt - len(x)) for x in tgt_ids]),
+        }
+
+
+class T5QA(FengshenModule):
+    def __init__(self, args):
+        super().__init__()
+        self.save_hyperparameters(args)
+        cfg = t5_tiny_config() if args.model_size == "tiny" else T5Config()
+        self.model = T5ForConditionalGeneration(cfg)
+
+    def training_step(self, batch, batch_idx):
+        out = self.model(**batch)
+        self.log("train_loss", out.loss)
+        return out.loss
+
+    def configure_optimizers(self):
+        return configure_optimizers(self)
+
+
+def synthetic_qa(n=256):
+    import random
+    rng = random.Random(0)
+    items = [("中国的首都是哪里？", "北京是中国的首都。", "北京"),
+             ("亚洲最长的河流是什么？", "长江是亚洲最长的河流。", "长江")]
+    return [dict(zip(("question", "context", "answer"), rng.choice(items)))
+            for _ in range(n)]
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--model_size", default="tiny",
+                        choices=["tiny", "base"])
+    parser.add_argument("--tokenizer", default=None)
+    add_module_args(parser)
+    UniversalDataModule.add_data_specific_args(parser)
+    Trainer.add_argparse_args(parser)
+    UniversalCheckpoint.add_argparse_args(parser)
+    args = parser.parse_args()
+
+    if args.tokenizer:
+        from transformers import AutoTokenizer
+        tokenizer = AutoTokenizer.from_pretrained(args.tokenizer)
+    else:
+        from fengshen_amd.tokenizer import SimpleCharTokenizer as FakeTokenizer
+        tokenizer = FakeTokenizer()
+
+    datasets = None
+    if not args.train_file and not args.datasets_name:
+        datasets = {"train": synthetic_qa()}
+    dm = UniversalDataModule(tokenizer, QACollator(tokenizer), args,
+                             datasets=datasets)
+    module = T5QA(args)
+    trainer = Trainer.from_argparse_args(
+        args, callbacks=[LearningRateMonitor(), ThroughputMonitor(),
+                         UniversalCheckpoint(args)])
+    trainer.fit(module, datamodule=dm)
+
+
+if __name__ == "__main__":
+    main()

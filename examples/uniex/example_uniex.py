@@ -1,0 +1,59 @@
+"""UniEX unified information-extraction demo.
+
+Behavioral parity: reference examples/uniex/example.py — span scorer +
+type matching against label prompts, fast extract mode.
+
+Run:  python example_uniex.py [--model_path ...]
+With no --model_path a tiny random-weight model runs (smoke mode).
+"""
+from __future__ import annotations
+
+import os
+import sys
+
+sys.path.insert(0, os.path.abspath(os.path.join(
+    os.path.dirname(__file__), "..", "..")))
+
+
+import argparse
+
+import torch
+
+from fengshen_amd.models.uniex.modeling_uniex import UniEXModel
+from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+    bert_tiny_config,
+)
+from fengshen_amd.tokenizer import SimpleCharTokenizer
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--model_path", default=None)
+    args = parser.parse_args()
+
+    tokenizer = SimpleCharTokenizer()
+    if args.model_path:
+        model = UniEXModel.from_pretrained(args.model_path)
+    else:
+        model = UniEXModel(bert_tiny_config(vocab_size=300,
+                                            torch_dtype="float32")).float()
+    model.eval()
+
+    entity_types = ["人名", "地名"]
+    text = "李明今天去了北京。"
+    input_ids = torch.tensor([tokenizer.encode(text)])
+    prompts = [tokenizer.encode(t) for t in entity_types]
+    L = max(len(p) for p in prompts)
+    label_prompt_ids = torch.tensor(
+        [p + [tokenizer.pad_token_id] * (L - len(p)) for p in prompts])
+
+    results = model.extract(input_ids, label_prompt_ids, threshold=0.0,
+                            max_spans=4)
+    for span in results[0]:
+        s, e = span["span"]
+        print(f"type={entity_types[span['type']]} span={text[s - 1:e]} "
+              f"score={span['score']:.3f}")
+
+
+if __name__ == "__main__":
+    main()
