@@ -58,3 +58,24 @@ class StepProfiler(Callback):
             print(table.splitlines()[0:5], flush=True)
             self._prof = None
             self._done = True
+
+
+class RocTXMarker(Callback):
+    """Per-step rocTX ranges (torch.cuda.nvtx maps to rocTX on ROCm).
+
+    Makes steps visible in `rocprofv3 --marker-trace` timelines so kernel
+    traces can be cut per step; no-op overhead when no profiler attached.
+    """
+
+    def __init__(self, prefix: str = "fengshen_step"):
+        self.prefix = prefix
+        self._open = False
+
+    def on_train_batch_start(self, trainer, module, batch, batch_idx):
+        torch.cuda.nvtx.range_push(f"{self.prefix}_{trainer.global_step}")
+        self._open = True
+
+    def on_train_batch_end(self, trainer, module, outputs, batch, batch_idx):
+        if self._open:
+            torch.cuda.nvtx.range_pop()
+            self._open = False

@@ -198,3 +198,16 @@ def test_step_profiler_callback(tmp_path):
                       default_root_dir=str(tmp_path))
     trainer.fit(model, train_dataloaders=loader)
     assert (tmp_path / "p" / "step_profile.txt").exists()
+
+
+def test_roctx_marker_callback(tmp_path):
+    from fengshen_amd.utils.profiling import RocTXMarker
+    args = _make_args()
+    torch.manual_seed(0)
+    model = ToyModule(args)
+    loader = DataLoader(ToyDataset(n=64), batch_size=16)
+    marker = RocTXMarker()
+    trainer = Trainer(max_steps=3, precision="fp32", callbacks=[marker],
+                      default_root_dir=str(tmp_path))
+    trainer.fit(model, train_dataloaders=loader)
+    assert not marker._open  # every pushed range was popped
