@@ -69,18 +69,26 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    assert torch.cuda.is_available(), "bench.py requires a GPU"
-    torch.cuda.set_device(local_rank)
-    device = torch.device("cuda", local_rank)
+    # FENGSHEN_BENCH_CPU=1: gloo/CPU smoke of the multi-rank path only —
+    # numbers from this mode are NOT benchmark results.
+    cpu_smoke = os.environ.get("FENGSHEN_BENCH_CPU") == "1"
+    if cpu_smoke:
+        device = torch.device("cpu")
+    else:
+        assert torch.cuda.is_available(), "bench.py requires a GPU"
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
 
     import torch.distributed as dist
     from fengshen_amd.parallel.groups import init_distributed
     from fengshen_amd.parallel.zero import ZeroOptimizer
-    from fengshen_amd.ops import has_ext
-    assert has_ext(), "HIP extension must be built (python -m fengshen_amd.ops.build)"
+    if not cpu_smoke:
+        from fengshen_amd.ops import has_ext
+        assert has_ext(), \
+            "HIP extension must be built (python -m fengshen_amd.ops.build)"
 
     if world > 1 or args.tensor_model_parallel_size > 1:
-        init_distributed(backend="nccl")
+        init_distributed(backend="gloo" if cpu_smoke else "nccl")
     if args.tensor_model_parallel_size > 1:
         from fengshen_amd.parallel.groups import initialize_model_parallel
         initialize_model_parallel(args.tensor_model_parallel_size)
@@ -144,11 +152,13 @@ def main():
 
     if world > 1:
         dist.barrier()
-    torch.cuda.synchronize()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         loss = step()
-    torch.cuda.synchronize()
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
     if world > 1:
         dist.barrier()
     elapsed = time.perf_counter() - t0
