@@ -84,7 +84,8 @@ def configure_optimizers(pl_model, model_params=None):
             betas=betas, eps=eps, weight_decay=args.weight_decay,
             process_group=pgroups.get_data_parallel_group(),
             bucket_numel=strategy.bucket_numel,
-            overlap_comm=strategy.overlap_comm)
+            overlap_comm=strategy.overlap_comm,
+            cpu_offload=getattr(strategy, "cpu_offload", False))
     else:
         from fengshen_amd.ops.adamw import FusedAdamW
         optimizer = FusedAdamW(optim_groups, lr=args.learning_rate,

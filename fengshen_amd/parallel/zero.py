@@ -90,11 +90,17 @@ class ZeroOptimizer(torch.optim.Optimizer):
                  betas=(0.9, 0.999), eps: float = 1e-8, weight_decay: float = 0.01,
                  process_group=None, bucket_numel: int = 128 * 1024 * 1024,
                  overlap_comm: bool = True, grad_dtype: Optional[torch.dtype] = None,
-                 clip_grad: float = 0.0):
+                 clip_grad: float = 0.0, cpu_offload: bool = False):
         defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
         super().__init__(params, defaults)
         assert stage in (0, 1, 2), "use Zero3Model for stage 3"
         self.stage = stage
+        # ZeRO-offload parity (reference: DeepSpeedCPUAdam + offload_optimizer,
+        # strategies/megatron_deepspeed.py:124-167): fp32 master + Adam moments
+        # live in pinned host memory; the step streams the grad shard to the
+        # host, updates there, and streams the new weights back.  Saves
+        # 12 bytes/param of HBM per shard at the cost of 2 PCIe transits.
+        self.cpu_offload = cpu_offload
         self.group = process_group
         if self.group is None and dist.is_available() and dist.is_initialized():
             from fengshen_amd.parallel import groups as pgroups
@@ -156,9 +162,22 @@ class ZeroOptimizer(torch.optim.Optimizer):
             src = b.flat_param[start:start + b.shard_numel]
         else:
             src = b.flat_param
-        b.master_shard = src.detach().to(torch.float32).clone()
-        b.exp_avg = torch.zeros_like(b.master_shard)
-        b.exp_avg_sq = torch.zeros_like(b.master_shard)
+        if self.cpu_offload:
+            pin = src.is_cuda
+            b.master_shard = src.detach().to(
+                torch.float32).cpu().clone()
+            if pin:
+                b.master_shard = b.master_shard.pin_memory()
+            b.exp_avg = torch.zeros_like(b.master_shard)
+            b.exp_avg_sq = torch.zeros_like(b.master_shard)
+            b.cpu_grad = torch.zeros(b.master_shard.numel(),
+                                     dtype=b.flat_grad.dtype)
+            if pin:
+                b.cpu_grad = b.cpu_grad.pin_memory()
+        else:
+            b.master_shard = src.detach().to(torch.float32).clone()
+            b.exp_avg = torch.zeros_like(b.master_shard)
+            b.exp_avg_sq = torch.zeros_like(b.master_shard)
 
     def _register_hooks(self):
         for b in self.buckets:
@@ -285,12 +304,22 @@ class ZeroOptimizer(torch.optim.Optimizer):
                 out_param = b.flat_param[start:start + b.shard_numel]
             else:
                 out_param = b.flat_param
-            fused_adamw_flat_(
-                b.master_shard, grad, b.exp_avg, b.exp_avg_sq,
-                out_param,
-                lr=g["lr"], beta1=beta1, beta2=beta2, eps=g["eps"],
-                weight_decay=g["weight_decay"], step=self._step_count,
-            )
+            if self.cpu_offload:
+                b.cpu_grad.copy_(grad)
+                fused_adamw_flat_(
+                    b.master_shard, b.cpu_grad, b.exp_avg, b.exp_avg_sq,
+                    None,
+                    lr=g["lr"], beta1=beta1, beta2=beta2, eps=g["eps"],
+                    weight_decay=g["weight_decay"], step=self._step_count,
+                )
+                out_param.copy_(b.master_shard, non_blocking=True)
+            else:
+                fused_adamw_flat_(
+                    b.master_shard, grad, b.exp_avg, b.exp_avg_sq,
+                    out_param,
+                    lr=g["lr"], beta1=beta1, beta2=beta2, eps=g["eps"],
+                    weight_decay=g["weight_decay"], step=self._step_count,
+                )
             if sharded:
                 shard = out_param.contiguous()
                 if dist.get_backend(self.group) == "nccl":

@@ -34,7 +34,11 @@ def parse_strategy(name: str) -> dict:
     if name in ("auto", "ddp", "single", "ddp_sharded"):
         return {"kind": name if name != "ddp_sharded" else "zero2", "stage": 0}
     if name.startswith("zero"):
-        return {"kind": "zero", "stage": int(name[4:])}
+        # 'zero2_offload' => ZeRO-offload (optimizer states in host RAM,
+        # reference parity: deepspeed offload_optimizer config)
+        offload = name.endswith("_offload")
+        stage = int(name[4:].split("_")[0])
+        return {"kind": "zero", "stage": stage, "cpu_offload": offload}
     if name.startswith("deepspeed_stage_"):  # reference-compat alias
         return {"kind": "zero", "stage": int(name.rsplit("_", 1)[1])}
     raise ValueError(f"unknown strategy {name!r}")
@@ -48,9 +52,11 @@ class Strategy:
                  pipe_model_parallel_size: int = 1,
                  mpu_seed: int = 42,
                  bucket_numel: int = 128 * 1024 * 1024,
-                 overlap_comm: bool = True):
+                 overlap_comm: bool = True,
+                 cpu_offload: bool = False):
         self.kind = kind
         self.stage = stage
+        self.cpu_offload = cpu_offload
         self.tp_size = tensor_model_parallel_size
         self.pp_size = pipe_model_parallel_size
         self.mpu_seed = mpu_seed
@@ -118,6 +124,7 @@ class Strategy:
                 process_group=pgroups.get_data_parallel_group(),
                 bucket_numel=self.bucket_numel,
                 overlap_comm=self.overlap_comm,
+                cpu_offload=self.cpu_offload,
             )
             self.zero_optimizer = zopt
             if scheduler_cfg is not None:

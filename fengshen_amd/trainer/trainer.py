@@ -100,7 +100,8 @@ class Trainer:
             tensor_model_parallel_size=tensor_model_parallel_size,
             pipe_model_parallel_size=pipe_model_parallel_size,
             mpu_seed=mpu_seed, bucket_numel=zero_bucket_numel,
-            overlap_comm=overlap_comm)
+            overlap_comm=overlap_comm,
+            cpu_offload=s.get("cpu_offload", False))
         self.max_steps = max_steps
         self.max_epochs = max_epochs
         self.precision = precision
