@@ -60,3 +60,22 @@ def test_dataset_collater_feeds_model():
     out = m(batch["source"], padding_mask=batch["padding_mask"],
             labels=batch["labels"])
     assert out.loss.isfinite()
+
+
+def test_hubert_save_load_roundtrip(tmp_path):
+    torch.manual_seed(0)
+    cfg = hubert_tiny_config(torch_dtype="float32")
+    m = HubertForPreTraining(cfg).float()
+    m.save_pretrained(tmp_path / "hubert")
+    m2 = HubertForPreTraining.from_pretrained(tmp_path / "hubert").float()
+    m.eval()
+    m2.eval()  # dropout off for a deterministic comparison
+    src = torch.randn(1, 3200)
+    mi = torch.zeros(1, m.hubert.frame_lengths(torch.tensor([3200]))[0],
+                     dtype=torch.bool)
+    mi[:, :4] = True
+    with torch.no_grad():
+        a = m.hubert(src, apply_mask=True, mask_time_indices=mi)
+        b = m2.hubert(src, apply_mask=True, mask_time_indices=mi)
+    assert torch.allclose(a.last_hidden_state, b.last_hidden_state,
+                          atol=1e-6)
