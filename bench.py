@@ -31,6 +31,9 @@ def parse_args():
     p.add_argument("--zero_stage", type=int, default=3)
     p.add_argument("--tensor_model_parallel_size", "--tp", type=int, default=1)
     p.add_argument("--lr", type=float, default=1e-5)
+    p.add_argument("--ckpt_skip", type=int, default=0,
+                   help="selective act-ckpt: every k-th layer skips "
+                        "recompute (0 = checkpoint all layers)")
     return p.parse_args()
 
 
@@ -102,7 +105,10 @@ def main():
     torch.set_default_dtype(torch.float32)
     model = model.to(torch.bfloat16).to(device)
     if hasattr(model, "gradient_checkpointing_enable"):
-        model.gradient_checkpointing_enable()
+        try:
+            model.gradient_checkpointing_enable(skip_interval=args.ckpt_skip)
+        except TypeError:
+            model.gradient_checkpointing_enable()
     model.train()
 
     if args.zero_stage == 3:

@@ -70,6 +70,9 @@ class LlamaModel(LlamaPreTrainedModel):
             for i in range(config.num_hidden_layers)])
         self.norm = RMSNorm(config.hidden_size, eps=config.rms_norm_epsilon)
         self.gradient_checkpointing = False
+        # selective activation ckpt: every k-th layer keeps activations
+        # (no recompute) — spends spare HBM3E to skip recompute FLOPs
+        self.gradient_checkpointing_skip_interval = 0
         self.post_init()
 
     def get_input_embeddings(self):
@@ -100,8 +103,13 @@ class LlamaModel(LlamaPreTrainedModel):
             from transformers.cache_utils import DynamicCache
             cache = DynamicCache()
 
-        for layer in self.layers:
-            if self.gradient_checkpointing and self.training and cache is None:
+        skip = self.gradient_checkpointing_skip_interval
+        for i, layer in enumerate(self.layers):
+            ckpt = self.gradient_checkpointing and self.training \
+                and cache is None
+            if ckpt and skip and (i % skip == skip - 1):
+                ckpt = False  # selective: keep this layer's activations
+            if ckpt:
                 h = activation_checkpoint(
                     lambda x, m, lyr=layer: lyr(x, attention_mask=m), h, mask)
             else:
@@ -126,8 +134,9 @@ class LlamaForCausalLM(LlamaPreTrainedModel, GenerationMixin):
     def get_output_embeddings(self):
         return self.lm_head
 
-    def gradient_checkpointing_enable(self, **_kw):
+    def gradient_checkpointing_enable(self, skip_interval: int = 0, **_kw):
         self.model.gradient_checkpointing = True
+        self.model.gradient_checkpointing_skip_interval = skip_interval
 
     def forward(self, input_ids: torch.Tensor,
                 attention_mask: Optional[torch.Tensor] = None,

@@ -134,3 +134,30 @@ def test_tp_shard_merge_roundtrip():
     merged = merge_state_dicts(m, shards)
     for k in full:
         assert torch.equal(full[k], merged[k]), k
+
+
+def test_selective_activation_checkpoint_grads_match():
+    """skip_interval recompute-skipping must not change gradients."""
+    from fengshen_amd.models.llama.configuration_llama import LlamaConfig
+    from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+    cfg = LlamaConfig(vocab_size=128, hidden_size=64, num_hidden_layers=4,
+                      num_attention_heads=4, intermediate_size=128,
+                      max_position_embeddings=64, torch_dtype="float32")
+    torch.manual_seed(0)
+    m1 = LlamaForCausalLM(cfg).float()
+    m2 = LlamaForCausalLM(cfg).float()
+    m2.load_state_dict(m1.state_dict())
+    m1.gradient_checkpointing_enable()
+    m2.gradient_checkpointing_enable(skip_interval=2)
+    m1.train()
+    m2.train()
+    ids = torch.randint(0, 128, (2, 32))
+    torch.manual_seed(1)
+    l1 = m1(ids, labels=ids).loss
+    l1.backward()
+    torch.manual_seed(1)
+    l2 = m2(ids, labels=ids).loss
+    l2.backward()
+    assert torch.allclose(l1, l2)
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1.grad, p2.grad, atol=1e-6)
