@@ -15,6 +15,11 @@ import json
 import os
 import time
 
+# expandable segments avoid fragmentation OOM with selective activation
+# checkpointing (large transient recompute allocations); must be set
+# before the first allocation.
+os.environ.setdefault("PYTORCH_HIP_ALLOC_CONF", "expandable_segments:True")
+
 import torch
 
 
@@ -31,9 +36,10 @@ def parse_args():
     p.add_argument("--zero_stage", type=int, default=3)
     p.add_argument("--tensor_model_parallel_size", "--tp", type=int, default=1)
     p.add_argument("--lr", type=float, default=1e-5)
-    p.add_argument("--ckpt_skip", type=int, default=0,
+    p.add_argument("--ckpt_skip", type=int, default=-1,
                    help="selective act-ckpt: every k-th layer skips "
-                        "recompute (0 = checkpoint all layers)")
+                        "recompute (0 = checkpoint all layers, -1 = auto "
+                        "from free HBM after optimizer init)")
     return p.parse_args()
 
 
@@ -123,6 +129,35 @@ def main():
     from fengshen_amd.parallel import groups as pgroups
     dp_rank = pgroups.get_data_parallel_rank()
     dp_world = pgroups.get_data_parallel_world_size()
+
+    # Selective activation checkpointing: spend spare HBM3E to skip
+    # recompute on every k-th layer (measured +5.4% at skip=4 on 1 GPU;
+    # with dp-sharded optimizer states at N=8 far more memory is free, so
+    # size k from what is actually available AFTER optimizer-state alloc).
+    skip = max(args.ckpt_skip, 0)
+    if args.ckpt_skip < 0 and torch.cuda.is_available() \
+            and args.model in ("ziya-llama-13b", "llama-tiny"):
+        free_b, _total = torch.cuda.mem_get_info()
+        L = model.config.num_hidden_layers
+        # kept-activation bytes per non-recomputed layer (bf16):
+        # ~21.2 hidden-sized tensors of [b, s] rows survive in the graph
+        act = 21.2 * model.config.hidden_size * args.micro_batch \
+            * args.seq_len * 2
+        n_store = int(free_b * 0.8 / act)
+        if n_store >= L:
+            skip = 1          # store everything: no recompute at all
+        elif n_store >= 2:
+            skip = max(2, -(-L // n_store))
+        else:
+            skip = 0
+        if skip and pgroups.get_data_parallel_world_size() == 1:
+            skip = max(skip, 4)  # skip<4 validated OOM at dp1 (skip3/skip2)
+    if skip:
+        try:
+            model.gradient_checkpointing_enable(skip_interval=skip)
+        except TypeError:
+            skip = 0
+
     is_bert = args.model == "erlangshen-1.3b"
     b, s = args.micro_batch, args.seq_len
     # TP ranks must see identical batches: seed by DP rank
@@ -209,6 +244,7 @@ def main():
                     + (f"_tp{args.tensor_model_parallel_size}"
                        if args.tensor_model_parallel_size > 1 else "")),
                 "activation_checkpointing": True,
+                "ckpt_skip_interval": skip,
             },
         }), flush=True)
 
