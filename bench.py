@@ -134,15 +134,16 @@ def main():
     # recompute on every k-th layer (measured +5.4% at skip=4 on 1 GPU;
     # with dp-sharded optimizer states at N=8 far more memory is free, so
     # size k from what is actually available AFTER optimizer-state alloc).
+    # kept-activation bytes per non-recomputed layer, in units of
+    # hidden * b * s * 2 (bf16): ~21.2 for SwiGLU llama, ~21 gpt2, ~19 bert
+    _ACT_MULT = {"ziya-llama-13b": 21.2, "llama-tiny": 21.2,
+                 "wenzhong-gpt2-3.5b": 21.0, "erlangshen-1.3b": 19.0}
     skip = max(args.ckpt_skip, 0)
-    if args.ckpt_skip < 0 and torch.cuda.is_available() \
-            and args.model in ("ziya-llama-13b", "llama-tiny"):
+    if args.ckpt_skip < 0 and torch.cuda.is_available():
         free_b, _total = torch.cuda.mem_get_info()
         L = model.config.num_hidden_layers
-        # kept-activation bytes per non-recomputed layer (bf16):
-        # ~21.2 hidden-sized tensors of [b, s] rows survive in the graph
-        act = 21.2 * model.config.hidden_size * args.micro_batch \
-            * args.seq_len * 2
+        act = _ACT_MULT[args.model] * model.config.hidden_size \
+            * args.micro_batch * args.seq_len * 2
         n_store = int(free_b * 0.8 / act)
         if n_store >= L:
             skip = 1          # store everything: no recompute at all
@@ -150,8 +151,9 @@ def main():
             skip = max(2, -(-L // n_store))
         else:
             skip = 0
-        if skip and pgroups.get_data_parallel_world_size() == 1:
-            skip = max(skip, 4)  # skip<4 validated OOM at dp1 (skip3/skip2)
+        if skip > 1 and args.model == "ziya-llama-13b" \
+                and pgroups.get_data_parallel_world_size() == 1:
+            skip = max(skip, 4)  # skip<4 validated OOM at 13B dp1
     if skip:
         try:
             model.gradient_checkpointing_enable(skip_interval=skip)

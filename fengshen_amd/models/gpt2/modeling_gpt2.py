@@ -62,6 +62,7 @@ class GPT2Model(GPT2PreTrainedModel):
             for i in range(config.num_hidden_layers)])
         self.ln_f = LayerNorm(config.hidden_size, eps=config.layer_norm_epsilon)
         self.gradient_checkpointing = False
+        self.gradient_checkpointing_skip_interval = 0
         self.post_init()
 
     def get_input_embeddings(self):
@@ -91,8 +92,13 @@ class GPT2Model(GPT2PreTrainedModel):
         elif attention_mask is not None:
             mask = attention_mask
 
-        for layer in self.h:
-            if self.gradient_checkpointing and self.training and cache is None:
+        skip = self.gradient_checkpointing_skip_interval
+        for i, layer in enumerate(self.h):
+            ckpt = self.gradient_checkpointing and self.training \
+                and cache is None
+            if ckpt and skip and (i % skip == skip - 1):
+                ckpt = False
+            if ckpt:
                 h = activation_checkpoint(
                     lambda x, m, lyr=layer: lyr(x, attention_mask=m), h, mask)
             else:
@@ -113,8 +119,9 @@ class GPT2LMHeadModel(GPT2PreTrainedModel, GenerationMixin):
     def get_output_embeddings(self):
         return None  # tied head applied via parallel_lm_logits
 
-    def gradient_checkpointing_enable(self, **_kw):
+    def gradient_checkpointing_enable(self, skip_interval: int = 0, **_kw):
         self.transformer.gradient_checkpointing = True
+        self.transformer.gradient_checkpointing_skip_interval = skip_interval
 
     def forward(self, input_ids, attention_mask=None, labels=None,
                 past_key_values=None, use_cache: bool = False,

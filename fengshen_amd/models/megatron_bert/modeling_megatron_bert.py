@@ -98,6 +98,9 @@ class MegatronBertModel(MegatronBertPreTrainedModel):
         self.pooler = nn.Linear(config.hidden_size, config.hidden_size) \
             if add_pooling_layer else None
         self.gradient_checkpointing = False
+        # selective activation ckpt (cf. LlamaModel): every k-th layer
+        # keeps activations instead of recomputing
+        self.gradient_checkpointing_skip_interval = 0
         self.post_init()
 
     def get_input_embeddings(self):
@@ -113,8 +116,12 @@ class MegatronBertModel(MegatronBertPreTrainedModel):
         if attention_mask is not None:
             # HF 1=keep [b,s] -> internal True=masked [b,1,1,s]
             mask = (attention_mask == 0)[:, None, None, :]
-        for layer in self.encoder:
-            if self.gradient_checkpointing and self.training:
+        skip = self.gradient_checkpointing_skip_interval
+        for i, layer in enumerate(self.encoder):
+            ckpt = self.gradient_checkpointing and self.training
+            if ckpt and skip and (i % skip == skip - 1):
+                ckpt = False
+            if ckpt:
                 h = activation_checkpoint(
                     lambda x, m, lyr=layer: lyr(x, attention_mask=m), h, mask)
             else:
@@ -175,8 +182,9 @@ class MegatronBertForPreTraining(MegatronBertPreTrainedModel):
     def get_input_embeddings(self):
         return self.bert.embeddings.word_embeddings
 
-    def gradient_checkpointing_enable(self, **_kw):
+    def gradient_checkpointing_enable(self, skip_interval: int = 0, **_kw):
         self.bert.gradient_checkpointing = True
+        self.bert.gradient_checkpointing_skip_interval = skip_interval
 
     def forward(self, input_ids, attention_mask=None, token_type_ids=None,
                 labels=None, next_sentence_label=None, position_ids=None,
@@ -212,8 +220,9 @@ class MegatronBertForMaskedLM(MegatronBertPreTrainedModel):
         self.cls = MegatronBertLMHead(config)
         self.post_init()
 
-    def gradient_checkpointing_enable(self, **_kw):
+    def gradient_checkpointing_enable(self, skip_interval: int = 0, **_kw):
         self.bert.gradient_checkpointing = True
+        self.bert.gradient_checkpointing_skip_interval = skip_interval
 
     def forward(self, input_ids, attention_mask=None, token_type_ids=None,
                 labels=None, return_dict=True, **_kw):
