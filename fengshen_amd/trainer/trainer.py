@@ -90,6 +90,7 @@ class Trainer:
         seed: int = 1234,
         num_sanity_val_steps: int = 0,
         activation_checkpointing: bool = False,
+        ckpt_skip_interval: int = 0,
         zero_bucket_numel: int = 128 * 1024 * 1024,
         overlap_comm: bool = True,
         **_unused,
@@ -118,6 +119,7 @@ class Trainer:
         self.seed_offset = 0
         self.num_sanity_val_steps = num_sanity_val_steps
         self.activation_checkpointing = activation_checkpointing
+        self.ckpt_skip_interval = ckpt_skip_interval
 
         # loop state
         self.global_step = 0
@@ -192,6 +194,9 @@ class Trainer:
         p.add_argument("--seed", type=int, default=1234)
         p.add_argument("--num_sanity_val_steps", type=int, default=0)
         p.add_argument("--activation_checkpointing", action="store_true", default=False)
+        p.add_argument("--ckpt_skip_interval", type=int, default=0,
+                       help="selective act-ckpt: every k-th layer keeps "
+                            "activations (0 = checkpoint all layers)")
         p.add_argument("--zero_bucket_numel", type=int, default=128 * 1024 * 1024)
         return parent_parser
 
@@ -252,6 +257,17 @@ class Trainer:
         if datamodule is not None and hasattr(datamodule, "setup"):
             datamodule.setup("fit")
         model.setup("fit")
+
+        if self.activation_checkpointing:
+            # enable on every submodule that supports it (apps usually hold
+            # the HF model at module.model / module.bert / ...)
+            for m in model.modules():
+                fn = getattr(m, "gradient_checkpointing_enable", None)
+                if fn is not None and m is not model:
+                    try:
+                        fn(skip_interval=self.ckpt_skip_interval)
+                    except TypeError:
+                        fn()
 
         dev = self.device
         wrapped = self.strategy.setup_model(model, dev, self.precision)
