@@ -376,7 +376,7 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
     __syncthreads();
     {
       const int tid = threadIdx.x;
-      for (int i = tid * 8; i < FB_T * FA_D; i += 256 * 8) {
+      for (int i = tid * 8; i < FB_T * FA_D; i += FA_WAVES * 64 * 8) {
         const int kr = i / FA_D;
         const int kc = i % FA_D;
         bf16x8 kk = *reinterpret_cast<const bf16x8*>(
@@ -533,7 +533,7 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
     __syncthreads();
     {
       const int tid = threadIdx.x;
-      for (int i = tid * 8; i < FB_T * FA_D; i += 256 * 8) {
+      for (int i = tid * 8; i < FB_T * FA_D; i += FA_WAVES * 64 * 8) {
         const int qr = i / FA_D;
         const int qc = i % FA_D;
         bf16x8 qq = *reinterpret_cast<const bf16x8*>(
