@@ -82,6 +82,8 @@ class VisionTransformer(nn.Module):
         self.ln_post = LayerNorm(hidden)
 
     def forward(self, pixel_values: torch.Tensor) -> torch.Tensor:
+        # accept fp32 images regardless of model dtype
+        pixel_values = pixel_values.to(self.patch_embed.weight.dtype)
         x = self.patch_embed(pixel_values)  # [b, h, gh, gw]
         x = x.flatten(2).transpose(1, 2)
         cls = self.cls_token.expand(x.shape[0], -1, -1).to(x.dtype)
