@@ -13,7 +13,12 @@ from transformers.generation import GenerationMixin
 from transformers.modeling_outputs import Seq2SeqLMOutput
 
 from fengshen_amd.models.encoder_decoder import DecoderLayer, EncoderLayer
-from fengshen_amd.models.layers import LayerNorm, init_normal, parallel_lm_logits, scaled_init_normal
+from fengshen_amd.models.layers import (
+    LayerNorm,
+    init_normal,
+    parallel_lm_logits,
+    scaled_init_normal,
+)
 from fengshen_amd.parallel import groups
 from fengshen_amd.parallel.cross_entropy import vocab_parallel_cross_entropy
 from fengshen_amd.parallel.layers import VocabParallelEmbedding

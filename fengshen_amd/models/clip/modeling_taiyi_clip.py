@@ -16,7 +16,12 @@ import torch.nn as nn
 from transformers import PretrainedConfig, PreTrainedModel
 from transformers.utils import ModelOutput
 
-from fengshen_amd.models.layers import LayerNorm, ParallelTransformerLayer, init_normal, scaled_init_normal
+from fengshen_amd.models.layers import (
+    LayerNorm,
+    ParallelTransformerLayer,
+    init_normal,
+    scaled_init_normal,
+)
 from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
     MegatronBertConfig,
 )
