@@ -18,7 +18,9 @@ import torch.nn as nn
 from transformers import PretrainedConfig, PreTrainedModel
 from transformers.modeling_outputs import (
     BaseModelOutput,
+    QuestionAnsweringModelOutput,
     SequenceClassifierOutput,
+    TokenClassifierOutput,
 )
 
 from fengshen_amd.models.layers import (
@@ -197,3 +199,58 @@ class ZenForSequenceClassification(ZenPreTrainedModel):
             loss = nn.functional.cross_entropy(
                 logits.float().view(-1, self.num_labels), labels.view(-1))
         return SequenceClassifierOutput(loss=loss, logits=logits)
+
+
+class ZenForTokenClassification(ZenPreTrainedModel):
+    """Token-level head (the zen2_finetune NER tasks; ref
+    modeling.py token classification head)."""
+
+    def __init__(self, config):
+        super().__init__(config)
+        self.num_labels = getattr(config, "num_labels", 2)
+        self.zen = ZenModel(config)
+        self.dropout = nn.Dropout(config.hidden_dropout)
+        self.classifier = nn.Linear(config.hidden_size, self.num_labels)
+        self.post_init()
+
+    def forward(self, input_ids, ngram_ids=None, ngram_position_matrix=None,
+                attention_mask=None, token_type_ids=None, labels=None, **_kw):
+        h = self.zen(input_ids, ngram_ids, ngram_position_matrix,
+                     attention_mask, token_type_ids).last_hidden_state
+        logits = self.classifier(self.dropout(h))
+        loss = None
+        if labels is not None:
+            loss = nn.functional.cross_entropy(
+                logits.float().view(-1, self.num_labels), labels.view(-1),
+                ignore_index=-100)
+        return TokenClassifierOutput(loss=loss, logits=logits)
+
+
+class ZenForQuestionAnswering(ZenPreTrainedModel):
+    """Span QA head — ZEN2's addition (ref zen2/modeling.py:1291)."""
+
+    def __init__(self, config):
+        super().__init__(config)
+        self.zen = ZenModel(config)
+        self.qa_outputs = nn.Linear(config.hidden_size, 2)
+        self.post_init()
+
+    def forward(self, input_ids, ngram_ids=None, ngram_position_matrix=None,
+                attention_mask=None, token_type_ids=None,
+                start_positions=None, end_positions=None, **_kw):
+        h = self.zen(input_ids, ngram_ids, ngram_position_matrix,
+                     attention_mask, token_type_ids).last_hidden_state
+        start_logits, end_logits = self.qa_outputs(h).split(1, dim=-1)
+        start_logits = start_logits.squeeze(-1)
+        end_logits = end_logits.squeeze(-1)
+        loss = None
+        if start_positions is not None and end_positions is not None:
+            s_loss = nn.functional.cross_entropy(
+                start_logits.float(), start_positions.clamp(
+                    0, start_logits.shape[1] - 1))
+            e_loss = nn.functional.cross_entropy(
+                end_logits.float(), end_positions.clamp(
+                    0, end_logits.shape[1] - 1))
+            loss = (s_loss + e_loss) / 2
+        return QuestionAnsweringModelOutput(
+            loss=loss, start_logits=start_logits, end_logits=end_logits)

@@ -179,3 +179,26 @@ def test_ner_metrics():
     labels[1, 2] = 5
     acc = metrics_mlm_acc(logits, labels)
     assert abs(acc.item() - 0.5) < 1e-6
+
+
+def test_zen_task_heads():
+    """ZEN2's token-classification and span-QA heads (ref zen2/modeling.py)."""
+    from fengshen_amd.models.zen.modeling_zen import (
+        ZenForQuestionAnswering,
+        ZenForTokenClassification,
+        zen_tiny_config,
+    )
+    torch.manual_seed(0)
+    cfg = zen_tiny_config(torch_dtype="float32", num_labels=5)
+    m = ZenForTokenClassification(cfg).float()
+    ids = torch.randint(0, 256, (2, 16))
+    lab = torch.randint(0, 5, (2, 16))
+    out = m(ids, labels=lab)
+    assert out.logits.shape == (2, 16, 5)
+    out.loss.backward()
+
+    q = ZenForQuestionAnswering(zen_tiny_config(torch_dtype="float32")).float()
+    out = q(ids, start_positions=torch.tensor([1, 2]),
+            end_positions=torch.tensor([3, 4]))
+    assert out.start_logits.shape == (2, 16)
+    assert out.loss.isfinite()
