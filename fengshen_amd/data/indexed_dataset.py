@@ -163,3 +163,131 @@ def make_builder(out_file: str, vocab_size: Optional[int] = None):
 def make_dataset(path: str, skip_warmup: bool = True) -> MMapIndexedDataset:
     assert MMapIndexedDataset.exists(path), f"no .bin/.idx at {path}"
     return MMapIndexedDataset(path, skip_warmup)
+
+
+# ----------------------------------------------------------------------
+# Legacy (non-mmap) fairseq format: TNTIDX header, explicit offset arrays
+# (reference indexed_dataset.py:130-216 IndexedDataset /
+#  :217 IndexedCachedDataset).  Kept for loading old preprocessed corpora.
+
+_LEGACY_MAGIC = b"TNTIDX\x00\x00"
+
+
+class IndexedDataset(torch.utils.data.Dataset):
+    """Reader for the legacy TNTIDX .bin/.idx pair (file reads, no mmap)."""
+
+    def __init__(self, path: str):
+        self.path = path
+        self.data_file = None
+        with open(index_file_path(path), "rb") as f:
+            magic = f.read(8)
+            assert magic == _LEGACY_MAGIC, "not a legacy TNTIDX index"
+            (version,) = struct.unpack("<Q", f.read(8))
+            assert version == 1
+            code, self.element_size = struct.unpack("<QQ", f.read(16))
+            self.dtype = np.dtype(_DTYPES[code])
+            self._len, s = struct.unpack("<QQ", f.read(16))
+            (doc_count,) = struct.unpack("<Q", f.read(8))
+            self.dim_offsets = np.fromfile(f, dtype=np.int64,
+                                           count=self._len + 1)
+            self.data_offsets = np.fromfile(f, dtype=np.int64,
+                                            count=self._len + 1)
+            self.sizes = np.fromfile(f, dtype=np.int64, count=s)
+            self.doc_idx = np.fromfile(f, dtype=np.int64, count=doc_count)
+
+    def __len__(self):
+        return self._len
+
+    def _read(self, idx: int) -> np.ndarray:
+        if self.data_file is None:
+            self.data_file = open(data_file_path(self.path), "rb",
+                                  buffering=0)
+        size = int(self.data_offsets[idx + 1] - self.data_offsets[idx])
+        a = np.empty(size, dtype=self.dtype)
+        self.data_file.seek(int(self.data_offsets[idx]) * self.element_size)
+        self.data_file.readinto(a)
+        return a
+
+    def __getitem__(self, idx: int) -> np.ndarray:
+        if idx < 0 or idx >= self._len:
+            raise IndexError("index out of range")
+        return self._read(idx)
+
+    def __del__(self):
+        if self.data_file:
+            self.data_file.close()
+
+    @staticmethod
+    def exists(path: str) -> bool:
+        if not os.path.exists(index_file_path(path)):
+            return False
+        with open(index_file_path(path), "rb") as f:
+            return f.read(8) == _LEGACY_MAGIC
+
+
+class IndexedCachedDataset(IndexedDataset):
+    """Legacy reader + in-RAM prefetch cache (reference :217-270)."""
+
+    def __init__(self, path: str):
+        super().__init__(path)
+        self.cache = {}
+
+    def prefetch(self, indices):
+        for i in sorted(set(indices)):
+            self.cache[i] = super()._read(i)
+
+    def __getitem__(self, idx: int) -> np.ndarray:
+        if idx in self.cache:
+            return self.cache[idx]
+        return super().__getitem__(idx)
+
+
+class IndexedDatasetBuilder:
+    """Writer for the legacy TNTIDX format (reference :271-343)."""
+
+    def __init__(self, out_file: str, dtype=np.int32):
+        self.out_file = open(out_file, "wb")
+        self.dtype = np.dtype(dtype)
+        self.element_size = self.dtype.itemsize
+        self.data_offsets = [0]
+        self.dim_offsets = [0]
+        self.sizes: List[int] = []
+        self.doc_idx = [0]
+
+    def add_item(self, tensor_or_array):
+        a = np.asarray(tensor_or_array, dtype=self.dtype)
+        self.out_file.write(a.tobytes(order="C"))
+        self.data_offsets.append(self.data_offsets[-1] + a.size)
+        self.sizes.append(a.size)
+        self.dim_offsets.append(self.dim_offsets[-1] + 1)
+
+    def end_document(self):
+        self.doc_idx.append(len(self.sizes))
+
+    def finalize(self, index_file: str):
+        self.out_file.close()
+        with open(index_file, "wb") as f:
+            f.write(_LEGACY_MAGIC)
+            f.write(struct.pack("<Q", 1))
+            f.write(struct.pack("<QQ", _DTYPE_CODES[self.dtype],
+                                self.element_size))
+            f.write(struct.pack("<QQ", len(self.data_offsets) - 1,
+                                len(self.sizes)))
+            f.write(struct.pack("<Q", len(self.doc_idx)))
+            np.asarray(self.dim_offsets, dtype=np.int64).tofile(f)
+            np.asarray(self.data_offsets, dtype=np.int64).tofile(f)
+            np.asarray(self.sizes, dtype=np.int64).tofile(f)
+            np.asarray(self.doc_idx, dtype=np.int64).tofile(f)
+
+
+def infer_dataset_impl(path: str) -> Optional[str]:
+    """'mmap' | 'cached' | None by magic (reference :33-46)."""
+    if not os.path.exists(index_file_path(path)):
+        return None
+    with open(index_file_path(path), "rb") as f:
+        magic = f.read(9)
+    if magic[:8] == _LEGACY_MAGIC:
+        return "cached"
+    if magic == _INDEX_HEADER:
+        return "mmap"
+    return None

@@ -248,3 +248,26 @@ def test_bart_mmap_dataset(tmp_path):
     assert torch.equal(item["input_ids"], bart_ds[0]["input_ids"])
     # different samples get different noise
     assert not torch.equal(bart_ds[0]["input_ids"], bart_ds[1]["input_ids"])
+
+
+def test_legacy_indexed_dataset_roundtrip(tmp_path):
+    from fengshen_amd.data.indexed_dataset import (
+        IndexedCachedDataset,
+        IndexedDataset,
+        IndexedDatasetBuilder,
+        infer_dataset_impl,
+    )
+    p = str(tmp_path / "legacy")
+    b = IndexedDatasetBuilder(p + ".bin", dtype=np.int32)
+    arrs = [np.arange(5), np.arange(3) + 100, np.arange(7) + 200]
+    for a in arrs:
+        b.add_item(a)
+    b.end_document()
+    b.finalize(p + ".idx")
+    assert infer_dataset_impl(p) == "cached"
+    assert IndexedDataset.exists(p)
+    ds = IndexedCachedDataset(p)
+    ds.prefetch([0, 2])
+    for i, a in enumerate(arrs):
+        assert (ds[i] == a).all()
+    assert len(ds) == 3
