@@ -7,6 +7,7 @@ equivalence oracle in tests.
 from __future__ import annotations
 
 import logging
+import os
 
 import numpy as np
 
@@ -96,3 +97,26 @@ def build_blending_indices(weights, size: int):
     if ext is not None:
         return ext.build_blending_indices(w, size)
     return py_build_blending_indices(w, size)
+
+
+def get_samples_mapping(indexed, data_prefix, num_epochs, max_num_samples,
+                        max_seq_length, short_seq_prob, seed, name):
+    """Cached samples mapping: build once, mmap-load thereafter
+    (reference dataset_utils.py:731-788 — '180GB loads in seconds').
+
+    The mapping is persisted next to the corpus as
+    {data_prefix}_{name}_{epochs}ep_{max_seq}msl_{seed}s_indexmap.npy.
+    """
+    cache = (f"{data_prefix}_{name}_{num_epochs}ep_{max_seq_length}msl_"
+             f"{seed}s_indexmap.npy")
+    if os.path.exists(cache):
+        return np.load(cache, mmap_mode="r")
+    mapping = build_mapping(
+        indexed.doc_idx, np.asarray(indexed.sizes, dtype=np.int32),
+        num_epochs, max_num_samples, max_seq_length, short_seq_prob, seed)
+    # write atomically so concurrent DP ranks never read a partial file
+    tmp = f"{cache}.tmp{os.getpid()}.npy"
+    with open(tmp, "wb") as f:
+        np.save(f, mapping)
+    os.replace(tmp, cache)
+    return np.load(cache, mmap_mode="r")

@@ -271,3 +271,25 @@ def test_legacy_indexed_dataset_roundtrip(tmp_path):
     for i, a in enumerate(arrs):
         assert (ds[i] == a).all()
     assert len(ds) == 3
+
+
+def test_get_samples_mapping_cache(tmp_path):
+    from fengshen_amd.data.helpers_py import get_samples_mapping
+    from fengshen_amd.data.indexed_dataset import (
+        MMapIndexedDataset,
+        MMapIndexedDatasetBuilder,
+    )
+    rng = np.random.RandomState(0)
+    p = str(tmp_path / "corp")
+    b = MMapIndexedDatasetBuilder(p + ".bin", dtype=np.int32)
+    for _ in range(5):
+        for _ in range(3):
+            b.add_item(rng.randint(10, 200, size=12))
+        b.end_document()
+    b.finalize(p + ".idx")
+    ds = MMapIndexedDataset(p)
+    m1 = get_samples_mapping(ds, p, 2, 2 ** 62, 32, 0.1, 1234, "t")
+    cache = [f for f in tmp_path.iterdir() if "indexmap" in f.name]
+    assert len(cache) == 1
+    m2 = get_samples_mapping(ds, p, 2, 2 ** 62, 32, 0.1, 1234, "t")
+    assert (np.asarray(m1) == np.asarray(m2)).all()
