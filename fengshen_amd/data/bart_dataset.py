@@ -34,7 +34,8 @@ class BartMmapDataset(torch.utils.data.Dataset):
                  random_ratio: float = 0.1,
                  seg_token_ids: Optional[Set[int]] = None,
                  short_seq_prob: float = 0.1, num_epochs: int = 1,
-                 max_num_samples: Optional[int] = None, seed: int = 1234):
+                 max_num_samples: Optional[int] = None, seed: int = 1234,
+                 data_prefix: Optional[str] = None):
         self.indexed = indexed
         self.vocab_id_to_token = vocab_id_to_token
         self.cls_id, self.sep_id = cls_id, sep_id
@@ -48,9 +49,16 @@ class BartMmapDataset(torch.utils.data.Dataset):
         self.seed = seed
         if max_num_samples is None:
             max_num_samples = 2 ** 62
-        self.samples_mapping = build_mapping(
-            indexed.doc_idx, indexed.sizes.astype(np.int32), num_epochs,
-            max_num_samples, max_seq_length - 3, short_seq_prob, seed)
+        if data_prefix is not None:
+            # persisted .npy indexmap so TB corpora build the mapping once
+            from fengshen_amd.data.helpers_py import get_samples_mapping
+            self.samples_mapping = get_samples_mapping(
+                indexed, data_prefix, num_epochs, max_num_samples,
+                max_seq_length - 3, short_seq_prob, seed, "bart")
+        else:
+            self.samples_mapping = build_mapping(
+                indexed.doc_idx, indexed.sizes.astype(np.int32), num_epochs,
+                max_num_samples, max_seq_length - 3, short_seq_prob, seed)
 
     def __len__(self):
         return len(self.samples_mapping)

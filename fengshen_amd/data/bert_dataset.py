@@ -31,7 +31,8 @@ class BertMmapDataset(torch.utils.data.Dataset):
                  sep_id: int, mask_id: int, pad_id: int,
                  max_seq_length: int = 512, masked_lm_prob: float = 0.15,
                  short_seq_prob: float = 0.1, num_epochs: int = 1,
-                 max_num_samples: Optional[int] = None, seed: int = 1234):
+                 max_num_samples: Optional[int] = None, seed: int = 1234,
+                 data_prefix: Optional[str] = None):
         self.indexed = indexed
         self.vocab_id_list = vocab_id_list
         self.vocab_id_to_token = vocab_id_to_token
@@ -42,9 +43,16 @@ class BertMmapDataset(torch.utils.data.Dataset):
         self.seed = seed
         if max_num_samples is None:
             max_num_samples = 2 ** 62
-        self.samples_mapping = build_mapping(
-            indexed.doc_idx, indexed.sizes.astype(np.int32), num_epochs,
-            max_num_samples, max_seq_length - 3, short_seq_prob, seed)
+        if data_prefix is not None:
+            # persisted .npy indexmap so TB corpora build the mapping once
+            from fengshen_amd.data.helpers_py import get_samples_mapping
+            self.samples_mapping = get_samples_mapping(
+                indexed, data_prefix, num_epochs, max_num_samples,
+                max_seq_length - 3, short_seq_prob, seed, "bert")
+        else:
+            self.samples_mapping = build_mapping(
+                indexed.doc_idx, indexed.sizes.astype(np.int32), num_epochs,
+                max_num_samples, max_seq_length - 3, short_seq_prob, seed)
 
     def __len__(self):
         return len(self.samples_mapping)
