@@ -293,3 +293,50 @@ def test_get_samples_mapping_cache(tmp_path):
     assert len(cache) == 1
     m2 = get_samples_mapping(ds, p, 2, 2 ** 62, 32, 0.1, 1234, "t")
     assert (np.asarray(m1) == np.asarray(m2)).all()
+
+
+def test_tagging_collators_feed_heads():
+    """span/biaffine collators produce exactly what BertSpan/BertBiaffine
+    consume (ref sequence_tagging_collator.py:9-206)."""
+    from fengshen_amd.data.tagging_collators import (
+        CollatorForBiaffine,
+        CollatorForCrf,
+        CollatorForLinear,
+        CollatorForSpan,
+    )
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config,
+    )
+    from fengshen_amd.models.tagging_models.bert_for_tagging import (
+        BertBiaffine,
+        BertSpan,
+    )
+    from fengshen_amd.tokenizer import SimpleCharTokenizer
+
+    tk = SimpleCharTokenizer()
+    samples = [{"text": "abcdef", "entities": [(0, 1, 2), (3, 5, 1)],
+                "labels": ["B-a", "I-a", "O", "B-b", "I-b", "I-b"]},
+               {"text": "abc", "entities": [(1, 2, 3)],
+                "labels": ["O", "B-c", "I-c"]}]
+
+    lin = CollatorForLinear(tk, {"O": 0, "B-a": 1, "I-a": 2, "B-b": 3,
+                                 "I-b": 4, "B-c": 5, "I-c": 6})
+    b = lin(samples)
+    assert b["labels"].shape == b["input_ids"].shape
+    assert CollatorForCrf is CollatorForLinear
+
+    cfg = bert_tiny_config(vocab_size=300, torch_dtype="float32")
+    span_batch = CollatorForSpan(tk)(samples)
+    m = BertSpan(cfg, num_labels=5).float()
+    out = m(**span_batch)
+    assert out.loss.isfinite()
+    # boundary labels landed (+1 for CLS)
+    assert span_batch["start_positions"][0, 1] == 2
+    assert span_batch["end_positions"][0, 2] == 2
+
+    bia_batch = CollatorForBiaffine(tk)(samples)
+    m2 = BertBiaffine(cfg, num_labels=5).float()
+    out2 = m2(**bia_batch)
+    assert out2.loss.isfinite()
+    assert bia_batch["span_labels"][0, 1, 2] == 2       # entity cell
+    assert bia_batch["span_labels"][0, 2, 1] == -100    # below diagonal
