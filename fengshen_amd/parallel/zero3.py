@@ -254,10 +254,18 @@ class Zero3Engine(torch.optim.Optimizer):
     def _register_module_hooks(self, m: nn.Module, u: _Unit, is_root=False):
         def pre(_m, _inp):
             u.ensure_gathered(self.group)
-            if self.prefetch and not torch.is_grad_enabled():
+            if self.prefetch:
                 i = self.units.index(u)
-                if i + 1 < len(self.units):
-                    self.units[i + 1].launch_gather(self.group)
+                if not torch.is_grad_enabled():
+                    # forward order: hide the NEXT unit's gather
+                    if i + 1 < len(self.units):
+                        self.units[i + 1].launch_gather(self.group)
+                elif i - 1 >= 0:
+                    # grad-enabled pass = activation-ckpt recompute during
+                    # backward, which walks layers in REVERSE: prefetch the
+                    # previous unit so its all-gather hides under this
+                    # unit's recompute+grads (no-op if still gathered)
+                    self.units[i - 1].launch_gather(self.group)
             return None
 
         def post(_m, _inp, _out):
