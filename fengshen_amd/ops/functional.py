@@ -117,6 +117,7 @@ class _RMSNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, eps):
         if use_hip(x):
+            x = x.contiguous()  # kernel reads rows as flat vectors
             out, invrms = get_ext().rms_norm_fwd(x, weight, eps)
         else:
             x32 = x.float()
@@ -159,6 +160,7 @@ class _LayerNorm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, eps):
         if use_hip(x):
+            x = x.contiguous()  # kernel reads rows as flat vectors
             out, mean, invstd = get_ext().layer_norm_fwd(x, weight, bias, eps)
             ctx.save_for_backward(x, weight, mean, invstd)
             ctx.has_bias = bias is not None
