@@ -220,8 +220,9 @@ def main():
 
     if rank == 0:
         print(json.dumps({
-            "metric": ("samples/sec Erlangshen-1.3B MLM pretrain" if is_bert
-                       else "tokens/sec Ziya-LLaMA-13B ZeRO"
+            "metric": ("samples/sec/node pretrain Erlangshen-1.3B"
+                       if is_bert
+                       else f"tokens/sec Ziya-LLaMA-13B ZeRO-{args.zero_stage}"
                        if args.model == "ziya-llama-13b"
                        else f"tokens/sec {model_name}"),
             "value": round(value, 2),
