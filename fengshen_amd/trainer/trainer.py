@@ -353,6 +353,11 @@ class Trainer:
                 if (self.limit_train_batches is not None
                         and batch_idx >= self.limit_train_batches):
                     break
+                # guard BEFORE the step so resuming at global_step ==
+                # max_steps trains zero further steps
+                if self.max_steps > 0 and self.global_step >= self.max_steps:
+                    self.should_stop = True
+                    break
                 batch = _move_to_device(batch, dev)
                 model.on_train_batch_start(batch, batch_idx)
                 self._call("on_train_batch_start", batch, batch_idx)
