@@ -10,7 +10,7 @@ import torch.nn as nn
 from tests.distributed_utils import run_distributed
 
 
-def _zero_resume_worker(rank, world_size, tmp_root):
+def _zero_resume_worker(rank, world_size, tmp_root, strategy="zero2"):
     import torch.distributed as dist
 
     from fengshen_amd import FengshenModule, Trainer, UniversalDataModule
@@ -70,13 +70,13 @@ def _zero_resume_worker(rank, world_size, tmp_root):
     # uninterrupted: 8 steps
     args = make_args()
     model_a = ToyModule(args)
-    tr_a = Trainer(max_steps=8, precision="fp32", strategy="zero2",
+    tr_a = Trainer(max_steps=8, precision="fp32", strategy=strategy,
                    default_root_dir=os.path.join(tmp_root, "a"))
     tr_a.fit(model_a, datamodule=dm(args))
 
     # interrupted: 4 steps -> save -> fresh model -> resume to 8
     model_b = ToyModule(args)
-    tr_b = Trainer(max_steps=4, precision="fp32", strategy="zero2",
+    tr_b = Trainer(max_steps=4, precision="fp32", strategy=strategy,
                    default_root_dir=os.path.join(tmp_root, "b"))
     tr_b.fit(model_b, datamodule=dm(args))
     ckpt = os.path.join(tmp_root, "ckpt4")
@@ -84,7 +84,7 @@ def _zero_resume_worker(rank, world_size, tmp_root):
     dist.barrier()
 
     model_c = ToyModule(args)
-    tr_c = Trainer(max_steps=8, precision="fp32", strategy="zero2",
+    tr_c = Trainer(max_steps=8, precision="fp32", strategy=strategy,
                    default_root_dir=os.path.join(tmp_root, "c"))
     tr_c.fit(model_c, datamodule=dm(args), ckpt_path=ckpt)
     assert tr_c.global_step == 8
@@ -101,3 +101,12 @@ def test_zero2_trainer_resume_exact(tmp_path):
                             args=(str(tmp_path),), timeout=300)
     for d in diffs:
         assert d < 1e-5, f"ZeRO-2 resume diverged: max param diff {d}"
+
+
+def test_zero2_offload_trainer_resume_exact(tmp_path):
+    """same exactness with host-RAM optimizer states."""
+    diffs = run_distributed(_zero_resume_worker, world_size=2,
+                            args=(str(tmp_path), "zero2_offload"),
+                            timeout=300)
+    for d in diffs:
+        assert d < 1e-5, f"ZeRO-offload resume diverged: {d}"
