@@ -274,7 +274,13 @@ class Zero3Engine(torch.optim.Optimizer):
             return None
 
         m.register_forward_pre_hook(pre)
-        if not is_root:
+        if is_root:
+            # training_step may invoke child modules directly without ever
+            # calling the root's forward (Lightning-style modules), so the
+            # root unit's gather must also trigger from its children
+            for child in m.children():
+                child.register_forward_pre_hook(pre)
+        else:
             # grad-enabled passes release in the grad hook instead (releasing
             # any earlier races with wgrad AccumulateGrad shape checks); the
             # root unit (embeddings early, head late) stays gathered all pass
@@ -379,6 +385,16 @@ class Zero3Engine(torch.optim.Optimizer):
     def gathered_params(self):
         """Context manager: all params materialized (for state_dict etc.)."""
         return Zero3Engine._GatherAll(self)
+
+    def refresh_shards_from_params(self):
+        """After externally writing the gathered full params (checkpoint
+        load inside gathered_params()), pull this rank's slice back into
+        the shard and refresh the fp32 master."""
+        for u in self.units:
+            assert u.gathered, "call inside gathered_params()"
+            start = u.rank * u.shard_numel
+            u.shard.copy_(u.full[start:start + u.shard_numel])
+            u.master.copy_(u.shard.float())
 
     def state_dict(self) -> dict:
         return {

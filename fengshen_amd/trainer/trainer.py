@@ -544,7 +544,14 @@ class Trainer:
         paths = self._ckpt_paths(ckpt_dir)
         meta = self._peek_checkpoint(ckpt_dir)
         state = torch.load(paths["model"], map_location="cpu", weights_only=False)
-        self.module.load_state_dict(state)
+        from fengshen_amd.parallel.zero3 import Zero3Engine
+        if isinstance(self.optimizer, Zero3Engine):
+            # params are sharded stubs: materialize, load, re-shard
+            with self.optimizer.gathered_params():
+                self.module.load_state_dict(state)
+                self.optimizer.refresh_shards_from_params()
+        else:
+            self.module.load_state_dict(state)
         # re-sync flat buffers in ZeRO (params were re-assigned by load? no:
         # load_state_dict copies INTO the flat views, so buffers are current)
         if os.path.exists(paths["optim"]) and self.optimizer is not None:
