@@ -10,7 +10,8 @@ import torch.nn as nn
 from tests.distributed_utils import run_distributed
 
 
-def _zero_resume_worker(rank, world_size, tmp_root, strategy="zero2"):
+def _zero_resume_worker(rank, world_size, tmp_root, strategy="zero2",
+                        accum=1):
     import torch.distributed as dist
 
     from fengshen_amd import FengshenModule, Trainer, UniversalDataModule
@@ -78,6 +79,7 @@ def _zero_resume_worker(rank, world_size, tmp_root, strategy="zero2"):
     args = make_args()
     model_a = ToyModule(args)
     tr_a = Trainer(max_steps=8, precision="fp32", strategy=strategy,
+                   accumulate_grad_batches=accum,
                    default_root_dir=os.path.join(tmp_root, "a"))
     tr_a.fit(model_a, datamodule=dm(args))
     snap_a = snapshot(tr_a, model_a)
@@ -85,6 +87,7 @@ def _zero_resume_worker(rank, world_size, tmp_root, strategy="zero2"):
     # interrupted: 4 steps -> save -> fresh model -> resume to 8
     model_b = ToyModule(args)
     tr_b = Trainer(max_steps=4, precision="fp32", strategy=strategy,
+                   accumulate_grad_batches=accum,
                    default_root_dir=os.path.join(tmp_root, "b"))
     tr_b.fit(model_b, datamodule=dm(args))
     ckpt = os.path.join(tmp_root, "ckpt4")
@@ -93,6 +96,7 @@ def _zero_resume_worker(rank, world_size, tmp_root, strategy="zero2"):
 
     model_c = ToyModule(args)
     tr_c = Trainer(max_steps=8, precision="fp32", strategy=strategy,
+                   accumulate_grad_batches=accum,
                    default_root_dir=os.path.join(tmp_root, "c"))
     tr_c.fit(model_c, datamodule=dm(args), ckpt_path=ckpt)
     assert tr_c.global_step == 8
@@ -126,3 +130,11 @@ def test_zero3_trainer_resume_exact(tmp_path):
                             args=(str(tmp_path), "zero3"), timeout=300)
     for d in diffs:
         assert d < 1e-5, f"ZeRO-3 resume diverged: {d}"
+
+
+def test_zero2_resume_with_grad_accumulation(tmp_path):
+    """resume mid-training with accumulate_grad_batches=2 stays exact."""
+    diffs = run_distributed(_zero_resume_worker, world_size=2,
+                            args=(str(tmp_path), "zero2", 2), timeout=300)
+    for d in diffs:
+        assert d < 1e-5, f"accum resume diverged: {d}"
