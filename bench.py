@@ -1,7 +1,7 @@
 #!/usr/bin/env python3
 """Flagship benchmark: Ziya-LLaMA-13B causal-LM training step, bf16, native
 ZeRO over RCCL, synthetic data, random-init weights (BASELINE.json metric:
-"tokens/sec Ziya-LLaMA-13B ZeRO @1/2/4/8 GPU", weak scaling).
+"tokens/sec Ziya-LLaMA-13B ZeRO-3 @1/2/4/8 GPU", weak scaling).
 
 Driver contract:
   python bench.py --gpus N --steps K --warmup W
@@ -112,7 +112,8 @@ def main():
     model = model.to(torch.bfloat16).to(device)
     if hasattr(model, "gradient_checkpointing_enable"):
         try:
-            model.gradient_checkpointing_enable(skip_interval=args.ckpt_skip)
+            model.gradient_checkpointing_enable(
+                skip_interval=max(args.ckpt_skip, 0))
         except TypeError:
             model.gradient_checkpointing_enable()
     model.train()
