@@ -121,7 +121,8 @@ class GPT2LMHeadModel(GPT2PreTrainedModel, GenerationMixin):
 
     def gradient_checkpointing_enable(self, skip_interval: int = 0, **_kw):
         self.transformer.gradient_checkpointing = True
-        self.transformer.gradient_checkpointing_
+        self.transformer.gradient_checkpointing_skip_interval = max(
+            skip_interval, 0)
 
     def forward(self, input_ids, attention_mask=None, labels=None,
                 past_key_values=None, use_cache: bool = False,

@@ -136,7 +136,7 @@ class LlamaForCausalLM(LlamaPreTrainedModel, GenerationMixin):
 
     def gradient_checkpointing_enable(self, skip_interval: int = 0, **_kw):
         self.model.gradient_checkpointing = True
-        self.model.gradient_checkpointing_
+        self.model.gradient_checkpointing_skip_interval = max(skip_interval, 0)
 
     def forward(self, input_ids: torch.Tensor,
                 attention_mask: Optional[torch.Tensor] = None,

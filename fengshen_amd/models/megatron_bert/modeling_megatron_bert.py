@@ -184,7 +184,7 @@ class MegatronBertForPreTraining(MegatronBertPreTrainedModel):
 
     def gradient_checkpointing_enable(self, skip_interval: int = 0, **_kw):
         self.bert.gradient_checkpointing = True
-        self.bert.gradient_checkpointing_
+        self.bert.gradient_checkpointing_skip_interval = max(skip_interval, 0)
 
     def forward(self, input_ids, attention_mask=None, token_type_ids=None,
                 labels=None, next_sentence_label=None, position_ids=None,
@@ -222,7 +222,7 @@ class MegatronBertForMaskedLM(MegatronBertPreTrainedModel):
 
     def gradient_checkpointing_enable(self, skip_interval: int = 0, **_kw):
         self.bert.gradient_checkpointing = True
-        self.bert.gradient_checkpointing_
+        self.bert.gradient_checkpointing_skip_interval = max(skip_interval, 0)
 
     def forward(self, input_ids, attention_mask=None, token_type_ids=None,
                 labels=None, return_dict=True, **_kw):
