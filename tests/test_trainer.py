@@ -211,3 +211,18 @@ def test_roctx_marker_callback(tmp_path):
                       default_root_dir=str(tmp_path))
     trainer.fit(model, train_dataloaders=loader)
     assert not marker._open  # every pushed range was popped
+
+
+def test_predict_loop(tmp_path):
+    class PredModule(ToyModule):
+        def predict_step(self, batch, batch_idx):
+            return self.net(batch["x"]).squeeze(-1)
+
+    args = _make_args()
+    torch.manual_seed(0)
+    model = PredModule(args)
+    loader = DataLoader(ToyDataset(n=32), batch_size=8)
+    trainer = Trainer(precision="fp32", default_root_dir=str(tmp_path))
+    outs = trainer.predict(model, loader)
+    assert len(outs) == 4
+    assert all(o.shape == (8,) for o in outs)
