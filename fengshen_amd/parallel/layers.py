@@ -51,15 +51,16 @@ def _init_partition(weight: torch.Tensor, init_method: Callable, full_shape=None
                     partition_dim: int = 0, stride: int = 1):
     """Initialize a TP-partitioned weight so the sharded init matches the
     unsharded init sliced (master-weight init, reference mpu/layers.py:180-220):
-    materialize the full fp32 weight with the shared TP RNG, then slice."""
+    materialize the full fp32 master with the TP-IDENTICAL default RNG
+    stream, then slice.  (The tracker's model-parallel stream is
+    tp-distinct by design — right for dropout, wrong for master init:
+    each rank would slice a different master.)"""
     tp = groups.get_tensor_model_parallel_world_size()
     if tp == 1 or full_shape is None:
-        with get_rng_tracker().fork():
-            init_method(weight)
+        init_method(weight)
         return
     master = torch.empty(full_shape, dtype=torch.float32, device=weight.device)
-    with get_rng_tracker().fork():
-        init_method(master)
+    init_method(master)
     rank = groups.get_tensor_model_parallel_rank()
     per = weight.size(partition_dim)
     shard = master.narrow(partition_dim, rank * per, per)
@@ -187,3 +188,61 @@ class RowParallelLinear(nn.Module):
         if self.bias is not None:
             output_ = output_ + self.bias
         return output_
+
+
+class ParallelRelativePositionBias(nn.Module):
+    """T5-style bucketed relative attention bias, sharded over heads
+    (reference mpu/layers.py:133-258: per-TP-rank head slice of the
+    [num_buckets, heads] embedding; bias added to attention scores).
+
+    forward(q_len, k_len) -> [1, heads/tp, q_len, k_len].
+    """
+
+    def __init__(self, num_buckets: int = 32, max_distance: int = 128,
+                 num_heads: int = 12, causal: bool = True,
+                 init_method: Callable = init.xavier_normal_):
+        super().__init__()
+        tp = groups.get_tensor_model_parallel_world_size()
+        self.num_buckets = num_buckets
+        self.max_distance = max_distance
+        self.causal = causal
+        self.heads_per_partition = divide(num_heads, tp)
+        self.weight = nn.Parameter(
+            torch.empty(num_buckets, self.heads_per_partition,
+                        dtype=torch.float32))
+        _init_partition(self.weight, init_method,
+                        full_shape=(num_buckets, num_heads),
+                        partition_dim=1)
+        self._cache = {}
+
+    def _bucket(self, relative_position: torch.Tensor) -> torch.Tensor:
+        """T5 relative_position_bucket: half the buckets exact, half
+        log-spaced out to max_distance."""
+        num_buckets = self.num_buckets
+        ret = torch.zeros_like(relative_position)
+        n = -relative_position
+        if not self.causal:
+            num_buckets //= 2
+            ret = ret + (n < 0).long() * num_buckets
+            n = n.abs()
+        else:
+            n = torch.clamp(n, min=0)
+        max_exact = num_buckets // 2
+        is_small = n < max_exact
+        import math as _m
+        val_large = max_exact + (
+            torch.log(n.float() / max_exact + 1e-6)
+            / _m.log(self.max_distance / max_exact)
+            * (num_buckets - max_exact)).long()
+        val_large = torch.clamp(val_large, max=num_buckets - 1)
+        return ret + torch.where(is_small, n, val_large)
+
+    def forward(self, q_len: int, k_len: int) -> torch.Tensor:
+        key = (q_len, k_len, str(self.weight.device))
+        if key not in self._cache:
+            ctx = torch.arange(q_len, device=self.weight.device)[:, None]
+            mem = torch.arange(k_len, device=self.weight.device)[None, :]
+            self._cache[key] = self._bucket(mem - ctx)
+        buckets = self._cache[key]
+        bias = F.embedding(buckets, self.weight)        # [q, k, h/tp]
+        return bias.permute(2, 0, 1).unsqueeze(0).to(self.weight.dtype)

@@ -152,3 +152,45 @@ def _mappings_worker(rank, world_size):
 
 def test_region_mappings():
     assert all(run_distributed(_mappings_worker, world_size=2))
+
+
+def _rel_bias_worker(rank, world_size, _):
+    import torch.distributed as dist
+    from fengshen_amd.parallel.groups import (
+        init_distributed,
+        initialize_model_parallel,
+    )
+    from fengshen_amd.parallel.layers import ParallelRelativePositionBias
+    from fengshen_amd.parallel.random import model_parallel_manual_seed
+
+    init_distributed(backend="gloo")
+    initialize_model_parallel(tensor_model_parallel_size=2)
+    model_parallel_manual_seed(7)
+    b = ParallelRelativePositionBias(num_buckets=16, max_distance=64,
+                                     num_heads=4)
+    out = b(6, 6).detach()
+    dist.destroy_process_group()
+    return out
+
+
+def test_parallel_relative_position_bias_shards_heads():
+    """TP-2 shards the head dim; concatenated shards == TP-1 full bias."""
+    import fengshen_amd.parallel.groups as pg
+    from fengshen_amd.parallel.layers import ParallelRelativePositionBias
+    from fengshen_amd.parallel.random import model_parallel_manual_seed
+
+    model_parallel_manual_seed(7)
+    full = ParallelRelativePositionBias(num_buckets=16, max_distance=64,
+                                        num_heads=4)
+    ref = full(6, 6).detach()
+    assert ref.shape == (1, 4, 6, 6)
+    # bucketing sanity: distance 0 on the diagonal maps to one bucket
+    ctx = torch.arange(6)
+    buckets = full._bucket(ctx[None, :] - ctx[:, None])
+    assert (buckets.diagonal() == buckets[0, 0]).all()
+
+    shards = run_distributed(_rel_bias_worker, world_size=2, args=(None,))
+    merged = torch.cat(shards, dim=1)
+    assert merged.shape == ref.shape
+    assert torch.allclose(merged, ref, atol=1e-6), \
+        (merged - ref).abs().max()
