@@ -50,7 +50,8 @@ class _ScaledMaskedSoftmax(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, mask, scale):
         if use_hip(x):
-            out = get_ext().scaled_masked_softmax_fwd(x, mask, scale)
+            out = get_ext().scaled_masked_softmax_fwd(x.contiguous(), mask,
+                                                      scale)
         else:
             out = eager_scaled_masked_softmax(x, mask, scale)
         ctx.save_for_backward(out)
@@ -74,7 +75,7 @@ class _ScaledCausalSoftmax(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, scale):
         if use_hip(x):
-            out = get_ext().scaled_causal_softmax_fwd(x, scale)
+            out = get_ext().scaled_causal_softmax_fwd(x.contiguous(), scale)
         else:
             out = eager_scaled_causal_softmax(x, scale)
         ctx.save_for_backward(out)
