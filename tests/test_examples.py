@@ -25,6 +25,7 @@ def _run(rel, *extra):
      ("--n", "2", "--seq_len", "8")),
     ("examples/transfo_xl_denoise/generate.py", ("--seq_len", "28")),
     ("examples/fastdemo/qa_demo.py", ("--smoke",)),
+    ("examples/disco_project/clip_guided_generate.py", ("--steps", "4")),
 ])
 def test_example_smokes(rel, extra):
     _run(rel, *extra)
