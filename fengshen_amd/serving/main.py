@@ -30,6 +30,7 @@ class APIConfig(BaseModel):
     port: int = 8000
     allow_origins: list = ["*"]
     pipeline_kwargs: dict = {}
+    log_file: Optional[str] = None  # file+console logging (ref API/utils.py:131-155)
 
     @classmethod
     def from_json(cls, path: str) -> "APIConfig":
@@ -49,6 +50,13 @@ def build_app(config: APIConfig, pipeline=None):
             if config.model else None
         pipeline = mod.Pipeline(model=config.model, tokenizer=tokenizer,
                                 **config.pipeline_kwargs)
+
+    if config.log_file:
+        fh = logging.FileHandler(config.log_file)
+        fh.setFormatter(logging.Formatter(
+            "%(asctime)s %(levelname)s %(message)s"))
+        logger.addHandler(fh)
+        logger.setLevel(logging.INFO)
 
     app = FastAPI(title=f"fengshen_amd {config.pipeline_type}")
     app.add_middleware(
