@@ -210,6 +210,8 @@ class ParallelRelativePositionBias(nn.Module):
         self.weight = nn.Parameter(
             torch.empty(num_buckets, self.heads_per_partition,
                         dtype=torch.float32))
+        setattr(self.weight, "tensor_model_parallel", True)
+        setattr(self.weight, "partition_dim", 1)
         _init_partition(self.weight, init_method,
                         full_shape=(num_buckets, num_heads),
                         partition_dim=1)
