@@ -194,3 +194,28 @@ def test_parallel_relative_position_bias_shards_heads():
     assert merged.shape == ref.shape
     assert torch.allclose(merged, ref, atol=1e-6), \
         (merged - ref).abs().max()
+
+
+def test_rel_bias_shard_merge_roundtrip():
+    """tp_convert handles the dim-1 (head-sharded) relative-bias weight."""
+    import torch.nn as nn
+
+    from fengshen_amd.parallel.layers import ParallelRelativePositionBias
+    from fengshen_amd.utils.tp_convert import (
+        merge_state_dicts,
+        shard_state_dict,
+    )
+
+    class M(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.bias = ParallelRelativePositionBias(num_buckets=8,
+                                                     num_heads=4)
+
+    torch.manual_seed(0)
+    m = M()
+    full = {k: v.clone() for k, v in m.state_dict().items()}
+    shards = [shard_state_dict(m, full, 2, r) for r in range(2)]
+    assert shards[0]["bias.weight"].shape == (8, 2)
+    merged = merge_state_dicts(m, shards)
+    assert torch.equal(merged["bias.weight"], full["bias.weight"])
