@@ -27,7 +27,6 @@ import torch
 from fengshen_amd import FengshenModule, Trainer, UniversalDataModule
 from fengshen_amd.data.collators import SftCollator
 from fengshen_amd.models.llama.configuration_llama import (
-    LlamaConfig,
     llama_tiny_config,
     ziya_llama_13b_config,
 )

@@ -21,7 +21,6 @@ from fengshen_amd.parallel.mappings import (
     reduce_from_tensor_model_parallel_region,
     scatter_to_tensor_model_parallel_region,
 )
-from fengshen_amd.parallel.random import get_rng_tracker
 
 
 def ensure_divisibility(numerator: int, denominator: int) -> None:

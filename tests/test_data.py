@@ -1,7 +1,6 @@
 """Data layer tests: mmap dataset roundtrip, GPT sample windows, blending,
 samplers, masking utils, collators."""
 import numpy as np
-import pytest
 import torch
 
 from fengshen_amd.data.indexed_dataset import (

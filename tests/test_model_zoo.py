@@ -2,7 +2,6 @@
 CRF, metrics."""
 import torch
 
-from tests.test_data import FakeTokenizer
 
 
 def test_t5_train_and_generate():

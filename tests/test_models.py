@@ -2,7 +2,6 @@
 models, plus TP=2 parity vs TP=1 via shard_state_dict (gloo)."""
 import os
 
-import pytest
 import torch
 
 from tests.distributed_utils import run_distributed

@@ -1,6 +1,5 @@
 """TP numerics on gloo world_size=2: parallel layers vs plain nn.Linear,
 vocab-parallel CE vs dense CE, broadcast_data."""
-import pytest
 import torch
 import torch.nn as nn
 
@@ -175,7 +174,6 @@ def _rel_bias_worker(rank, world_size, _):
 
 def test_parallel_relative_position_bias_shards_heads():
     """TP-2 shards the head dim; concatenated shards == TP-1 full bias."""
-    import fengshen_amd.parallel.groups as pg
     from fengshen_amd.parallel.layers import ParallelRelativePositionBias
     from fengshen_amd.parallel.random import model_parallel_manual_seed
 

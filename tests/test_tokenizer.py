@@ -1,6 +1,4 @@
 """Tokenizer tooling: char tokenizer + sentencepiece training."""
-import torch
-
 from fengshen_amd.tokenizer import SimpleCharTokenizer
 
 

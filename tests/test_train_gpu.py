@@ -1,9 +1,6 @@
 """GPU end-to-end training smokes: tiny LLaMA + ZeRO on 1 MI355X, bf16,
 with the HIP kernels on the hot path (extension required, no eager fallback)."""
-import json
 import os
-import subprocess
-import sys
 
 import pytest
 import torch

@@ -4,7 +4,6 @@ import argparse
 import math
 import os
 
-import pytest
 import torch
 import torch.nn as nn
 from torch.utils.data import DataLoader, Dataset

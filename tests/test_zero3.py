@@ -1,6 +1,5 @@
 """ZeRO-3 numerics: sharded params + gather-on-forward across 2 gloo ranks
 must match a single-process AdamW run with the same decay split."""
-import pytest
 import torch
 import torch.nn as nn
 
