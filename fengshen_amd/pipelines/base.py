@@ -53,6 +53,10 @@ class _TaskModule(FengshenModule):
 
 
 class BasePipeline:
+    """Shared pipeline base: ctor(model=hub-id-or-instance, tokenizer),
+    `__call__(inputs)` for inference, `.train(datasets)` spins a Trainer
+    (reference pipelines/*.Pipeline interface)."""
+
     task_name = "base"
 
     def __init__(self, args=None, model=None, tokenizer=None):

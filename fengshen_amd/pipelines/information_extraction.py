@@ -8,6 +8,8 @@ from fengshen_amd.pipelines.base import BasePipeline
 
 
 class InformationExtractionPipeline(BasePipeline):
+    """UBERT span extraction (entity types as prompts)."""
+
     task_name = "information_extraction"
 
     def __init__(self, args=None, model=None, tokenizer=None, config=None):

@@ -8,6 +8,8 @@ from fengshen_amd.pipelines.base import BasePipeline
 
 
 class MultipleChoicePipeline(BasePipeline):
+    """UniMC zero-shot label-as-option choice."""
+
     task_name = "multiplechoice"
 
     def __init__(self, args=None, model=None, tokenizer=None, config=None,

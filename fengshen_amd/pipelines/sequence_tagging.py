@@ -47,6 +47,8 @@ class _TagCollator:
 
 
 class SequenceTaggingPipeline(BasePipeline):
+    """NER with linear/CRF heads + BIO decoding."""
+
     task_name = "sequence_tagging"
 
     def __init__(self, args=None, model=None, tokenizer=None, id2label=None,

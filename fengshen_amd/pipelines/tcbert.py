@@ -8,6 +8,8 @@ from fengshen_amd.pipelines.base import BasePipeline
 
 
 class TCBertPipeline(BasePipeline):
+    """Prompt-based topic classification (label read at [MASK])."""
+
     task_name = "tcbert"
 
     def __init__(self, args=None, model=None, tokenizer=None, config=None,

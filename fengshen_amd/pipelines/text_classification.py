@@ -62,6 +62,8 @@ class _ClsCollator:
 
 
 class TextClassificationPipeline(BasePipeline):
+    """Sequence classification over Erlangshen-family backbones."""
+
     task_name = "text_classification"
 
     def __init__(self, args=None, model=None, tokenizer=None,

@@ -12,6 +12,7 @@ import torch
 
 
 class Callback:
+    """Hook interface: subclass and override any on_* method."""
     def on_fit_start(self, trainer, module):
         pass
 

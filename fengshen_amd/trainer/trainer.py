@@ -29,6 +29,7 @@ logger = logging.getLogger(__name__)
 
 
 def seed_everything(seed: int):
+    """Seed python/numpy/torch (+CUDA) for reproducible runs."""
     random.seed(seed)
     np.random.seed(seed % (2 ** 32))
     torch.manual_seed(seed)
@@ -48,6 +49,8 @@ def _move_to_device(batch, device):
 
 
 class CSVLogger:
+    """Rank-0 metrics.csv sink (header re-emitted when keys change)."""
+
     def __init__(self, root: str, rank: int):
         self.rank = rank
         self.path = os.path.join(root, "metrics.csv")
@@ -69,6 +72,13 @@ class CSVLogger:
 
 
 class Trainer:
+    """The training orchestrator (see module docstring for scope).
+
+    Construct with keyword flags (or `Trainer.from_argparse_args`), then
+    `fit(module, datamodule=... | train_dataloaders=...)`;
+    `validate` / `predict` reuse the same strategy setup.
+    """
+
     def __init__(
         self,
         max_steps: int = -1,
