@@ -20,7 +20,6 @@ sys.path.insert(0, os.path.abspath(os.path.join(
 
 
 import argparse
-import os
 
 import torch
 
