@@ -117,6 +117,7 @@ def main():
         except TypeError:
             model.gradient_checkpointing_enable()
     model.train()
+    n_params = sum(p.numel() for p in model.parameters())
 
     if args.zero_stage == 3:
         from fengshen_amd.parallel.zero3 import Zero3Engine
@@ -165,9 +166,10 @@ def main():
     b, s = args.micro_batch, args.seq_len
     # TP ranks must see identical batches: seed by DP rank
     g = torch.Generator(device="cpu").manual_seed(42 + dp_rank)
-    # several distinct synthetic batches (avoids single-batch memorization)
+    # one fresh synthetic batch per step (a fixed 4-batch cycle lets a 13B
+    # memorize them and prints a meaningless ~0 loss)
     batches = []
-    for _ in range(4):
+    for _ in range(args.warmup + args.steps):
         ids = torch.randint(3, vocab, (b, s), generator=g).to(device)
         if is_bert:
             labels = ids.clone()
@@ -219,6 +221,15 @@ def main():
         * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
+    # model-FLOPs utilisation (PaLM convention: no recompute counted):
+    # 6*N per token dense + 12*L*h*s^2*b attention (fwd 4bs^2h + bwd 8bs^2h)
+    # n_params was captured before ZeRO-3 sharded the param storage.
+    L = getattr(model.config, "num_hidden_layers", 0)
+    h = getattr(model.config, "hidden_size", 0)
+    flops_step = 6.0 * n_params * b * s + 12.0 * L * h * s * s * b
+    tflops_per_gpu = flops_step * args.steps / elapsed / 1e12
+    mfu = tflops_per_gpu / 2500.0  # MI355X dense bf16 peak ~2.5 PF/s
+
     if rank == 0:
         print(json.dumps({
             "metric": ("samples/sec/node pretrain Erlangshen-1.3B"
@@ -238,6 +249,8 @@ def main():
             "dtype": "bf16",
             "data": "synthetic",
             "loss": round(float(loss.item()), 4),
+            "tflops_per_gpu": round(tflops_per_gpu, 1),
+            "mfu": round(mfu, 4),
             "config": {
                 "model": model_name,
                 "global_batch": b * world,

@@ -180,9 +180,11 @@ class ParallelAttention(nn.Module):
         if cache is not None:
             k, v = cache.update(k, v, self.layer_idx)
 
-        causal = self.causal and q.shape[-2] == k.shape[-2]
-        if offset > 0 and q.shape[-2] == 1:
+        causal = self.causal
+        if causal and offset > 0 and q.shape[-2] == 1:
             causal = False  # single-token decode: all past is visible
+        # chunked prefill (offset>0, sq>1) stays causal: functional.attention
+        # builds the offset-aware rectangular mask (triu(1+sk-sq)).
         ctx = F_ops.attention(q, k, v, causal=causal, mask=attention_mask,
                               dropout_p=self.attention_dropout,
                               training=self.training, scale=self.norm_factor)

@@ -398,9 +398,13 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         probs = probs.view(b, np_, sq, sk)
     else:
         m = mask
-        if causal and sq == sk and mask is not None:
-            cm = torch.ones(sq, sk, dtype=torch.bool, device=q.device).triu(1)
-            m = mask.bool() | cm
+        if causal:
+            # Rectangular causal for chunked prefill (sq < sk with a KV
+            # cache): query i sits at global position (sk - sq + i), so key
+            # j is visible iff j <= sk - sq + i.
+            cm = torch.ones(sq, sk, dtype=torch.bool,
+                            device=q.device).triu(1 + sk - sq)
+            m = cm if mask is None else (mask.bool() | cm)
         probs = scaled_masked_softmax(scores, m, scale)
     if dropout_p > 0.0 and training:
         from fengshen_amd.parallel.random import get_rng_tracker

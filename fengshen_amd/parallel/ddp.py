@@ -62,7 +62,10 @@ class GradReducer:
         off = 0
         for p in b["params"]:
             n = p.numel()
-            b["flat"][off:off + n].copy_(p.grad.reshape(-1))
+            if p.grad is None:
+                b["flat"][off:off + n].zero_()
+            else:
+                b["flat"][off:off + n].copy_(p.grad.reshape(-1))
             off += n
         b["flat"].div_(self.world_size)
         b["work"] = dist.all_reduce(b["flat"], group=self.group, async_op=True)
@@ -72,7 +75,12 @@ class GradReducer:
         if not self.enabled or not self._sync:
             return
         for b in self.buckets:
-            if b["work"] is None and b["ready"] > 0:
+            # ready==0 with grads present happens when every micro-batch ran
+            # under set_sync(False) (trailing grad-accum flush): hooks never
+            # fired with sync on, so launch the reduction here.
+            if b["work"] is None and (
+                    b["ready"] > 0
+                    or any(p.grad is not None for p in b["params"])):
                 self._launch(b)
         for b in self.buckets:
             if b["work"] is None:
@@ -81,7 +89,8 @@ class GradReducer:
             off = 0
             for p in b["params"]:
                 n = p.numel()
-                p.grad.reshape(-1).copy_(b["flat"][off:off + n])
+                if p.grad is not None:
+                    p.grad.reshape(-1).copy_(b["flat"][off:off + n])
                 off += n
             b["work"] = None
             b["ready"] = 0

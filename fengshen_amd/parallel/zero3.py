@@ -208,22 +208,36 @@ class Zero3Engine(torch.optim.Optimizer):
             unit_classes = (ParallelTransformerLayer,)
 
         # ---- partition into units ----------------------------------------
+        # Shared/tied params: a unit may only claim a param whose EVERY
+        # occurrence in the module tree lies inside that unit.  A param tied
+        # into a second module (e.g. embedding tied to the LM head) must live
+        # in the "(rest)" unit, which is gathered at root pre-forward and
+        # stays resident through the whole step — otherwise the outside use
+        # would see released (sharded) storage.
+        appear: Dict[int, int] = {}
+        for _n, p in module.named_parameters(remove_duplicate=False):
+            appear[id(p)] = appear.get(id(p), 0) + 1
+
         unit_modules: List[Tuple[str, nn.Module]] = []
         claimed = set()
         for name, m in module.named_modules():
             if isinstance(m, tuple(unit_classes)):
-                unit_modules.append((name, m))
-                for p in m.parameters():
-                    claimed.add(id(p))
+                inside: Dict[int, int] = {}
+                for _pn, p in m.named_parameters(remove_duplicate=False):
+                    inside[id(p)] = inside.get(id(p), 0) + 1
+                owned = {pid for pid, c in inside.items()
+                         if c == appear.get(pid, c) and pid not in claimed}
+                unit_modules.append((name, m, owned))
+                claimed |= owned
         rest = [(n, p) for n, p in module.named_parameters()
                 if id(p) not in claimed and p.requires_grad]
 
         self.units: List[_Unit] = []
         self._unit_of_module: Dict[int, _Unit] = {}
         self._unit_of_param: Dict[int, _Unit] = {}
-        for name, m in unit_modules:
+        for name, m, owned in unit_modules:
             named = [(f"{name}.{pn}", p) for pn, p in m.named_parameters()
-                     if p.requires_grad]
+                     if p.requires_grad and id(p) in owned]
             if not named:
                 continue
             u = _Unit(name, named, self.world, self.rank)
@@ -312,6 +326,14 @@ class Zero3Engine(torch.optim.Optimizer):
                 sq += (g * g).sum()
         if self.world > 1:
             dist.all_reduce(sq, group=self.group)
+        from fengshen_amd.parallel import groups as pgroups
+        if pgroups.get_tensor_model_parallel_world_size() > 1 and dist.is_initialized():
+            # Sum parallel shards across the TP group so every TP rank clips
+            # with the same coefficient (TP-replicated params like RMSNorm
+            # weights would otherwise drift apart).  Mirrors
+            # ZeroOptimizer._global_grad_norm (zero.py); replicated params are
+            # counted tp times — a tiny, conservative overcount.
+            dist.all_reduce(sq, group=pgroups.get_tensor_model_parallel_group())
         return sq.sqrt()
 
     @torch.no_grad()

@@ -354,9 +354,14 @@ class Trainer:
             model.train()
             model.on_train_epoch_start()
             self._call("on_train_epoch_start")
-            if hasattr(train_loader, "sampler") and hasattr(
-                    train_loader.sampler, "set_epoch"):
-                train_loader.sampler.set_epoch(self.current_epoch)
+            # Prefer batch_sampler: UniversalDataModule builds DataLoader with
+            # batch_sampler=, leaving .sampler a default SequentialSampler, so
+            # checking only .sampler would never reseed the epoch shuffle.
+            for samp in (getattr(train_loader, "batch_sampler", None),
+                         getattr(train_loader, "sampler", None)):
+                if samp is not None and hasattr(samp, "set_epoch"):
+                    samp.set_epoch(self.current_epoch)
+                    break
 
             micro_idx = 0
             for batch_idx, batch in enumerate(train_loader):
@@ -404,6 +409,19 @@ class Trainer:
                     if self.max_steps > 0 and self.global_step >= self.max_steps:
                         self.should_stop = True
                         break
+
+            if accum > 1 and micro_idx % accum != 0 and not self._did_step:
+                # Trailing micro-batches (epoch length not divisible by accum)
+                # accumulated gradients but never stepped; flush a final
+                # optimizer step so they don't leak into the next epoch.
+                self.strategy.set_sync(True)
+                self.strategy.pre_step()
+                self._clip_and_step(model)
+                self._did_step = True
+                self.global_step += 1
+                if (self.scheduler_cfg is not None
+                        and self.scheduler_cfg.get("interval", "step") == "step"):
+                    self.scheduler_cfg["scheduler"].step()
 
             model.on_train_epoch_end()
             self._call("on_train_epoch_end")

@@ -153,3 +153,107 @@ def test_zero3_grad_accumulation():
     for sd in results:
         for k, v in ref.items():
             assert torch.allclose(v, sd[k], atol=2e-5, rtol=1e-4), k
+
+
+class TiedBlock(nn.Module):
+    """Unit-class module that CONTAINS the embedding later tied to the head."""
+
+    def __init__(self, d, vocab):
+        super().__init__()
+        self.emb = nn.Embedding(vocab, d)
+        self.ln = nn.LayerNorm(d)
+
+    def forward(self, ids):
+        return self.ln(self.emb(ids))
+
+
+class TiedModel(nn.Module):
+    """Embedding inside a unit module, tied into an outside LM head — the
+    hazardous ZeRO-3 case: the head's forward must not see released storage.
+    The engine must place the tied weight in the resident "(rest)" unit."""
+
+    def __init__(self, d=32, vocab=50, seed=5):
+        super().__init__()
+        torch.manual_seed(seed)
+        self.block = TiedBlock(d, vocab)
+        self.mid = nn.Sequential(nn.Linear(d, d), nn.GELU())
+        self.head = nn.Linear(d, vocab, bias=False)
+        self.head.weight = self.block.emb.weight  # tie
+
+    def forward(self, ids):
+        return self.head(self.mid(self.block(ids)))
+
+
+def _tied_data(seed=13, n=16, vocab=50):
+    g = torch.Generator().manual_seed(seed)
+    ids = torch.randint(0, vocab, (n, 8), generator=g)
+    return ids
+
+
+def _tied_oracle(steps=4, lr=1e-2, wd=0.01):
+    from fengshen_amd.parallel.zero3 import _is_no_decay
+    model = TiedModel()
+    decay, nodecay = [], []
+    seen = set()
+    for n_, p in model.named_parameters():
+        if id(p) in seen:
+            continue
+        seen.add(id(p))
+        (nodecay if _is_no_decay(n_, p) else decay).append(p)
+    opt = torch.optim.AdamW(
+        [{"params": decay, "weight_decay": wd},
+         {"params": nodecay, "weight_decay": 0.0}],
+        lr=lr, betas=(0.9, 0.999), eps=1e-8)
+    ids = _tied_data()
+    tgt = ids.roll(-1, dims=1)
+    for _ in range(steps):
+        logits = model(ids)
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, logits.shape[-1]), tgt.reshape(-1))
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+    return {k: v.detach().clone() for k, v in model.state_dict().items()}
+
+
+def _tied_worker(rank, world_size, steps=4, lr=1e-2, wd=0.01):
+    import torch.distributed as dist
+    from fengshen_amd.parallel.groups import init_distributed
+    from fengshen_amd.parallel.zero3 import Zero3Engine
+
+    init_distributed(backend="gloo")
+    model = TiedModel()
+    eng = Zero3Engine(model, lr=lr, betas=(0.9, 0.999), eps=1e-8,
+                      weight_decay=wd,
+                      unit_classes=(TiedBlock, nn.Sequential))
+    # tied weight must NOT be claimed by the TiedBlock unit
+    for u in eng.units:
+        if u.name != "(rest)":
+            assert all(p is not model.head.weight for p in u.params), u.name
+    assert any(p is model.head.weight
+               for u in eng.units if u.name == "(rest)" for p in u.params)
+
+    ids = _tied_data()
+    tgt = ids.roll(-1, dims=1)
+    n = ids.shape[0] // world_size
+    ids_r, tgt_r = ids[rank * n:(rank + 1) * n], tgt[rank * n:(rank + 1) * n]
+    for _ in range(steps):
+        logits = model(ids_r)
+        loss = torch.nn.functional.cross_entropy(
+            logits.reshape(-1, logits.shape[-1]), tgt_r.reshape(-1))
+        eng.zero_grad()
+        loss.backward()
+        eng.step()
+    with eng.gathered_params():
+        out = {k: v.detach().clone() for k, v in model.state_dict().items()}
+    dist.destroy_process_group()
+    return out
+
+
+def test_zero3_tied_parameters_world2():
+    ref = _tied_oracle()
+    results = run_distributed(_tied_worker, world_size=2)
+    for sd in results:
+        for k, v in ref.items():
+            assert torch.allclose(v, sd[k], atol=2e-5, rtol=1e-4), (
+                k, (v - sd[k]).abs().max().item())
