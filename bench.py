@@ -43,6 +43,41 @@ def parse_args():
     return p.parse_args()
 
 
+_REPO = os.path.dirname(os.path.abspath(__file__))
+
+
+def setup_tunableop(model: str, local_rank: int) -> str:
+    """hipBLASLt GEMM algorithm selection via PyTorch TunableOp.
+
+    Replay mode (default): load the committed per-model tuning cache from
+    profiles/tunableop/<model>.csv so every hot GEMM shape runs its
+    offline-selected fastest hipBLASLt solution.
+    Tune mode (FENGSHEN_TUNE=1): search algorithms online and write the
+    cache to gpurun_out/tunableop_<model>_<rank>.csv for committing.
+    FENGSHEN_TUNABLEOP=0 disables both.
+    """
+    mode = "off"
+    if os.environ.get("FENGSHEN_TUNABLEOP", "1") == "0" \
+            or not torch.cuda.is_available():
+        return mode
+    import torch.cuda.tunable as tunable
+    cache = os.path.join(_REPO, "profiles", "tunableop", f"{model}.csv")
+    if os.environ.get("FENGSHEN_TUNE", "0") == "1":
+        out_dir = os.path.join(_REPO, "gpurun_out")
+        os.makedirs(out_dir, exist_ok=True)
+        tunable.enable(True)
+        tunable.tuning_enable(True)
+        tunable.set_filename(
+            os.path.join(out_dir, f"tunableop_{model}_{local_rank}.csv"))
+        mode = "tune"
+    elif os.path.exists(cache):
+        tunable.enable(True)
+        tunable.tuning_enable(False)
+        tunable.read_file(cache)
+        mode = "replay"
+    return mode
+
+
 def build_model(name: str, seq_len: int):
     if name == "ziya-llama-13b":
         from fengshen_amd.models.llama.configuration_llama import (
@@ -95,6 +130,8 @@ def main():
         from fengshen_amd.ops import has_ext
         assert has_ext(), \
             "HIP extension must be built (python -m fengshen_amd.ops.build)"
+
+    tunable_mode = setup_tunableop(args.model, local_rank)
 
     if world > 1 or args.tensor_model_parallel_size > 1:
         init_distributed(backend="gloo" if cpu_smoke else "nccl")
@@ -251,6 +288,7 @@ def main():
             "loss": round(float(loss.item()), 4),
             "tflops_per_gpu": round(tflops_per_gpu, 1),
             "mfu": round(mfu, 4),
+            "tunableop": tunable_mode,
             "config": {
                 "model": model_name,
                 "global_batch": b * world,
@@ -264,6 +302,10 @@ def main():
                 "ckpt_skip_interval": skip,
             },
         }), flush=True)
+
+    if tunable_mode == "tune":
+        import torch.cuda.tunable as tunable
+        tunable.write_file()
 
     if world > 1:
         dist.destroy_process_group()

@@ -114,8 +114,13 @@ class MegatronBertModel(MegatronBertPreTrainedModel):
         h = self.embeddings(input_ids, token_type_ids, position_ids)
         mask = None
         if attention_mask is not None:
-            # HF 1=keep [b,s] -> internal True=masked [b,1,1,s]
-            mask = (attention_mask == 0)[:, None, None, :]
+            if attention_mask.dim() == 3:
+                # per-sample 2D mask [b, sq, sk] (e.g. UniMC option
+                # isolation): 1=keep -> internal True=masked [b,1,sq,sk]
+                mask = (attention_mask == 0)[:, None, :, :]
+            else:
+                # HF 1=keep [b,s] -> internal True=masked [b,1,1,s]
+                mask = (attention_mask == 0)[:, None, None, :]
         skip = self.gradient_checkpointing_skip_interval
         for i, layer in enumerate(self.encoder):
             ckpt = self.gradient_checkpointing and self.training
@@ -225,8 +230,9 @@ class MegatronBertForMaskedLM(MegatronBertPreTrainedModel):
         self.bert.gradient_checkpointing_skip_interval = max(skip_interval, 0)
 
     def forward(self, input_ids, attention_mask=None, token_type_ids=None,
-                labels=None, return_dict=True, **_kw):
-        out = self.bert(input_ids, attention_mask, token_type_ids)
+                labels=None, position_ids=None, return_dict=True, **_kw):
+        out = self.bert(input_ids, attention_mask, token_type_ids,
+                        position_ids)
         logits_parallel = self.cls(out.last_hidden_state,
                                    self.bert.embeddings.word_embeddings.weight)
         loss = None

@@ -53,5 +53,26 @@ class InformationExtractionPipeline(BasePipeline):
             results.append(per_text)
         return results[0] if single else results
 
+    def predict(self, predict_data, batch_size: int = 8,
+                max_length: int = 128, threshold: float = 0.5):
+        """Reference UbertPipelines.predict (modeling_ubert.py:714-740):
+        task-typed items ({task_type, subtask_type, text, choices}) are
+        bucketed by choice count and decoded into entity structures."""
+        from fengshen_amd.models.ubert.modeling_ubert import UbertExtractor
+        extractor = UbertExtractor(self.model, self.tokenizer,
+                                   max_length=max_length,
+                                   threshold=threshold)
+        result = []
+        start = 0
+        while start < len(predict_data):
+            batch = predict_data[start:start + batch_size]
+            start += batch_size
+            buckets = {}
+            for item in batch:
+                buckets.setdefault(len(item["choices"]), []).append(item)
+            for _n, items in buckets.items():
+                result.extend(extractor.extract(items))
+        return result
+
 
 Pipeline = InformationExtractionPipeline

@@ -23,17 +23,62 @@ class SimpleCharTokenizer:
         self.bos_token_id = 5
         self.eos_token_id = 6
 
+        self.pad_token = "[PAD]"
+        self.cls_token = "[CLS]"
+        self.sep_token = "[SEP]"
+        self.mask_token = "[MASK]"
+        self.unk_token = "[UNK]"
+
     def get_vocab(self):
         return self._vocab
 
     def __len__(self):
         return len(self._vocab)
 
+    @property
+    def vocab_size(self):
+        return len(self._vocab)
+
+    def _tokenize_ids(self, text: str) -> List[int]:
+        """Char-level with special-token substrings kept whole."""
+        ids: List[int] = []
+        i = 0
+        specials = ("[PAD]", "[CLS]", "[SEP]", "[MASK]", "[UNK]")
+        while i < len(text):
+            matched = False
+            if text[i] == "[":
+                for sp in specials:
+                    if text.startswith(sp, i):
+                        ids.append(self._vocab[sp])
+                        i += len(sp)
+                        matched = True
+                        break
+            if not matched:
+                ids.append(self._vocab.get(text[i], self.unk_token_id))
+                i += 1
+        return ids
+
     def encode(self, text: str, add_special_tokens: bool = True) -> List[int]:
-        ids = [self._vocab.get(c, self.unk_token_id) for c in text]
+        ids = self._tokenize_ids(text)
         if add_special_tokens:
             return [self.cls_token_id] + ids + [self.sep_token_id]
         return ids
+
+    def encode_plus(self, text: str, max_length: int = 512,
+                    padding: str = "max_length",
+                    truncation="longest_first", **_kw):
+        ids = self.encode(text)
+        if truncation and len(ids) > max_length:
+            ids = ids[:max_length - 1] + [self.sep_token_id]
+        attn = [1] * len(ids)
+        tok_type = [0] * len(ids)
+        if padding == "max_length" and len(ids) < max_length:
+            pad = max_length - len(ids)
+            ids = ids + [self.pad_token_id] * pad
+            attn = attn + [0] * pad
+            tok_type = tok_type + [0] * pad
+        return {"input_ids": ids, "attention_mask": attn,
+                "token_type_ids": tok_type}
 
     def decode(self, ids, skip_special_tokens: bool = True) -> str:
         out = []

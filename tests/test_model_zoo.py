@@ -201,3 +201,160 @@ def test_zen_task_heads():
             end_positions=torch.tensor([3, 4]))
     assert out.start_logits.shape == (2, 16)
     assert out.loss.isfinite()
+
+
+def test_unimc_option_isolation_mask():
+    """Ref get_att_mask (modeling_unimc.py:92-112): options attend only to
+    themselves + question/text, never to each other."""
+    import numpy as np
+    from fengshen_amd.models.unimc.modeling_unimc import UniMCEncoder
+    am = np.ones(12, dtype=np.int64)
+    # [CLS] | opt A = idx 1..3 | opt B = idx 4..6 | text 7..11
+    label_idx = [1, 4, 7]
+    att = UniMCEncoder.get_att_mask(am, label_idx, question_len=1)
+    assert att.shape == (12, 12)
+    assert att[1:4, 4:7].sum() == 0      # A does not see B
+    assert att[4:7, 1:4].sum() == 0      # B does not see A
+    assert (att[1:4, 1:4] == 1).all()    # A sees itself
+    assert (att[1:4, 7:] == 1).all()     # A sees text
+    assert (att[8, :] == 1).all()        # text row sees everything
+
+
+def test_unimc_position_ids_restart_per_option():
+    from fengshen_amd.models.unimc.modeling_unimc import UniMCEncoder
+    pos = UniMCEncoder.get_position_ids([1, 4, 7], 16, question_len=1)
+    assert pos[:7] == [0, 1, 2, 3, 1, 2, 3]  # option positions restart
+    assert pos[7] == 4                        # text continues after max
+
+
+def test_unimc_isolation_changes_logits():
+    """With the isolation mask, changing option B's tokens must NOT move
+    option A's anchor logit (1-layer model); with a plain 1D mask it does."""
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.models.unimc.modeling_unimc import UniMCEncoder, UniMCModel
+    import numpy as np
+    torch.manual_seed(0)
+    cfg = bert_tiny_config()
+    cfg.num_hidden_layers = 1
+    cfg.hidden_dropout = 0.0
+    cfg.attention_dropout = 0.0
+    m = UniMCModel(cfg, yes_token_id=5).eval()
+
+    label_idx = [1, 4, 7]
+    ids = torch.randint(3, 256, (1, 12))
+    ids2 = ids.clone()
+    ids2[0, 4:6] = (ids2[0, 4:6] + 7) % 250 + 3  # perturb option B only
+    att = torch.tensor(UniMCEncoder.get_att_mask(
+        np.ones(12, dtype=np.int64), label_idx, 1)).float().unsqueeze(0)
+    opt = torch.tensor([[1, 4]])
+    with torch.no_grad():
+        la = m(ids, attention_mask=att, option_positions=opt).cls_logits
+        lb = m(ids2, attention_mask=att, option_positions=opt).cls_logits
+        pa = m(ids, attention_mask=torch.ones(1, 12),
+               option_positions=opt).cls_logits
+        pb = m(ids2, attention_mask=torch.ones(1, 12),
+               option_positions=opt).cls_logits
+    assert torch.allclose(la[0, 0], lb[0, 0], atol=1e-5)   # isolated
+    assert not torch.allclose(pa[0, 0], pb[0, 0], atol=1e-5)  # plain leaks
+
+
+def test_unimc_reference_style_forward():
+    """clslabels = anchor position; cls CE over positions + MLM aux loss."""
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.models.unimc.modeling_unimc import UniMCModel
+    torch.manual_seed(0)
+    m = UniMCModel(bert_tiny_config(), yes_token_id=5)
+    b, s = 2, 14
+    ids = torch.randint(3, 256, (b, s))
+    clsmask = torch.full((b, s), -10000.0)
+    clsmask[:, [1, 4]] = 0.0
+    clslabels = torch.tensor([1, 4])
+    mlml = ids.clone()
+    mlml[:, 7:] = -100
+    out = m(ids, mlmlabels=mlml, clslabels=clslabels, clslabels_mask=clsmask)
+    assert out.loss.isfinite()
+    assert out.cls_logits.shape == (b, s)
+    # masked positions cannot win
+    assert (out.cls_logits.argmax(-1) < 5).all()
+    out.loss.backward()
+
+
+def test_ubert_reference_loss_and_mask():
+    """Additive span_labels_mask + 10*(100*BCE + soft1 + soft2) loss
+    (ref modeling_ubert.py:298-309)."""
+    from fengshen_amd.models.ubert.modeling_ubert import UbertModel
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    torch.manual_seed(0)
+    m = UbertModel(bert_tiny_config())
+    b, nl, s = 2, 3, 10
+    ids = torch.randint(3, 256, (b, nl, s))
+    labels = torch.zeros(b, nl, s, s)
+    labels[:, :, 2, 4] = 1
+    slm = torch.full((b, nl, s, s), -10000.0)
+    slm[:, :, 2:, 2:] = 0.0
+    out = m(ids, span_labels=labels, span_labels_mask=slm)
+    assert out.loss.isfinite()
+    out.loss.backward()
+    # masked cells are pushed to -10000: sigmoid ~ 0
+    assert (out.span_logits[:, :, 0, 0].sigmoid() < 1e-3).all()
+
+
+def test_ubert_encoder_schema():
+    """UbertEncoder builds prompt rows + span labels from char-level
+    entity_idx (ref UbertDataset.encode :56-190)."""
+    from fengshen_amd.models.ubert.modeling_ubert import UbertEncoder
+    from fengshen_amd.tokenizer import SimpleCharTokenizer
+    tk = SimpleCharTokenizer()
+    enc = UbertEncoder(tk, max_length=48, num_labels=4)
+    item = {
+        "task_type": "抽取任务", "subtask_type": "实体识别",
+        "text": "abc def",
+        "choices": [
+            {"entity_type": "x", "entity_list": [{"entity_idx": [[0, 2]]}]},
+            {"entity_type": "y", "entity_list": []},
+        ],
+    }
+    s = enc.encode(item)
+    assert s["input_ids"].shape == (4, 48)
+    assert s["span_labels"].shape == (4, 48, 48)
+    # positive row exists: entity "abc" chars 0..2 -> token positions
+    qlen = len(tk.encode("抽取任务[SEP]实体识别[SEP]x"))
+    assert s["span_labels"][0, qlen, qlen + 2] == 1
+    # masked region: question cells are -10000
+    assert s["span_labels_mask"][0, 0, 0] == -10000.0
+    assert s["span_labels_mask"][0, qlen, qlen] == 0.0
+
+
+def test_ubert_extractor_decode():
+    """UbertExtractor returns entity structures with entity_name text
+    (ref extractModel.extract :486-675)."""
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.models.ubert.modeling_ubert import (
+        UbertExtractor, UbertModel)
+    from fengshen_amd.tokenizer import SimpleCharTokenizer
+    torch.manual_seed(0)
+    tk = SimpleCharTokenizer()
+    m = UbertModel(bert_tiny_config()).eval()
+    ex = UbertExtractor(m, tk, max_length=32, threshold=0.0)
+    items = [{"task_type": "抽取任务", "subtask_type": "实体识别",
+              "text": "ab cd", "choices": [{"entity_type": "x"}]},
+             {"task_type": "分类任务", "subtask_type": "情感分析",
+              "text": "ab", "choices": [{"entity_type": "好"},
+                                        {"entity_type": "坏"}]}]
+    out = ex.extract([items[0]])
+    assert "entity_list" in out[0]["choices"][0]
+    for e in out[0]["choices"][0]["entity_list"]:
+        assert isinstance(e["entity_name"], str) and "score" in e
+    out2 = ex.extract([items[1]])
+    assert any(c.get("label") == 1 for c in out2[0]["choices"])
+
+
+def test_ubert_offset_mapping():
+    from fengshen_amd.models.ubert.modeling_ubert import OffsetMapping
+    mapping = OffsetMapping().rematch("Hello ab", list("hello ab"))
+    assert mapping[0] == [0]
+    assert mapping[6] == [6]
