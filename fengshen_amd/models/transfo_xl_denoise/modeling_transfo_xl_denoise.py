@@ -97,7 +97,9 @@ class _XlLayer(nn.Module):
                                init_method=im, output_init_method=om)
 
     def forward(self, x, mem=None):
-        x = x + self.attn(self.ln1(x), mem)
+        # memory stores pre-LN hiddens; normalize at use (ref :460)
+        m = self.ln1(mem) if mem is not None else None
+        x = x + self.attn(self.ln1(x), m)
         return x + self.mlp(self.ln2(x))
 
 
@@ -141,8 +143,15 @@ class TransfoXLDenoiseModel(PreTrainedModel):
         h = self.wte(input_ids) + self.wpe(pos)
         new_mems = []
         for i, layer in enumerate(self.layers):
-            new_mems.append(h.detach()[:, -self.mem_len:])
-            h = layer(h, mems[i] if mems else None)
+            old = mems[i] if mems else None
+            # accumulate memory across segments (ref update_mems :649-662):
+            # concat old memory with this segment, keep the last mem_len
+            if old is not None:
+                new_mems.append(
+                    torch.cat([old, h.detach()], dim=1)[:, -self.mem_len:])
+            else:
+                new_mems.append(h.detach()[:, -self.mem_len:])
+            h = layer(h, old)
         h = self.ln_f(h)
         logits = h @ self.wte.weight.t().to(h.dtype)
         loss = None

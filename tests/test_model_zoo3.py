@@ -100,3 +100,81 @@ def test_deltalm_interleaved_decoder():
     m.eval()
     gen = m.generate(src, max_new_tokens=4, do_sample=False)
     assert gen.shape[0] == 2
+
+
+# ---------------------------------------------------------------------------
+# Transfo-XL paraphrase / reasoning variants + sampling helpers
+# (ref models/transfo_xl_{paraphrase,reasoning}, utils/transfo_xl_utils.py)
+# ---------------------------------------------------------------------------
+def _tiny_xl():
+    from fengshen_amd.models.transfo_xl_denoise.modeling_transfo_xl_denoise \
+        import TransfoXLDenoiseConfig, TransfoXLDenoiseModel
+    torch.manual_seed(0)
+    cfg = TransfoXLDenoiseConfig(
+        vocab_size=120, hidden_size=32, num_hidden_layers=2,
+        num_attention_heads=4, intermediate_size=64,
+        max_position_embeddings=128, mem_len=64)
+    return TransfoXLDenoiseModel(cfg).eval()
+
+
+def test_transfo_xl_mems_accumulate():
+    """Incremental decode must see the FULL history through mems
+    (ref update_mems :649-662: concat + truncate)."""
+    m = _tiny_xl()
+    ids = torch.randint(3, 100, (1, 10))
+    full = m(input_ids=ids)
+    # token-by-token with mems
+    out = m(input_ids=ids[:, :5])
+    mems = out.mems
+    assert mems[0].shape[1] == 5
+    for t in range(5, 10):
+        out = m(input_ids=ids[:, t:t + 1], mems=mems)
+        mems = out.mems
+    assert mems[0].shape[1] == 10  # accumulated, not reset to 1
+    assert torch.allclose(out.logits[0, -1], full.logits[0, -1], atol=1e-4)
+
+
+def test_top_k_logits_filtering():
+    from fengshen_amd.utils.transfo_xl_utils import top_k_logits
+    logits = torch.tensor([[1.0, 5.0, 3.0, 0.5]])
+    k = top_k_logits(logits.clone(), top_k=2)
+    assert k[0, 1] == 5.0 and k[0, 2] == 3.0
+    assert k[0, 0] == -float("inf") and k[0, 3] == -float("inf")
+    p = top_k_logits(logits.clone(), top_p=0.6)
+    assert p[0, 1] == 5.0          # best token always kept
+    assert p[0, 3] == -float("inf")
+
+
+def test_sample_sequence_batch_ragged_prompts():
+    """Ragged prompts: shorter sequences keep copying their real prompt
+    tokens (switch) until consumed; outputs restore input order."""
+    from fengshen_amd.utils.transfo_xl_utils import sample_sequence_batch
+    m = _tiny_xl()
+    torch.manual_seed(1)
+    prompts = torch.randint(3, 100, (2, 8))
+    lengths = torch.tensor([8, 4])
+    outs, probs = sample_sequence_batch(
+        m, prompts, lengths, max_out_seq=6, end_token_id=119, top_p=0.9)
+    assert len(outs) == 2 and len(probs) == 2
+    assert outs[0][:8] == prompts[0].tolist()  # full prompt preserved
+    assert outs[1][:4] == prompts[1, :4].tolist()
+
+
+def test_paraphrase_and_reasoning_generate():
+    from fengshen_amd.models.transfo_xl_paraphrase import paraphrase_generate
+    from fengshen_amd.models.transfo_xl_reasoning import (
+        abduction_generate, deduction_generate, en_to_zh)
+    from fengshen_amd.tokenizer import SimpleCharTokenizer
+    m = _tiny_xl()
+    tk = SimpleCharTokenizer()
+    torch.manual_seed(2)
+    res = paraphrase_generate(m, tk, "天气很好", max_out_seq=8,
+                              eod_token=119)
+    assert isinstance(res, str)
+    outs = deduction_generate(m, tk, ["下雨了"], max_out_seq=6,
+                              end_token_id=119)
+    assert len(outs) == 1 and isinstance(outs[0], str)
+    outs = abduction_generate(m, tk, "地面湿了", max_out_seq=6,
+                              end_token_id=119)
+    assert len(outs) == 1
+    assert en_to_zh("a,b.") == "a，b。"
