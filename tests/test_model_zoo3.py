@@ -111,7 +111,7 @@ def _tiny_xl():
         import TransfoXLDenoiseConfig, TransfoXLDenoiseModel
     torch.manual_seed(0)
     cfg = TransfoXLDenoiseConfig(
-        vocab_size=120, hidden_size=32, num_hidden_layers=2,
+        vocab_size=300, hidden_size=32, num_hidden_layers=2,
         num_attention_heads=4, intermediate_size=64,
         max_position_embeddings=128, mem_len=64)
     return TransfoXLDenoiseModel(cfg).eval()
