@@ -339,3 +339,61 @@ def test_tagging_collators_feed_heads():
     assert out2.loss.isfinite()
     assert bia_batch["span_labels"][0, 1, 2] == 2       # entity cell
     assert bia_batch["span_labels"][0, 2, 1] == -100    # below diagonal
+
+
+# ---------------------------------------------------------------------------
+# t5_gen_datasets: knowledge-grounded dialog dataset
+# (ref data/t5_dataloader/t5_gen_datasets.py)
+# ---------------------------------------------------------------------------
+def test_dialog_dataset_layout():
+    from fengshen_amd.data.t5_gen_datasets import (
+        DialogCollator, DialogDataset, add_dialog_special_tokens)
+    from fengshen_amd.tokenizer import SimpleCharTokenizer
+
+    class Tok(SimpleCharTokenizer):
+        def add_special_tokens(self, d):
+            for t in d["additional_special_tokens"]:
+                self._vocab.setdefault(t, len(self._vocab))
+            self._inv = {v: k for k, v in self._vocab.items()}
+
+        def convert_tokens_to_ids(self, t):
+            return self._vocab.get(t, self.unk_token_id)
+
+    tk = add_dialog_special_tokens(Tok())
+    data = [{"context": ["你好", "你好啊", "天气如何"],
+             "knowledge": "今天晴",
+             "target": "天气很好"}]
+    ds = DialogDataset(data, tk, max_seq_length=64,
+                       max_knowledge_length=16, max_target_length=8,
+                       eos_token_id=tk.eos_token_id)
+    s = ds[0]
+    kn_start = tk.convert_tokens_to_ids("[KNSTART]")
+    kn_end = tk.convert_tokens_to_ids("[KNEND]")
+    ct_start = tk.convert_tokens_to_ids("[CTSTART]")
+    ids = s["input_ids"].tolist()
+    assert ids[0] == ct_start and kn_start in ids and ids[-1] == kn_end
+    # knowledge region typed 2, context alternates 0/1
+    tt = s["token_types"].tolist()
+    kn_i = ids.index(kn_start)
+    assert all(t == 2 for t in tt[kn_i:])
+    assert set(tt[:kn_i]) <= {0, 1}
+    assert len(ids) == len(tt) == len(s["attention_mask"])
+    assert s["labels"][-1] == tk.eos_token_id
+
+
+def test_dialog_collator_shift_right():
+    import numpy as np
+    from fengshen_amd.data.t5_gen_datasets import (
+        DialogCollator, shift_tokens_right)
+    labels = np.array([[7, 8, 9, -100], [5, 6, -100, -100]])
+    shifted = shift_tokens_right(labels, pad_token_id=0,
+                                 decoder_start_token_id=2)
+    assert shifted.tolist() == [[2, 7, 8, 9], [2, 5, 6, 0]]
+    samples = [{"input_ids": [3, 4], "token_types": [0, 0],
+                "attention_mask": [1, 1], "labels": [7, 8]},
+               {"input_ids": [3, 4, 5], "token_types": [0, 0, 1],
+                "attention_mask": [1, 1, 1], "labels": [9]}]
+    batch = DialogCollator(pad_token_id=0, decoder_start_token_id=2)(samples)
+    assert batch["input_ids"].shape == (2, 3)
+    assert batch["labels"][1].tolist() == [9, -100]
+    assert batch["decoder_input_ids"][0].tolist() == [2, 7]
