@@ -103,6 +103,7 @@ class Trainer:
         ckpt_skip_interval: int = 0,
         zero_bucket_numel: int = 128 * 1024 * 1024,
         overlap_comm: bool = True,
+        logger=None,
         **_unused,
     ):
         s = parse_strategy(strategy)
@@ -143,6 +144,14 @@ class Trainer:
         self._metrics: Dict[str, float] = {}
         self._did_step = False
         self._csv = None
+        # experiment loggers (TensorBoardLogger / WandbLogger / custom);
+        # CSV always written in addition (ref: PL logger= semantics)
+        if logger is None:
+            self.loggers = []
+        elif isinstance(logger, (list, tuple)):
+            self.loggers = list(logger)
+        else:
+            self.loggers = [logger]
 
     # ------------------------------------------------------------------
     @property
@@ -229,6 +238,8 @@ class Trainer:
         if self.global_rank == 0 and self._metrics:
             if self._csv is not None:
                 self._csv.log(self.global_step, self._metrics)
+            for lg in self.loggers:
+                lg.log_metrics(self._metrics, self.global_step)
             msg = " | ".join(
                 f"{k} {v:.6g}" if isinstance(v, float) else f"{k} {v}"
                 for k, v in self._metrics.items())
@@ -245,6 +256,12 @@ class Trainer:
         self.strategy.setup_environment(self)
         seed_everything(self.seed)
         self._csv = CSVLogger(self.default_root_dir, self.global_rank)
+        if self.global_rank == 0:
+            hp = getattr(model, "hparams", None)
+            if hp:
+                for lg in self.loggers:
+                    lg.log_hyperparams(dict(hp) if not isinstance(hp, dict)
+                                       else hp)
         self.module = model
         model.trainer = self
         self.datamodule = datamodule
@@ -435,6 +452,8 @@ class Trainer:
                 model.train()
             self.current_epoch += 1
         self._flush_logs()
+        for lg in self.loggers:
+            lg.finalize()
 
     def _clip_and_step(self, model):
         from fengshen_amd.parallel.zero import ZeroOptimizer
