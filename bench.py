@@ -30,7 +30,8 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--model", type=str, default="ziya-llama-13b",
                    choices=["ziya-llama-13b", "wenzhong-gpt2-3.5b",
-                            "erlangshen-1.3b", "llama-tiny"])
+                            "erlangshen-1.3b", "llama-tiny", "taiyi-sd"])
+    p.add_argument("--image_size", type=int, default=512)
     p.add_argument("--seq_len", type=int, default=2048)
     p.add_argument("--micro_batch", type=int, default=16)
     p.add_argument("--zero_stage", type=int, default=3)
@@ -78,7 +79,61 @@ def setup_tunableop(model: str, local_rank: int) -> str:
     return mode
 
 
+class _TaiyiSDTrainStep(torch.nn.Module):
+    """Taiyi-SD-1B finetune step (BASELINE config 5): frozen VAE encode ->
+    noise/timestep -> BERT text encode -> SD-1 UNet -> mse.  Trains UNet +
+    text encoder (the bilingual stage-2 recipe)."""
+
+    def __init__(self):
+        super().__init__()
+        from fengshen_amd.models.megatron_bert.configuration_megatron_bert \
+            import bert_tiny_config
+        from fengshen_amd.models.megatron_bert.modeling_megatron_bert \
+            import MegatronBertModel
+        from fengshen_amd.models.taiyi_sd import (
+            AutoencoderKL, DDPMScheduler, UNet2DConditionModel)
+        from fengshen_amd.models.taiyi_sd.unet import taiyi_sd_1b_config
+        self.text_encoder = MegatronBertModel(
+            bert_tiny_config(hidden_size=768, num_hidden_layers=12,
+                             num_attention_heads=12,
+                             intermediate_size=3072,
+                             max_position_embeddings=512),
+            add_pooling_layer=False)
+        self.vae = AutoencoderKL()
+        for p in self.vae.parameters():
+            p.requires_grad = False
+        self.unet = UNet2DConditionModel(taiyi_sd_1b_config())
+        self.noise_scheduler = DDPMScheduler()
+
+    class _Out:
+        def __init__(self, loss):
+            self.loss = loss
+
+    @property
+    def config(self):
+        return self.unet.config
+
+    def gradient_checkpointing_enable(self, **_kw):
+        self.unet.gradient_checkpointing_enable()
+
+    def forward(self, pixel_values, input_ids):
+        with torch.no_grad():
+            latents = self.vae.encode(pixel_values.to(
+                self.unet.conv_in.weight.dtype))
+        noise = torch.randn_like(latents)
+        t = torch.randint(0, self.noise_scheduler.num_train_timesteps,
+                          (latents.shape[0],), device=latents.device)
+        noisy = self.noise_scheduler.add_noise(latents, noise, t)
+        ctx = self.text_encoder(input_ids).last_hidden_state
+        pred = self.unet(noisy, t, ctx)
+        loss = torch.nn.functional.mse_loss(pred.float(), noise.float())
+        return self._Out(loss)
+
+
 def build_model(name: str, seq_len: int):
+    if name == "taiyi-sd":
+        m = _TaiyiSDTrainStep()
+        return m, m.text_encoder.config.vocab_size, "Taiyi-SD-1B"
     if name == "ziya-llama-13b":
         from fengshen_amd.models.llama.configuration_llama import (
             ziya_llama_13b_config)
@@ -142,12 +197,15 @@ def main():
     torch.manual_seed(1234)
     # construct directly on the GPU in bf16: 8 ranks x fp32-on-CPU would
     # exhaust host RAM for 13B, and GPU-side init is much faster
-    torch.set_default_dtype(torch.bfloat16)
+    # cpu_smoke stays fp32 (CPU group_norm/conv kernels lack bf16 paths)
+    bench_dtype = torch.float32 if cpu_smoke else torch.bfloat16
+    torch.set_default_dtype(bench_dtype)
     with device:
         model, vocab, model_name = build_model(args.model, args.seq_len)
     torch.set_default_dtype(torch.float32)
-    model = model.to(torch.bfloat16).to(device)
-    if hasattr(model, "gradient_checkpointing_enable"):
+    model = model.to(bench_dtype).to(device)
+    if hasattr(model, "gradient_checkpointing_enable") \
+            and args.model != "taiyi-sd":
         try:
             model.gradient_checkpointing_enable(
                 skip_interval=max(args.ckpt_skip, 0))
@@ -178,7 +236,9 @@ def main():
     _ACT_MULT = {"ziya-llama-13b": 21.2, "llama-tiny": 21.2,
                  "wenzhong-gpt2-3.5b": 21.0, "erlangshen-1.3b": 19.0}
     skip = max(args.ckpt_skip, 0)
-    if args.ckpt_skip < 0 and torch.cuda.is_available():
+    if args.model == "taiyi-sd":
+        skip = 0  # conv UNet at 512^2 fits 288 GB HBM without recompute
+    elif args.ckpt_skip < 0 and torch.cuda.is_available():
         free_b, _total = torch.cuda.mem_get_info()
         L = model.config.num_hidden_layers
         act = _ACT_MULT[args.model] * model.config.hidden_size \
@@ -200,7 +260,10 @@ def main():
             skip = 0
 
     is_bert = args.model == "erlangshen-1.3b"
+    is_sd = args.model == "taiyi-sd"
     b, s = args.micro_batch, args.seq_len
+    if is_sd:
+        s = 77  # CLIP-style caption length
     # TP ranks must see identical batches: seed by DP rank
     g = torch.Generator(device="cpu").manual_seed(42 + dp_rank)
     # one fresh synthetic batch per step (a fixed 4-batch cycle lets a 13B
@@ -208,7 +271,11 @@ def main():
     batches = []
     for _ in range(args.warmup + args.steps):
         ids = torch.randint(3, vocab, (b, s), generator=g).to(device)
-        if is_bert:
+        if is_sd:
+            px = torch.randn(b, 3, args.image_size, args.image_size,
+                             generator=g).to(device)
+            batches.append(dict(pixel_values=px, input_ids=ids))
+        elif is_bert:
             labels = ids.clone()
             mask_pos = torch.rand(b, s, generator=g) < 0.15
             labels[~mask_pos.to(device)] = -100
@@ -254,7 +321,7 @@ def main():
 
     tokens_per_step = b * s * dp_world
     samples_per_step = b * dp_world
-    value = (samples_per_step if is_bert else tokens_per_step) \
+    value = (samples_per_step if (is_bert or is_sd) else tokens_per_step) \
         * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
@@ -266,16 +333,20 @@ def main():
     flops_step = 6.0 * n_params * b * s + 12.0 * L * h * s * s * b
     tflops_per_gpu = flops_step * args.steps / elapsed / 1e12
     mfu = tflops_per_gpu / 2500.0  # MI355X dense bf16 peak ~2.5 PF/s
+    if is_sd:  # token-based MFU formula does not apply to the conv UNet
+        tflops_per_gpu, mfu = None, None
 
     if rank == 0:
         print(json.dumps({
             "metric": ("samples/sec/node pretrain Erlangshen-1.3B"
                        if is_bert
+                       else "samples/sec Taiyi-SD-1B 512x512 finetune"
+                       if is_sd
                        else f"tokens/sec Ziya-LLaMA-13B ZeRO-{args.zero_stage}"
                        if args.model == "ziya-llama-13b"
                        else f"tokens/sec {model_name}"),
             "value": round(value, 2),
-            "unit": "samples/s" if is_bert else "tokens/s",
+            "unit": "samples/s" if (is_bert or is_sd) else "tokens/s",
             "n_gpus": world,
             "steps": args.steps,
             "warmup": args.warmup,
@@ -283,17 +354,19 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "bf16" if not cpu_smoke else "fp32(cpu-smoke)",
             "data": "synthetic",
             "loss": round(float(loss.item()), 4),
-            "tflops_per_gpu": round(tflops_per_gpu, 1),
-            "mfu": round(mfu, 4),
+            "tflops_per_gpu": (round(tflops_per_gpu, 1)
+                               if tflops_per_gpu is not None else None),
+            "mfu": round(mfu, 4) if mfu is not None else None,
             "tunableop": tunable_mode,
             "config": {
                 "model": model_name,
                 "global_batch": b * world,
                 "micro_batch": b,
                 "seq_len": s,
+                **({"image_size": args.image_size} if is_sd else {}),
                 "parallelism": (
                     f"zero{args.zero_stage}_dp{dp_world}"
                     + (f"_tp{args.tensor_model_parallel_size}"

@@ -31,7 +31,10 @@ from fengshen_amd.models.taiyi_sd import (
     DDPMScheduler,
     UNet2DConditionModel,
 )
-from fengshen_amd.models.taiyi_sd.unet import UNetConfig, unet_tiny_config
+from fengshen_amd.models.taiyi_sd.unet import (
+    taiyi_sd_1b_config,
+    unet_tiny_config,
+)
 from fengshen_amd.trainer.callbacks import ThroughputMonitor
 from fengshen_amd.utils.universal_checkpoint import UniversalCheckpoint
 from fengshen_amd.utils.utils import report_memory
@@ -45,14 +48,13 @@ class TaiyiSD(FengshenModule):
             text_cfg = bert_tiny_config()
             unet_cfg = unet_tiny_config()
         else:
+            # Taiyi-SD-1B: BERT text tower (768 hidden) + the SD-1 UNet
+            # (860M, diffusers weight layout)
             text_cfg = bert_tiny_config(hidden_size=768,
                                         num_hidden_layers=12,
                                         num_attention_heads=12,
                                         intermediate_size=3072)
-            unet_cfg = UNetConfig(block_channels=(320, 640, 1280),
-                                  layers_per_block=2,
-                                  num_attention_heads=8,
-                                  cross_attention_dim=768)
+            unet_cfg = taiyi_sd_1b_config()
         self.text_encoder = MegatronBertModel(text_cfg,
                                               add_pooling_layer=False)
         self.vae = AutoencoderKL()
@@ -86,6 +88,19 @@ class TaiyiSD(FengshenModule):
 
     def configure_optimizers(self):
         return configure_optimizers(self)
+
+    def on_save_checkpoint(self, checkpoint):
+        """Export HF pipeline layout (ref finetune.py:154-158): one subdir
+        per component, loadable with from_pretrained."""
+        if self.global_rank == 0:
+            root = os.path.join(
+                self.hparams.default_root_dir,
+                f"hf_out_{self.trainer.current_epoch}"
+                f"_{self.trainer.global_step}")
+            self.unet.save_pretrained(os.path.join(root, "unet"))
+            self.vae.save_pretrained(os.path.join(root, "vae"))
+            self.text_encoder.save_pretrained(
+                os.path.join(root, "text_encoder"))
 
 
 class _SDCollator:

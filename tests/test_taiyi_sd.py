@@ -70,3 +70,54 @@ def test_sd_training_step_end_to_end():
     assert loss.isfinite()
     loss.backward()
     assert text_encoder.embeddings.word_embeddings.weight.grad is not None
+
+
+def test_unet_diffusers_weight_layout():
+    """state_dict keys follow diffusers UNet2DConditionModel naming so
+    real SD-1/Taiyi checkpoints map 1:1."""
+    from fengshen_amd.models.taiyi_sd.unet import (
+        UNet2DConditionModel, unet_tiny_config)
+    m = UNet2DConditionModel(unet_tiny_config())
+    keys = set(m.state_dict().keys())
+    expected = [
+        "conv_in.weight",
+        "time_embedding.linear_1.weight",
+        "time_embedding.linear_2.bias",
+        "down_blocks.0.resnets.0.norm1.weight",
+        "down_blocks.0.resnets.0.time_emb_proj.weight",
+        "down_blocks.0.attentions.0.proj_in.weight",
+        "down_blocks.0.attentions.0.transformer_blocks.0.attn1.to_q.weight",
+        "down_blocks.0.attentions.0.transformer_blocks.0.attn2.to_k.weight",
+        "down_blocks.0.attentions.0.transformer_blocks.0.attn2.to_out.0.bias",
+        "down_blocks.0.attentions.0.transformer_blocks.0.ff.net.0.proj.weight",
+        "down_blocks.0.attentions.0.transformer_blocks.0.ff.net.2.weight",
+        "down_blocks.0.downsamplers.0.conv.weight",
+        "mid_block.resnets.0.conv1.weight",
+        "mid_block.attentions.0.transformer_blocks.0.norm3.weight",
+        "up_blocks.0.resnets.0.conv_shortcut.weight",
+        "up_blocks.0.upsamplers.0.conv.weight",
+        "conv_norm_out.weight",
+        "conv_out.bias",
+    ]
+    for k in expected:
+        assert k in keys, k
+    # cross-attn to_q/k/v have no bias (diffusers)
+    assert ("down_blocks.0.attentions.0.transformer_blocks.0.attn1.to_q.bias"
+            not in keys)
+
+
+def test_unet_sd1b_param_count():
+    """taiyi_sd_1b_config builds the SD-1 UNet shape (~860M params)."""
+    from fengshen_amd.models.taiyi_sd.unet import (
+        UNet2DConditionModel, taiyi_sd_1b_config)
+    with torch.device("meta"):
+        m = UNet2DConditionModel(taiyi_sd_1b_config())
+    n = sum(p.numel() for p in m.parameters())
+    assert 820e6 < n < 900e6, n
+    # 12 skip connections on the down path: 1 conv_in + 4 blocks x 2 res
+    # + 3 downsamplers
+    n_res_down = sum(len(b.resnets) for b in m.down_blocks)
+    n_ds = sum(1 for b in m.down_blocks if b.downsamplers is not None)
+    assert 1 + n_res_down + n_ds == 12
+    n_res_up = sum(len(b.resnets) for b in m.up_blocks)
+    assert n_res_up == 12  # consumes every skip
