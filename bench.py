@@ -354,7 +354,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if not cpu_smoke else "fp32(cpu-smoke)",
+            "dtype": "bf16",
             "data": "synthetic",
             "loss": round(float(loss.item()), 4),
             "tflops_per_gpu": (round(tflops_per_gpu, 1)
