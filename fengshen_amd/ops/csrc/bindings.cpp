@@ -48,10 +48,12 @@ void fs_bias_gelu(const void*, const void*, const void*, void*, long, int, int,
 void fs_fused_adamw(float*, const void*, float*, float*, void*, long, float,
                     float, float, float, float, int, int, int, hipStream_t);
 void fs_flash_attn_fwd(const void*, const void*, const void*, void*, float*,
-                       int, int, int, float, hipStream_t);
+                       const int*, int, int, int, int, int, int, float,
+                       float, unsigned long long, hipStream_t);
 void fs_flash_attn_bwd(const void*, const void*, const void*, const void*,
                        const void*, const float*, void*, void*, void*, float*,
-                       int, int, int, float, hipStream_t);
+                       const int*, int, int, int, int, int, int, float,
+                       float, unsigned long long, hipStream_t);
 void fs_w8_gemv(const void*, const float*, const void*, void*, int, int, int,
                 hipStream_t);
 }
@@ -265,36 +267,67 @@ static void fused_adamw(at::Tensor master, at::Tensor grad, at::Tensor m,
                  outp ? fs_dtype(out_param) : FS_F32, cur_stream());
 }
 
-// q,k,v: [b, h, s, 128] bf16 contiguous, causal, seq %% 64 == 0
-static std::vector<at::Tensor> flash_attn_fwd(at::Tensor q, at::Tensor k,
-                                              at::Tensor v, double scale) {
+static const int kFlashDims[] = {40, 64, 80, 96, 128, 160};
+
+static bool flash_dim_ok(int d) {
+  for (int x : kFlashDims)
+    if (x == d) return true;
+  return false;
+}
+
+// q: [b, h, sq, d] bf16 contiguous; k,v: [b, h, sk, d].
+// causal: sq == sk, sq % 64 == 0, klens must be absent.
+// non-causal: cross-attention (sq != sk) and ragged sk allowed; optional
+// klens int32 [b] = per-batch key length (suffix padding mask).
+static std::vector<at::Tensor> flash_attn_fwd(
+    at::Tensor q, at::Tensor k, at::Tensor v, double scale, bool causal,
+    c10::optional<at::Tensor> klens, double drop_p, int64_t seed) {
   TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "flash_attn: bf16 only");
-  TORCH_CHECK(q.dim() == 4 && q.size(3) == 128, "flash_attn: head_dim 128");
-  const int b = q.size(0), h = q.size(1), s = q.size(2);
-  TORCH_CHECK(s % 64 == 0, "flash_attn: seq must be divisible by 64");
+  TORCH_CHECK(q.dim() == 4 && flash_dim_ok(q.size(3)),
+              "flash_attn: head_dim must be one of 40/64/80/96/128/160");
+  const int b = q.size(0), h = q.size(1), sq = q.size(2), d = q.size(3);
+  const int sk = k.size(2);
+  if (causal) {
+    TORCH_CHECK(sq == sk && sq % 64 == 0 && !klens.has_value(),
+                "flash_attn causal: sq==sk, sq%64==0, no klens");
+  } else {
+    TORCH_CHECK(sq % 16 == 0 && sq >= 16,
+                "flash_attn: sq must be a multiple of 16");
+  }
+  const int* klp = nullptr;
+  if (klens.has_value()) {
+    TORCH_CHECK(klens->scalar_type() == at::kInt && klens->numel() == b
+                && klens->is_contiguous());
+    klp = klens->data_ptr<int>();
+  }
   auto o = at::empty_like(q);
-  auto lse = at::empty({b, h, s}, q.options().dtype(at::kFloat));
+  auto lse = at::empty({b, h, sq}, q.options().dtype(at::kFloat));
   fs_flash_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                    lse.data_ptr<float>(), b, h, s, (float)scale,
-                    cur_stream());
+                    lse.data_ptr<float>(), klp, b, h, sq, sk, d,
+                    causal ? 1 : 0, (float)scale, (float)drop_p,
+                    (unsigned long long)seed, cur_stream());
   return {o, lse};
 }
 
-static std::vector<at::Tensor> flash_attn_bwd(at::Tensor q, at::Tensor k,
-                                              at::Tensor v, at::Tensor o,
-                                              at::Tensor dout, at::Tensor lse,
-                                              double scale) {
-  const int b = q.size(0), h = q.size(1), s = q.size(2);
+static std::vector<at::Tensor> flash_attn_bwd(
+    at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor o, at::Tensor dout,
+    at::Tensor lse, double scale, bool causal,
+    c10::optional<at::Tensor> klens, double drop_p, int64_t seed) {
+  const int b = q.size(0), h = q.size(1), sq = q.size(2), d = q.size(3);
+  const int sk = k.size(2);
+  const int* klp = nullptr;
+  if (klens.has_value()) klp = klens->data_ptr<int>();
   auto dq = at::empty_like(q);
   auto dk = at::empty_like(k);
   auto dv = at::empty_like(v);
-  auto delta = at::empty({b, h, s}, q.options().dtype(at::kFloat));
+  auto delta = at::empty({b, h, sq}, q.options().dtype(at::kFloat));
   fs_flash_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                     dout.contiguous().data_ptr(), lse.data_ptr<float>(),
                     dq.data_ptr(), dk.data_ptr(), dv.data_ptr(),
-                    delta.data_ptr<float>(), b, h, s, (float)scale,
-                    cur_stream());
+                    delta.data_ptr<float>(), klp, b, h, sq, sk, d,
+                    causal ? 1 : 0, (float)scale, (float)drop_p,
+                    (unsigned long long)seed, cur_stream());
   return {dq, dk, dv};
 }
 

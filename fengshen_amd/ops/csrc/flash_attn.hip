@@ -1,45 +1,105 @@
-// Flash-style fused causal attention FORWARD for MI355X (gfx950).
+// Flash-style fused attention (fwd + FA2-style bwd) for MI355X (gfx950).
 //
 // Replaces the reference's flash_attn_cuda import (flash_attention.py:7) with
-// a native CDNA4 kernel: per Q-tile, stream K/V tiles through LDS, QK^T and
-// PV on MFMA (v_mfma_f32_16x16x32_bf16), online softmax in fp32 registers.
-// No S×S materialization (removes the reference's fused-softmax sk<=2048 cap
-// and the 3x HBM round trip of the bmm path).
+// a native CDNA4 kernel family: per Q-tile, stream K/V tiles through LDS,
+// QK^T and PV on MFMA (v_mfma_f32_16x16x32_bf16), online softmax in fp32
+// registers.  No SxS materialization (removes the reference's fused-softmax
+// sk<=2048 cap and the 3x HBM round trip of the bmm path).
 //
-// v2 structure:
-//   block = 4 waves; wave w owns 16 q-rows  => QBLK = 64
-//   KVBLK = 64; K staged [64][D] with XOR swizzle (guide §6 G4: row-major
-//   [*][128] bf16 is a 16-way ds_read_b128 bank conflict; byte ^= (row&7)<<4
-//   fixes it); V staged transposed [D][64+pad] so PV B-fragments read
-//   16B-contiguous kv runs.  D = 128, bf16, causal, seq % 64 == 0.
-// Outputs: O [b,h,s,D] and LSE [b,h,s] (softmax log-sum-exp, for backward).
+// Generality (round 2): kernels are templated on <int D, bool CAUSAL>:
+//   D in {40, 64, 80, 96, 128, 160}  — covers LLaMA/GPT (128/96), BERT (64),
+//       SD UNet self/cross attention (40/80/160).  D is padded to DP
+//       (multiple of 32) in LDS/fragments; global loads are guarded at
+//       8-element granularity (D % 8 == 0), pad lanes carry zeros.
+//   CAUSAL=true : dense causal self-attention (sq == sk, sq % 64 == 0) —
+//       byte-identical schedule to the round-1 kernel at D=128.
+//   CAUSAL=false: bidirectional, supports cross-attention (sq != sk),
+//       ragged sk (any length; tail masked), and per-batch key lengths
+//       (klens[b], suffix padding masks a la BERT) — key col >= klen
+//       contributes nothing to softmax, P, dK, dV.
+//
+// Structure (guide §B "fused attention prefill"):
+//   block = 8 waves; wave w owns 16 q-rows => QBLK = 128; KVBLK = 64.
+//   K staged [64][DP] with XOR swizzle (guide §6 G4 / T2: row-major bf16 is
+//   a wide ds_read_b128 bank conflict; byte ^= (row&7)<<4 fixes it; LDS row
+//   stride is padded to a power of two so the XOR stays a bijection).
+//   V staged transposed [DP][64+pad] with kv-index swizzle so PV B-fragments
+//   read 16B-contiguous kv runs.  async-STAGE split (T14): next tile's
+//   global loads issue during current tile's MFMA.
+// Outputs: O [b,h,sq,D] and LSE [b,h,sq] (softmax log-sum-exp, for bwd).
 
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) short bf16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
-#define FA_D 128
-#define FA_QBLK 128          // 8 waves x 16 q rows: 2x compute per barrier
+#define FA_QBLK 128          // 8 waves x 16 q rows
 #define FA_KVBLK 64
 #define FA_WAVES 8
 #define FA_VPAD 8
+#define FB_T 32              // backward streaming tile
 
-// XOR swizzle for k_lds rows (applies to 16B-aligned byte offsets)
+// padded head dim (fragment/LDS layout granularity)
+constexpr int fa_dpad(int d) { return (d + 31) / 32 * 32; }
+// LDS row stride in BYTES: power of two >= 2*DP so the (row&7)<<4 XOR
+// swizzle is a bijection that never crosses rows
+constexpr int fa_swb(int dp) {
+  int need = 2 * dp, p = 128;
+  while (p < need) p <<= 1;
+  return p;
+}
+
+// XOR swizzle for row-major K/Q/V LDS tiles (16B-granular byte offsets)
 __device__ __forceinline__ int kswz(int row, int byte_in_row) {
   return byte_in_row ^ ((row & 7) << 4);
 }
 
+// splitmix64 counter hash for attention dropout: fwd and bwd regenerate
+// the SAME keep/drop decision per (bh, qrow, kcol) from (seed, index) —
+// no mask tensor is ever materialized.
+__device__ __forceinline__ unsigned int fa_hash(unsigned long long seed,
+                                                unsigned long long idx) {
+  unsigned long long z = seed + idx * 0x9E3779B97F4A7C15ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  return (unsigned int)(z >> 32);
+}
+
+__device__ __forceinline__ short fa_bf16bits(float f) {
+  return (short)__hip_bfloat16_raw(__float2bfloat16(f)).x;
+}
+
+// guarded 8-elem global load: zero vector beyond D (D % 8 == 0)
+template <int D>
+__device__ __forceinline__ bf16x8 fa_load8(const bf16_t* rowp, int d0) {
+  if (d0 < D) return *reinterpret_cast<const bf16x8*>(rowp + d0);
+  bf16x8 z;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) z[j] = 0;
+  return z;
+}
+
+// ===========================================================================
+// FORWARD
+// ===========================================================================
+template <int D, bool CAUSAL>
 __global__ __launch_bounds__(FA_WAVES * 64)
 void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
                            const bf16_t* __restrict__ K,
                            const bf16_t* __restrict__ V,
                            bf16_t* __restrict__ O,
                            float* __restrict__ LSE,
-                           int b, int h, int s, float scale) {
-  // raw bf16 bits in short storage (short=bf16_t assignment would convert)
-  __shared__ short k_lds[FA_KVBLK][FA_D];              // swizzled rows
-  __shared__ short vt_lds[FA_D][FA_KVBLK + FA_VPAD];   // V transposed
+                           const int* __restrict__ klens,
+                           int b, int h, int sq, int sk, float scale,
+                           unsigned int drop_thresh, float keep_scale,
+                           unsigned long long drop_seed) {
+  constexpr int DP = fa_dpad(D);
+  constexpr int NC = DP / 32;    // MFMA K-chunks
+  constexpr int NTO = DP / 16;   // output d-tiles
+  constexpr int SWB = fa_swb(DP);
+
+  __shared__ char k_raw[FA_KVBLK * SWB];               // swizzled K rows
+  __shared__ short vt_lds[DP][FA_KVBLK + FA_VPAD];     // V transposed
   __shared__ short p_lds[FA_WAVES][16][FA_KVBLK + FA_VPAD];
 
   const int lane = threadIdx.x & 63;
@@ -48,95 +108,110 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
   const int head = blockIdx.y;
   const int batch = blockIdx.z;
 
-  const long bh_off = ((long)batch * h + head) * s * FA_D;
-  const bf16_t* Qp = Q + bh_off;
-  const bf16_t* Kp = K + bh_off;
-  const bf16_t* Vp = V + bh_off;
-  bf16_t* Op = O + bh_off;
+  const bf16_t* Qp = Q + ((long)batch * h + head) * sq * D;
+  const bf16_t* Kp = K + ((long)batch * h + head) * sk * D;
+  const bf16_t* Vp = V + ((long)batch * h + head) * sk * D;
+  bf16_t* Op = O + ((long)batch * h + head) * sq * D;
+
+  const int klen = CAUSAL ? sk
+                          : min(klens ? klens[batch] : sk, sk);
 
   const int q0 = qb * FA_QBLK + wave * 16;
-  // partial blocks (s % FA_QBLK != 0): OOB waves compute on clamped rows
-  // (must still hit every __syncthreads) and skip their stores
-  const bool q_active = q0 < s;
-  const int q0c = q_active ? q0 : s - 16;
+  // partial blocks: OOB waves compute on clamped rows (must still hit every
+  // __syncthreads) and skip their stores
+  const bool q_active = q0 < sq;
+  const int q0c = q_active ? q0 : (sq > 16 ? sq - 16 : 0);
 
   // ---- Q tile -> A-fragments (layout LA0: row=l%16, k=(l/16)*8+j;
   // verified on HW by scripts/mfma_probe), pre-scaled -----------------------
-  bf16x8 q_frag[4];
+  bf16x8 q_frag[NC];
   {
     const int row = lane & 15;
     const int k0 = (lane >> 4) * 8;
+    const int qrow = min(q0c + row, sq - 1);
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      bf16x8 raw = *reinterpret_cast<const bf16x8*>(
-          Qp + (long)(q0c + row) * FA_D + c * 32 + k0);
+    for (int c = 0; c < NC; ++c) {
+      bf16x8 raw = fa_load8<D>(Qp + (long)qrow * D, c * 32 + k0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         unsigned short u = (unsigned short)raw[j];
         float f = __uint_as_float(((unsigned int)u) << 16) * scale;
-        raw[j] = (short)__hip_bfloat16_raw(__float2bfloat16(f)).x;
+        raw[j] = fa_bf16bits(f);
       }
       q_frag[c] = raw;
     }
   }
 
   float m_run[4], l_run[4];
-  f32x4 o_acc[8];
+  f32x4 o_acc[NTO];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     m_run[r] = -INFINITY;
     l_run[r] = 0.f;
   }
 #pragma unroll
-  for (int t = 0; t < 8; ++t) o_acc[t] = f32x4{0, 0, 0, 0};
+  for (int t = 0; t < NTO; ++t) o_acc[t] = f32x4{0, 0, 0, 0};
 
-  const int q_hi = min(qb * FA_QBLK + FA_QBLK, s) - 1;
-  const int n_kv_tiles = (q_hi / FA_KVBLK) + 1;
+  int n_kv_tiles;
+  if (CAUSAL) {
+    const int q_hi = min(qb * FA_QBLK + FA_QBLK, sq) - 1;
+    n_kv_tiles = (q_hi / FA_KVBLK) + 1;
+  } else {
+    n_kv_tiles = (klen + FA_KVBLK - 1) / FA_KVBLK;
+    if (n_kv_tiles < 1) n_kv_tiles = 1;  // empty rows still write O=0
+  }
 
-  // async-STAGE split (guide §6 G15): global loads for tile t+1 are issued
-  // DURING tile t's compute (registers k_reg/v_reg), and only the cheap
-  // LDS writes sit between the barriers — HBM latency hides under MFMA.
+  // async-STAGE split (guide T14): global loads for tile t+1 are issued
+  // DURING tile t's compute (registers k_reg/v_reg); only the cheap LDS
+  // writes sit between the barriers — HBM latency hides under MFMA.
   const int tid = threadIdx.x;
-  bf16x8 k_reg[2], v_reg[2];
+  constexpr int ELEMS = FA_KVBLK * DP;          // elements per tile
+  constexpr int SWEEPS = (ELEMS + FA_WAVES * 64 * 8 - 1) / (FA_WAVES * 64 * 8);
+  bf16x8 k_reg[SWEEPS], v_reg[SWEEPS];
 #pragma unroll
-  for (int sweep = 0; sweep < 2; ++sweep) {
-    const int i = tid * 8 + sweep * 4096;
-    const int kr = i / FA_D;
-    const int kc = i % FA_D;
-    k_reg[sweep] = *reinterpret_cast<const bf16x8*>(Kp + (long)kr * FA_D + kc);
-    v_reg[sweep] = *reinterpret_cast<const bf16x8*>(Vp + (long)kr * FA_D + kc);
+  for (int sweep = 0; sweep < SWEEPS; ++sweep) {
+    const int i = tid * 8 + sweep * (FA_WAVES * 64 * 8);
+    if (i < ELEMS) {
+      const int kr = i / DP;
+      const int kc = i % DP;
+      const long krg = min(kr, sk - 1);
+      k_reg[sweep] = fa_load8<D>(Kp + krg * D, kc);
+      v_reg[sweep] = fa_load8<D>(Vp + krg * D, kc);
+    }
   }
 
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
     const int k_base = kt * FA_KVBLK;
     __syncthreads();
 #pragma unroll
-    for (int sweep = 0; sweep < 2; ++sweep) {
-      const int i = tid * 8 + sweep * 4096;
-      const int kr = i / FA_D;
-      const int kc = i % FA_D;
-      *reinterpret_cast<bf16x8*>(
-          reinterpret_cast<char*>(&k_lds[kr][0]) + kswz(kr, kc * 2)) =
-          k_reg[sweep];
-      bf16x8 vv = v_reg[sweep];
-      // kv index XOR-swizzled by d bits 3-5 (the transpose scatter would
-      // otherwise put 16 lanes of one K-row into ONE bank)
+    for (int sweep = 0; sweep < SWEEPS; ++sweep) {
+      const int i = tid * 8 + sweep * (FA_WAVES * 64 * 8);
+      if (i < ELEMS) {
+        const int kr = i / DP;
+        const int kc = i % DP;
+        *reinterpret_cast<bf16x8*>(k_raw + kr * SWB + kswz(kr, kc * 2)) =
+            k_reg[sweep];
+        bf16x8 vv = v_reg[sweep];
+        // kv index XOR-swizzled by d bits 3-5 (the transpose scatter would
+        // otherwise put 16 lanes of one K-row into ONE bank)
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        vt_lds[kc + j][kr ^ ((kc + j) & 0x38)] = vv[j];
+        for (int j = 0; j < 8; ++j)
+          vt_lds[kc + j][kr ^ ((kc + j) & 0x38)] = vv[j];
+      }
     }
     __syncthreads();
     if (kt + 1 < n_kv_tiles) {
       const int nb = (kt + 1) * FA_KVBLK;
 #pragma unroll
-      for (int sweep = 0; sweep < 2; ++sweep) {
-        const int i = tid * 8 + sweep * 4096;
-        const int kr = i / FA_D;
-        const int kc = i % FA_D;
-        k_reg[sweep] = *reinterpret_cast<const bf16x8*>(
-            Kp + (long)(nb + kr) * FA_D + kc);
-        v_reg[sweep] = *reinterpret_cast<const bf16x8*>(
-            Vp + (long)(nb + kr) * FA_D + kc);
+      for (int sweep = 0; sweep < SWEEPS; ++sweep) {
+        const int i = tid * 8 + sweep * (FA_WAVES * 64 * 8);
+        if (i < ELEMS) {
+          const int kr = i / DP;
+          const int kc = i % DP;
+          const long krg = min(nb + kr, sk - 1);
+          k_reg[sweep] = fa_load8<D>(Kp + krg * D, kc);
+          v_reg[sweep] = fa_load8<D>(Vp + krg * D, kc);
+        }
       }
     }
 
@@ -150,17 +225,17 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
       for (int nt = 0; nt < 4; ++nt) {
         const int krow = nt * 16 + col;
 #pragma unroll
-        for (int c = 0; c < 4; ++c) {
+        for (int c = 0; c < NC; ++c) {
           const int d0 = c * 32 + (lane >> 4) * 8;
           bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(&k_lds[krow][0]) + kswz(krow, d0 * 2));
+              k_raw + krow * SWB + kswz(krow, d0 * 2));
           s_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               q_frag[c], bfrag, s_acc[nt], 0, 0, 0);
         }
       }
     }
 
-    // ---- causal mask + online softmax -------------------------------------
+    // ---- mask + online softmax -------------------------------------------
     const int col = lane & 15;
     float p[4][4];
     float tile_max[4];
@@ -172,7 +247,7 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
       for (int nt = 0; nt < 4; ++nt) {
         const int kcol = k_base + nt * 16 + col;
         float v = s_acc[nt][r];
-        if (kcol > qrow) v = -INFINITY;
+        if (CAUSAL ? (kcol > qrow) : (kcol >= klen)) v = -INFINITY;
         p[nt][r] = v;
         tm = fmaxf(tm, v);
       }
@@ -185,10 +260,12 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
         tile_max[r] = fmaxf(tile_max[r], __shfl_xor(tile_max[r], off, 64));
     }
     float alpha[4], rowsum[4];
+    bool need_rescale = false;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const float m_new = fmaxf(m_run[r], tile_max[r]);
       alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - m_new);
+      if (m_new != m_run[r]) need_rescale = true;
       m_run[r] = m_new;
       float ps = 0.f;
 #pragma unroll
@@ -208,10 +285,31 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
     for (int r = 0; r < 4; ++r)
       l_run[r] = l_run[r] * alpha[r] + rowsum[r];
+    // attention dropout on P (normalizer l uses the UNdropped sum, like
+    // eager dropout(softmax(S))); uniform branch, free when p == 0
+    if (drop_thresh) {
+      const unsigned long long bh_base =
+          ((unsigned long long)batch * h + head) * sq;
 #pragma unroll
-    for (int t = 0; t < 8; ++t) {
+      for (int r = 0; r < 4; ++r) {
+        const unsigned long long qi =
+            (bh_base + (unsigned long long)(q0c + (lane >> 4) * 4 + r)) *
+            (unsigned long long)sk;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) o_acc[t][r] *= alpha[r];
+        for (int nt = 0; nt < 4; ++nt) {
+          const int kcol = k_base + nt * 16 + col;
+          p[nt][r] = (fa_hash(drop_seed, qi + kcol) < drop_thresh)
+                         ? 0.f : p[nt][r] * keep_scale;
+        }
+      }
+    }
+    // skip the O-wide rescale when no row max moved (defer-max lite, T13)
+    if (__any(need_rescale)) {
+#pragma unroll
+      for (int t = 0; t < NTO; ++t) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) o_acc[t][r] *= alpha[r];
+      }
     }
 
     // ---- P -> per-wave LDS, then PV ---------------------------------------
@@ -220,8 +318,7 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
       const int row = (lane >> 4) * 4 + r;
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
-        p_lds[wave][row][nt * 16 + col] =
-            (short)__hip_bfloat16_raw(__float2bfloat16(p[nt][r])).x;
+        p_lds[wave][row][nt * 16 + col] = fa_bf16bits(p[nt][r]);
       }
     }
     __builtin_amdgcn_s_waitcnt(0);  // wave-local LDS visibility
@@ -233,7 +330,7 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
       const int kv0 = kc * 32 + (lane >> 4) * 8;
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wave][row][kv0]);
 #pragma unroll
-      for (int t = 0; t < 8; ++t) {
+      for (int t = 0; t < NTO; ++t) {
         const int dcol = t * 16 + (lane & 15);
         bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
             &vt_lds[dcol][kv0 ^ (dcol & 0x38)]);
@@ -249,27 +346,18 @@ void flash_attn_fwd_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int qrow = q0 + (lane >> 4) * 4 + r;
+    if (qrow >= sq) continue;
     const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
 #pragma unroll
-    for (int t = 0; t < 8; ++t) {
-      Op[(long)qrow * FA_D + t * 16 + col] =
-          __float2bfloat16(o_acc[t][r] * inv_l);
+    for (int t = 0; t < NTO; ++t) {
+      const int d = t * 16 + col;
+      if (d < D) Op[(long)qrow * D + d] = __float2bfloat16(o_acc[t][r] * inv_l);
     }
     if (col == 0 && LSE) {
-      LSE[((long)batch * h + head) * s + qrow] =
+      LSE[((long)batch * h + head) * sq + qrow] =
           m_run[r] + logf(fmaxf(l_run[r], 1e-30f));
     }
   }
-}
-
-extern "C" void fs_flash_attn_fwd(const void* q, const void* k, const void* v,
-                                  void* o, float* lse, int b, int h, int s,
-                                  float scale, hipStream_t stream) {
-  dim3 grid((s + FA_QBLK - 1) / FA_QBLK, h, b);
-  dim3 block(FA_WAVES * 64);
-  hipLaunchKernelGGL(flash_attn_fwd_kernel, grid, block, 0, stream,
-                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
-                     (bf16_t*)o, lse, b, h, s, scale);
 }
 
 // ===========================================================================
@@ -280,12 +368,10 @@ extern "C" void fs_flash_attn_fwd(const void* q, const void* k, const void* v,
 //                          dS=P*(dP-delta)*s; dQ += dS K
 //   dKdV kernel (kv-major): St=(sK)Q^T; Pt=exp(St-LSE[col]); dPt=V dO^T;
 //                          dV += Pt dO; dSt=Pt*(dPt-delta[col])*s; dK += dSt Q
-// Same fragment layouts / swizzles as forward (HW-verified via mfma_probe).
-// Streaming tiles are 32-wide (FB_T) so LDS stays ~40KB -> 3-4 blocks/CU
-// (the 64-wide first cut sat at 1 block/CU and was latency-bound).
+// Same fragment layouts / swizzles as forward.  Streaming tiles are 32-wide
+// (FB_T) so LDS stays small -> 3-4 blocks/CU.
 
-#define FB_T 32
-
+template <int D>
 __global__ __launch_bounds__(256)
 void flash_delta_kernel(const bf16_t* __restrict__ dO,
                         const bf16_t* __restrict__ O,
@@ -295,18 +381,21 @@ void flash_delta_kernel(const bf16_t* __restrict__ dO,
   const long row0 = (long)blockIdx.x * 4 + wid;
   const long stride = (long)gridDim.x * 4;
   for (long row = row0; row < rows; row += stride) {
-    const bf16_t* dop = dO + row * FA_D + lane * 2;
-    const bf16_t* op = O + row * FA_D + lane * 2;
+    const bf16_t* dop = dO + row * D;
+    const bf16_t* op = O + row * D;
     float sacc = 0.f;
+    for (int i = lane * 2; i < D; i += 128) {
 #pragma unroll
-    for (int j = 0; j < 2; ++j)
-      sacc += __bfloat162float(dop[j]) * __bfloat162float(op[j]);
+      for (int j = 0; j < 2; ++j)
+        sacc += __bfloat162float(dop[i + j]) * __bfloat162float(op[i + j]);
+    }
     sacc = wave_reduce_sum(sacc);
     if (lane == 0) delta[row] = sacc;
   }
 }
 
-// Q-major: each block owns 64 q rows (wave -> 16), streams KV in 32-tiles.
+// Q-major: each block owns 128 q rows (wave -> 16), streams KV in 32-tiles.
+template <int D, bool CAUSAL>
 __global__ __launch_bounds__(FA_WAVES * 64)
 void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
                               const bf16_t* __restrict__ K,
@@ -315,77 +404,89 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
                               const float* __restrict__ LSE,
                               const float* __restrict__ Delta,
                               bf16_t* __restrict__ dQ,
-                              int b, int h, int s, float scale) {
-  __shared__ short k_lds[FB_T][FA_D];                 // K rows, swizzled
-  __shared__ short v_lds[FB_T][FA_D];                 // V rows, swizzled
-  __shared__ short kt_lds[FA_D][FB_T + FA_VPAD];      // K^T, kv-swizzled
+                              const int* __restrict__ klens,
+                              int b, int h, int sq, int sk, float scale,
+                              unsigned int drop_thresh, float keep_scale,
+                              unsigned long long drop_seed) {
+  constexpr int DP = fa_dpad(D);
+  constexpr int NC = DP / 32;
+  constexpr int NTO = DP / 16;
+  constexpr int SWB = fa_swb(DP);
+
+  __shared__ char k_raw[FB_T * SWB];                  // K rows, swizzled
+  __shared__ char v_raw[FB_T * SWB];                  // V rows, swizzled
+  __shared__ short kt_lds[DP][FB_T + FA_VPAD];        // K^T, kv-swizzled
   __shared__ short p_lds[FA_WAVES][16][FB_T + FA_VPAD];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const long bh = (long)blockIdx.z * h + blockIdx.y;
-  const bf16_t* Qp = Q + bh * s * FA_D;
-  const bf16_t* Kp = K + bh * s * FA_D;
-  const bf16_t* Vp = V + bh * s * FA_D;
-  const bf16_t* dOp = dO + bh * s * FA_D;
-  bf16_t* dQp = dQ + bh * s * FA_D;
-  const float* lse = LSE + bh * s;
-  const float* dlt = Delta + bh * s;
+  const bf16_t* Qp = Q + bh * sq * D;
+  const bf16_t* Kp = K + bh * sk * D;
+  const bf16_t* Vp = V + bh * sk * D;
+  const bf16_t* dOp = dO + bh * sq * D;
+  bf16_t* dQp = dQ + bh * sq * D;
+  const float* lse = LSE + bh * sq;
+  const float* dlt = Delta + bh * sq;
+
+  const int klen = CAUSAL ? sk
+                          : min(klens ? klens[blockIdx.z] : sk, sk);
 
   const int q0 = blockIdx.x * FA_QBLK + wave * 16;
-  const bool q_active = q0 < s;
-  const int q0c = q_active ? q0 : s - 16;
+  const bool q_active = q0 < sq;
+  const int q0c = q_active ? q0 : (sq > 16 ? sq - 16 : 0);
 
-  bf16x8 q_frag[4], do_frag[4];
+  bf16x8 q_frag[NC], do_frag[NC];
   {
     const int row = lane & 15;
     const int k0 = (lane >> 4) * 8;
+    const int qrow = min(q0c + row, sq - 1);
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      bf16x8 raw = *reinterpret_cast<const bf16x8*>(
-          Qp + (long)(q0c + row) * FA_D + c * 32 + k0);
+    for (int c = 0; c < NC; ++c) {
+      bf16x8 raw = fa_load8<D>(Qp + (long)qrow * D, c * 32 + k0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         unsigned short u = (unsigned short)raw[j];
         float f = __uint_as_float(((unsigned int)u) << 16) * scale;
-        raw[j] = (short)__hip_bfloat16_raw(__float2bfloat16(f)).x;
+        raw[j] = fa_bf16bits(f);
       }
       q_frag[c] = raw;
-      do_frag[c] = *reinterpret_cast<const bf16x8*>(
-          dOp + (long)(q0c + row) * FA_D + c * 32 + k0);
+      do_frag[c] = fa_load8<D>(dOp + (long)qrow * D, c * 32 + k0);
     }
   }
   float lse_r[4], dlt_r[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    const int qrow = q0c + (lane >> 4) * 4 + r;
+    const int qrow = min(q0c + (lane >> 4) * 4 + r, sq - 1);
     lse_r[r] = lse[qrow];
     dlt_r[r] = dlt[qrow];
   }
 
-  f32x4 dq_acc[8];
+  f32x4 dq_acc[NTO];
 #pragma unroll
-  for (int t = 0; t < 8; ++t) dq_acc[t] = f32x4{0, 0, 0, 0};
+  for (int t = 0; t < NTO; ++t) dq_acc[t] = f32x4{0, 0, 0, 0};
 
-  const int q_hi = min(blockIdx.x * FA_QBLK + FA_QBLK, s) - 1;
-  const int n_kv_tiles = (q_hi / FB_T) + 1;
+  int n_kv_tiles;
+  if (CAUSAL) {
+    const int q_hi = min(blockIdx.x * FA_QBLK + FA_QBLK, sq) - 1;
+    n_kv_tiles = (q_hi / FB_T) + 1;
+  } else {
+    n_kv_tiles = (klen + FB_T - 1) / FB_T;
+  }
 
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
     const int k_base = kt * FB_T;
     __syncthreads();
     {
       const int tid = threadIdx.x;
-      for (int i = tid * 8; i < FB_T * FA_D; i += FA_WAVES * 64 * 8) {
-        const int kr = i / FA_D;
-        const int kc = i % FA_D;
-        bf16x8 kk = *reinterpret_cast<const bf16x8*>(
-            Kp + (long)(k_base + kr) * FA_D + kc);
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(&k_lds[kr][0]) + kswz(kr, kc * 2)) = kk;
-        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
-            Vp + (long)(k_base + kr) * FA_D + kc);
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(&v_lds[kr][0]) + kswz(kr, kc * 2)) = vv;
+      for (int i = tid * 8; i < FB_T * DP; i += FA_WAVES * 64 * 8) {
+        const int kr = i / DP;
+        const int kc = i % DP;
+        const long krg = min(k_base + kr, sk - 1);
+        bf16x8 kk = fa_load8<D>(Kp + krg * D, kc);
+        *reinterpret_cast<bf16x8*>(k_raw + kr * SWB + kswz(kr, kc * 2)) = kk;
+        bf16x8 vv = fa_load8<D>(Vp + krg * D, kc);
+        *reinterpret_cast<bf16x8*>(v_raw + kr * SWB + kswz(kr, kc * 2)) = vv;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           kt_lds[kc + j][kr ^ ((kc + j) & 0x18)] = kk[j];
@@ -405,14 +506,14 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
       for (int nt = 0; nt < 2; ++nt) {
         const int krow = nt * 16 + col;
 #pragma unroll
-        for (int c = 0; c < 4; ++c) {
+        for (int c = 0; c < NC; ++c) {
           const int d0 = c * 32 + (lane >> 4) * 8;
           bf16x8 kb = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(&k_lds[krow][0]) + kswz(krow, d0 * 2));
+              k_raw + krow * SWB + kswz(krow, d0 * 2));
           s_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               q_frag[c], kb, s_acc[nt], 0, 0, 0);
           bf16x8 vb = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(&v_lds[krow][0]) + kswz(krow, d0 * 2));
+              v_raw + krow * SWB + kswz(krow, d0 * 2));
           dp_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               do_frag[c], vb, dp_acc[nt], 0, 0, 0);
         }
@@ -427,10 +528,18 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
       for (int nt = 0; nt < 2; ++nt) {
         const int kcol = k_base + nt * 16 + col;
-        float pv = (kcol > qrow) ? 0.f : __expf(s_acc[nt][r] - lse_r[r]);
-        float ds = pv * (dp_acc[nt][r] - dlt_r[r]) * scale;
-        p_lds[wave][row][nt * 16 + col] =
-            (short)__hip_bfloat16_raw(__float2bfloat16(ds)).x;
+        const bool masked = CAUSAL ? (kcol > qrow) : (kcol >= klen);
+        float pv = masked ? 0.f : __expf(s_acc[nt][r] - lse_r[r]);
+        float dpv = dp_acc[nt][r];
+        if (drop_thresh) {
+          const unsigned long long idx =
+              (bh * (unsigned long long)sq + qrow) *
+                  (unsigned long long)sk + kcol;
+          dpv = (fa_hash(drop_seed, idx) < drop_thresh)
+                    ? 0.f : dpv * keep_scale;
+        }
+        float ds = pv * (dpv - dlt_r[r]) * scale;
+        p_lds[wave][row][nt * 16 + col] = fa_bf16bits(ds);
       }
     }
     __builtin_amdgcn_s_waitcnt(0);
@@ -440,7 +549,7 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
       const int kv0 = (lane >> 4) * 8;
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wave][row][kv0]);
 #pragma unroll
-      for (int t = 0; t < 8; ++t) {
+      for (int t = 0; t < NTO; ++t) {
         const int dcol = t * 16 + (lane & 15);
         bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
             &kt_lds[dcol][kv0 ^ (dcol & 0x18)]);
@@ -455,13 +564,17 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int qrow = q0 + (lane >> 4) * 4 + r;
+    if (qrow >= sq) continue;
 #pragma unroll
-    for (int t = 0; t < 8; ++t)
-      dQp[(long)qrow * FA_D + t * 16 + col] = __float2bfloat16(dq_acc[t][r]);
+    for (int t = 0; t < NTO; ++t) {
+      const int d = t * 16 + col;
+      if (d < D) dQp[(long)qrow * D + d] = __float2bfloat16(dq_acc[t][r]);
+    }
   }
 }
 
-// KV-major: each block owns 64 kv rows (wave -> 16), streams Q in 32-tiles.
+// KV-major: each block owns 128 kv rows (wave -> 16), streams Q in 32-tiles.
+template <int D, bool CAUSAL>
 __global__ __launch_bounds__(FA_WAVES * 64)
 void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
                                const bf16_t* __restrict__ K,
@@ -471,78 +584,84 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
                                const float* __restrict__ Delta,
                                bf16_t* __restrict__ dK,
                                bf16_t* __restrict__ dV,
-                               int b, int h, int s, float scale) {
-  __shared__ short q_lds[FB_T][FA_D];
-  __shared__ short do_lds[FB_T][FA_D];
-  __shared__ short qt_lds[FA_D][FB_T + FA_VPAD];
-  __shared__ short dot_lds[FA_D][FB_T + FA_VPAD];
+                               const int* __restrict__ klens,
+                               int b, int h, int sq, int sk, float scale,
+                               unsigned int drop_thresh, float keep_scale,
+                               unsigned long long drop_seed) {
+  constexpr int DP = fa_dpad(D);
+  constexpr int NC = DP / 32;
+  constexpr int NTO = DP / 16;
+  constexpr int SWB = fa_swb(DP);
+
+  __shared__ char q_raw[FB_T * SWB];
+  __shared__ char do_raw[FB_T * SWB];
+  __shared__ short qt_lds[DP][FB_T + FA_VPAD];
+  __shared__ short dot_lds[DP][FB_T + FA_VPAD];
   __shared__ short p_lds[FA_WAVES][16][FB_T + FA_VPAD];
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const long bh = (long)blockIdx.z * h + blockIdx.y;
-  const bf16_t* Qp = Q + bh * s * FA_D;
-  const bf16_t* Kp = K + bh * s * FA_D;
-  const bf16_t* Vp = V + bh * s * FA_D;
-  const bf16_t* dOp = dO + bh * s * FA_D;
-  bf16_t* dKp = dK + bh * s * FA_D;
-  bf16_t* dVp = dV + bh * s * FA_D;
-  const float* lse = LSE + bh * s;
-  const float* dlt = Delta + bh * s;
+  const bf16_t* Qp = Q + bh * sq * D;
+  const bf16_t* Kp = K + bh * sk * D;
+  const bf16_t* Vp = V + bh * sk * D;
+  const bf16_t* dOp = dO + bh * sq * D;
+  bf16_t* dKp = dK + bh * sk * D;
+  bf16_t* dVp = dV + bh * sk * D;
+  const float* lse = LSE + bh * sq;
+  const float* dlt = Delta + bh * sq;
 
-  // each block owns FA_WAVES*16 kv rows (grid is s / FA_QBLK with
-  // FA_QBLK == FA_WAVES*16)
+  const int klen = CAUSAL ? sk
+                          : min(klens ? klens[blockIdx.z] : sk, sk);
+
+  // each block owns FA_WAVES*16 kv rows
   const int kv0_blk = blockIdx.x * (FA_WAVES * 16);
   const int kv0_wave = kv0_blk + wave * 16;
-  const bool kv_active = kv0_wave < s;
-  const int kv0c = kv_active ? kv0_wave : s - 16;
+  const bool kv_active = kv0_wave < sk;
+  const int kv0c = kv_active ? kv0_wave : (sk > 16 ? sk - 16 : 0);
 
-  bf16x8 k_frag[4], v_frag[4];
+  bf16x8 k_frag[NC], v_frag[NC];
   {
     const int row = lane & 15;
     const int c0 = (lane >> 4) * 8;
+    const int kvrow = min(kv0c + row, sk - 1);
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      bf16x8 raw = *reinterpret_cast<const bf16x8*>(
-          Kp + (long)(kv0c + row) * FA_D + c * 32 + c0);
+    for (int c = 0; c < NC; ++c) {
+      bf16x8 raw = fa_load8<D>(Kp + (long)kvrow * D, c * 32 + c0);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         unsigned short u = (unsigned short)raw[j];
         float f = __uint_as_float(((unsigned int)u) << 16) * scale;
-        raw[j] = (short)__hip_bfloat16_raw(__float2bfloat16(f)).x;
+        raw[j] = fa_bf16bits(f);
       }
       k_frag[c] = raw;
-      v_frag[c] = *reinterpret_cast<const bf16x8*>(
-          Vp + (long)(kv0c + row) * FA_D + c * 32 + c0);
+      v_frag[c] = fa_load8<D>(Vp + (long)kvrow * D, c * 32 + c0);
     }
   }
 
-  f32x4 dv_acc[8], dk_acc[8];
+  f32x4 dv_acc[NTO], dk_acc[NTO];
 #pragma unroll
-  for (int t = 0; t < 8; ++t) {
+  for (int t = 0; t < NTO; ++t) {
     dv_acc[t] = f32x4{0, 0, 0, 0};
     dk_acc[t] = f32x4{0, 0, 0, 0};
   }
 
-  const int first_qt = kv0_blk / FB_T;
-  const int n_q_tiles = s / FB_T;
+  const int first_qt = CAUSAL ? kv0_blk / FB_T : 0;
+  const int n_q_tiles = (sq + FB_T - 1) / FB_T;
 
   for (int qt = first_qt; qt < n_q_tiles; ++qt) {
     const int q_base = qt * FB_T;
     __syncthreads();
     {
       const int tid = threadIdx.x;
-      for (int i = tid * 8; i < FB_T * FA_D; i += FA_WAVES * 64 * 8) {
-        const int qr = i / FA_D;
-        const int qc = i % FA_D;
-        bf16x8 qq = *reinterpret_cast<const bf16x8*>(
-            Qp + (long)(q_base + qr) * FA_D + qc);
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(&q_lds[qr][0]) + kswz(qr, qc * 2)) = qq;
-        bf16x8 dd = *reinterpret_cast<const bf16x8*>(
-            dOp + (long)(q_base + qr) * FA_D + qc);
-        *reinterpret_cast<bf16x8*>(
-            reinterpret_cast<char*>(&do_lds[qr][0]) + kswz(qr, qc * 2)) = dd;
+      for (int i = tid * 8; i < FB_T * DP; i += FA_WAVES * 64 * 8) {
+        const int qr = i / DP;
+        const int qc = i % DP;
+        const long qrg = min(q_base + qr, sq - 1);
+        bf16x8 qq = fa_load8<D>(Qp + qrg * D, qc);
+        *reinterpret_cast<bf16x8*>(q_raw + qr * SWB + kswz(qr, qc * 2)) = qq;
+        bf16x8 dd = fa_load8<D>(dOp + qrg * D, qc);
+        *reinterpret_cast<bf16x8*>(do_raw + qr * SWB + kswz(qr, qc * 2)) = dd;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           qt_lds[qc + j][qr ^ ((qc + j) & 0x18)] = qq[j];
@@ -564,14 +683,14 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
       for (int nt = 0; nt < 2; ++nt) {
         const int qrow = nt * 16 + col;
 #pragma unroll
-        for (int c = 0; c < 4; ++c) {
+        for (int c = 0; c < NC; ++c) {
           const int d0 = c * 32 + (lane >> 4) * 8;
           bf16x8 qb = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(&q_lds[qrow][0]) + kswz(qrow, d0 * 2));
+              q_raw + qrow * SWB + kswz(qrow, d0 * 2));
           st_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               k_frag[c], qb, st_acc[nt], 0, 0, 0);
           bf16x8 db = *reinterpret_cast<const bf16x8*>(
-              reinterpret_cast<char*>(&do_lds[qrow][0]) + kswz(qrow, d0 * 2));
+              do_raw + qrow * SWB + kswz(qrow, d0 * 2));
           dpt_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               v_frag[c], db, dpt_acc[nt], 0, 0, 0);
         }
@@ -583,14 +702,24 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
     for (int nt = 0; nt < 2; ++nt) {
       const int qcol = q_base + nt * 16 + col;
-      const float lse_c = lse[qcol];
-      const float dlt_c = dlt[qcol];
+      const int qcolc = min(qcol, sq - 1);
+      const float lse_c = lse[qcolc];
+      const float dlt_c = dlt[qcolc];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int kvrow = kv0c + (lane >> 4) * 4 + r;
-        float pv = (qcol < kvrow) ? 0.f : __expf(st_acc[nt][r] - lse_c);
-        pt[nt][r] = pv;
-        dpt_acc[nt][r] = pv * (dpt_acc[nt][r] - dlt_c) * scale;  // = dSt
+        bool masked = CAUSAL ? (qcol < kvrow)
+                             : (kvrow >= klen || qcol >= sq);
+        float pv = masked ? 0.f : __expf(st_acc[nt][r] - lse_c);
+        float dm = 1.f;
+        if (drop_thresh) {
+          const unsigned long long idx =
+              (bh * (unsigned long long)sq + qcol) *
+                  (unsigned long long)sk + kvrow;
+          dm = (fa_hash(drop_seed, idx) < drop_thresh) ? 0.f : keep_scale;
+        }
+        pt[nt][r] = pv * dm;                       // dropped P feeds dV
+        dpt_acc[nt][r] = pv * (dm * dpt_acc[nt][r] - dlt_c) * scale;  // dSt
       }
     }
 
@@ -599,8 +728,7 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
       const int row = (lane >> 4) * 4 + r;
 #pragma unroll
       for (int nt = 0; nt < 2; ++nt)
-        p_lds[wave][row][nt * 16 + col] =
-            (short)__hip_bfloat16_raw(__float2bfloat16(pt[nt][r])).x;
+        p_lds[wave][row][nt * 16 + col] = fa_bf16bits(pt[nt][r]);
     }
     __builtin_amdgcn_s_waitcnt(0);
     {
@@ -608,7 +736,7 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
       const int q0f = (lane >> 4) * 8;
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wave][row][q0f]);
 #pragma unroll
-      for (int t = 0; t < 8; ++t) {
+      for (int t = 0; t < NTO; ++t) {
         const int dcol = t * 16 + (lane & 15);
         bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
             &dot_lds[dcol][q0f ^ (dcol & 0x18)]);
@@ -623,8 +751,7 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
       const int row = (lane >> 4) * 4 + r;
 #pragma unroll
       for (int nt = 0; nt < 2; ++nt)
-        p_lds[wave][row][nt * 16 + col] =
-            (short)__hip_bfloat16_raw(__float2bfloat16(dpt_acc[nt][r])).x;
+        p_lds[wave][row][nt * 16 + col] = fa_bf16bits(dpt_acc[nt][r]);
     }
     __builtin_amdgcn_s_waitcnt(0);
     {
@@ -632,7 +759,7 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
       const int q0f = (lane >> 4) * 8;
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(&p_lds[wave][row][q0f]);
 #pragma unroll
-      for (int t = 0; t < 8; ++t) {
+      for (int t = 0; t < NTO; ++t) {
         const int dcol = t * 16 + (lane & 15);
         bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
             &qt_lds[dcol][q0f ^ (dcol & 0x18)]);
@@ -647,35 +774,120 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int kvrow = kv0_wave + (lane >> 4) * 4 + r;
+    if (kvrow >= sk) continue;
 #pragma unroll
-    for (int t = 0; t < 8; ++t) {
-      dKp[(long)kvrow * FA_D + t * 16 + col] = __float2bfloat16(dk_acc[t][r]);
-      dVp[(long)kvrow * FA_D + t * 16 + col] = __float2bfloat16(dv_acc[t][r]);
+    for (int t = 0; t < NTO; ++t) {
+      const int d = t * 16 + col;
+      if (d < D) {
+        dKp[(long)kvrow * D + d] = __float2bfloat16(dk_acc[t][r]);
+        dVp[(long)kvrow * D + d] = __float2bfloat16(dv_acc[t][r]);
+      }
     }
+  }
+}
+
+// ===========================================================================
+// launchers
+// ===========================================================================
+template <int D>
+static void launch_fwd(bool causal, const void* q, const void* k,
+                       const void* v, void* o, float* lse, const int* klens,
+                       int b, int h, int sq, int sk, float scale,
+                       unsigned int dth, float ksc, unsigned long long seed,
+                       hipStream_t stream) {
+  dim3 grid((sq + FA_QBLK - 1) / FA_QBLK, h, b);
+  dim3 block(FA_WAVES * 64);
+  if (causal)
+    hipLaunchKernelGGL((flash_attn_fwd_kernel<D, true>), grid, block, 0,
+                       stream, (const bf16_t*)q, (const bf16_t*)k,
+                       (const bf16_t*)v, (bf16_t*)o, lse, klens, b, h, sq, sk,
+                       scale, dth, ksc, seed);
+  else
+    hipLaunchKernelGGL((flash_attn_fwd_kernel<D, false>), grid, block, 0,
+                       stream, (const bf16_t*)q, (const bf16_t*)k,
+                       (const bf16_t*)v, (bf16_t*)o, lse, klens, b, h, sq, sk,
+                       scale, dth, ksc, seed);
+}
+
+extern "C" void fs_flash_attn_fwd(const void* q, const void* k, const void* v,
+                                  void* o, float* lse, const int* klens,
+                                  int b, int h, int sq, int sk, int d,
+                                  int causal, float scale, float drop_p,
+                                  unsigned long long seed,
+                                  hipStream_t stream) {
+  const unsigned int dth =
+      (drop_p > 0.f) ? (unsigned int)(drop_p * 4294967296.0) : 0u;
+  const float ksc = (drop_p > 0.f) ? 1.f / (1.f - drop_p) : 1.f;
+#define FWD_CASE(DD) case DD: launch_fwd<DD>(causal, q, k, v, o, lse, klens, \
+    b, h, sq, sk, scale, dth, ksc, seed, stream); break;
+  switch (d) {
+    FWD_CASE(40) FWD_CASE(64) FWD_CASE(80)
+    FWD_CASE(96) FWD_CASE(128) FWD_CASE(160)
+  }
+#undef FWD_CASE
+}
+
+template <int D>
+static void launch_bwd(bool causal, const void* q, const void* k,
+                       const void* v, const void* o, const void* dout,
+                       const float* lse, void* dq, void* dk, void* dv,
+                       float* delta_ws, const int* klens, int b, int h,
+                       int sq, int sk, float scale, unsigned int dth,
+                       float ksc, unsigned long long seed,
+                       hipStream_t stream) {
+  const long rows = (long)b * h * sq;
+  {
+    long blocks = (rows + 3) / 4;
+    if (blocks > 4096) blocks = 4096;
+    hipLaunchKernelGGL((flash_delta_kernel<D>), dim3((unsigned)blocks),
+                       dim3(256), 0, stream, (const bf16_t*)dout,
+                       (const bf16_t*)o, delta_ws, rows);
+  }
+  dim3 gq((sq + FA_QBLK - 1) / FA_QBLK, h, b);
+  dim3 gkv((sk + FA_WAVES * 16 - 1) / (FA_WAVES * 16), h, b);
+  dim3 block(FA_WAVES * 64);
+  if (causal) {
+    hipLaunchKernelGGL((flash_attn_bwd_dq_kernel<D, true>), gq, block, 0,
+                       stream, (const bf16_t*)q, (const bf16_t*)k,
+                       (const bf16_t*)v, (const bf16_t*)dout, lse, delta_ws,
+                       (bf16_t*)dq, klens, b, h, sq, sk, scale, dth, ksc,
+                       seed);
+    hipLaunchKernelGGL((flash_attn_bwd_dkv_kernel<D, true>), gkv, block, 0,
+                       stream, (const bf16_t*)q, (const bf16_t*)k,
+                       (const bf16_t*)v, (const bf16_t*)dout, lse, delta_ws,
+                       (bf16_t*)dk, (bf16_t*)dv, klens, b, h, sq, sk, scale,
+                       dth, ksc, seed);
+  } else {
+    hipLaunchKernelGGL((flash_attn_bwd_dq_kernel<D, false>), gq, block, 0,
+                       stream, (const bf16_t*)q, (const bf16_t*)k,
+                       (const bf16_t*)v, (const bf16_t*)dout, lse, delta_ws,
+                       (bf16_t*)dq, klens, b, h, sq, sk, scale, dth, ksc,
+                       seed);
+    hipLaunchKernelGGL((flash_attn_bwd_dkv_kernel<D, false>), gkv, block, 0,
+                       stream, (const bf16_t*)q, (const bf16_t*)k,
+                       (const bf16_t*)v, (const bf16_t*)dout, lse, delta_ws,
+                       (bf16_t*)dk, (bf16_t*)dv, klens, b, h, sq, sk, scale,
+                       dth, ksc, seed);
   }
 }
 
 extern "C" void fs_flash_attn_bwd(const void* q, const void* k, const void* v,
                                   const void* o, const void* dout,
                                   const float* lse, void* dq, void* dk,
-                                  void* dv, float* delta_ws, int b, int h,
-                                  int s, float scale, hipStream_t stream) {
-  const long rows = (long)b * h * s;
-  {
-    long blocks = (rows + 3) / 4;
-    if (blocks > 4096) blocks = 4096;
-    hipLaunchKernelGGL(flash_delta_kernel, dim3((unsigned)blocks), dim3(256),
-                       0, stream, (const bf16_t*)dout, (const bf16_t*)o,
-                       delta_ws, rows);
+                                  void* dv, float* delta_ws, const int* klens,
+                                  int b, int h, int sq, int sk, int d,
+                                  int causal, float scale, float drop_p,
+                                  unsigned long long seed,
+                                  hipStream_t stream) {
+  const unsigned int dth =
+      (drop_p > 0.f) ? (unsigned int)(drop_p * 4294967296.0) : 0u;
+  const float ksc = (drop_p > 0.f) ? 1.f / (1.f - drop_p) : 1.f;
+#define BWD_CASE(DD) case DD: launch_bwd<DD>(causal, q, k, v, o, dout, lse, \
+    dq, dk, dv, delta_ws, klens, b, h, sq, sk, scale, dth, ksc, seed, \
+    stream); break;
+  switch (d) {
+    BWD_CASE(40) BWD_CASE(64) BWD_CASE(80)
+    BWD_CASE(96) BWD_CASE(128) BWD_CASE(160)
   }
-  dim3 grid((s + FA_QBLK - 1) / FA_QBLK, h, b);
-  dim3 block(FA_WAVES * 64);
-  hipLaunchKernelGGL(flash_attn_bwd_dq_kernel, grid, block, 0, stream,
-                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
-                     (const bf16_t*)dout, lse, delta_ws, (bf16_t*)dq, b, h, s,
-                     scale);
-  hipLaunchKernelGGL(flash_attn_bwd_dkv_kernel, grid, block, 0, stream,
-                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
-                     (const bf16_t*)dout, lse, delta_ws, (bf16_t*)dk,
-                     (bf16_t*)dv, b, h, s, scale);
+#undef BWD_CASE
 }

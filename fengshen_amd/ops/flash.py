@@ -1,67 +1,103 @@
-"""Flash attention wrapper: fused HIP forward, composed backward.
+"""Flash attention wrapper: fused HIP forward + fused FA2-style backward.
 
-Forward: csrc/flash_attn.hip (no SxS materialization, LSE saved).
-Backward (v1): recompute-from-QKV composition on hipBLASLt bmm + the fused
-softmax-backward identity — dV = P^T dO; dP = dO V^T;
-dS = P*(dP - rowsum(dP*P)); dQ = dS K * scale; dK = dS^T Q * scale.
-P is rebuilt row-block-exactly from the saved LSE, so forward and backward
-agree bitwise on the softmax normalizer.  A fully fused backward kernel is
-the planned next step.
+csrc/flash_attn.hip (no SxS materialization, LSE saved).  Round-2
+generality (ref flash_attention.py:174 call semantics):
+- causal self-attention (LLaMA/GPT): sq == sk, sq % 64 == 0;
+- bidirectional with optional per-batch key lengths (BERT suffix padding
+  masks become klens int32 [b]);
+- cross-attention (sq != sk, ragged sk) for the SD UNet;
+- head_dim in {40, 64, 80, 96, 128, 160} (BERT 64, GPT2-3.5B 96,
+  LLaMA 128, SD 40/80/160).
 """
 from __future__ import annotations
+
+from typing import Optional
 
 import torch
 
 from fengshen_amd.ops import get_ext
 
+_FLASH_DIMS = (40, 64, 80, 96, 128, 160)
+
 
 class _FlashAttention(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, scale):
-        o, lse = get_ext().flash_attn_fwd(q, k, v, scale)
+    def forward(ctx, q, k, v, scale, causal, klens, dropout_p, seed):
+        o, lse = get_ext().flash_attn_fwd(q, k, v, scale, causal, klens,
+                                          dropout_p, seed)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.scale = scale
+        ctx.causal = causal
+        ctx.klens = klens
+        ctx.dropout_p = dropout_p
+        ctx.seed = seed
         return o
 
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
-        scale = ctx.scale
-        if hasattr(get_ext(), "flash_attn_bwd"):
-            dq, dk, dv = get_ext().flash_attn_bwd(q, k, v, o, do.contiguous(),
-                                                  lse, scale)
-            return dq, dk, dv, None
-        b, h, s, d = q.shape
-        qf = q.reshape(b * h, s, d)
-        kf = k.reshape(b * h, s, d)
-        vf = v.reshape(b * h, s, d)
-        dof = do.contiguous().reshape(b * h, s, d)
-        # rebuild P from QK^T and the saved LSE (fp32 softmax, bf16 P)
-        scores = torch.baddbmm(
-            torch.empty(b * h, s, s, dtype=q.dtype, device=q.device),
-            qf, kf.transpose(1, 2), beta=0.0, alpha=scale).float()
-        causal = torch.ones(s, s, dtype=torch.bool, device=q.device).triu(1)
-        scores.masked_fill_(causal, float("-inf"))
-        p = torch.exp(scores - lse.reshape(b * h, s, 1))
-        pb = p.to(q.dtype)
-        dv = torch.bmm(pb.transpose(1, 2), dof)
-        dp = torch.bmm(dof, vf.transpose(1, 2)).float()
-        delta = (dp * p).sum(-1, keepdim=True)
-        ds = (p * (dp - delta)).to(q.dtype)
-        dq = torch.bmm(ds, kf) * scale
-        dk = torch.bmm(ds.transpose(1, 2), qf) * scale
-        return (dq.view(b, h, s, d), dk.view(b, h, s, d),
-                dv.view(b, h, s, d), None)
+        dq, dk, dv = get_ext().flash_attn_bwd(
+            q, k, v, o, do.contiguous(), lse, ctx.scale, ctx.causal,
+            ctx.klens, ctx.dropout_p, ctx.seed)
+        return dq, dk, dv, None, None, None, None, None
+
+
+def _draw_seed(device) -> int:
+    """Per-call dropout seed: CPU torch RNG (deterministic under
+    torch.manual_seed) mixed with the TP rank so TP-local heads get
+    decorrelated masks (eager path forks the mpu RNG tracker for the
+    same reason)."""
+    base = int(torch.randint(0, 2 ** 62, (1,)).item())
+    try:
+        from fengshen_amd.parallel import groups as pg
+        base += pg.get_tensor_model_parallel_rank() * 0x9E3779B9
+    except Exception:
+        pass
+    return base
 
 
 def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
-                    scale: float) -> torch.Tensor:
-    """q,k,v [b, h, s, 128] bf16 contiguous; causal."""
+                    scale: float, causal: bool = True,
+                    klens: Optional[torch.Tensor] = None,
+                    dropout_p: float = 0.0) -> torch.Tensor:
+    """q [b,h,sq,d], k/v [b,h,sk,d] bf16 contiguous.  dropout_p > 0
+    applies attention dropout with an in-kernel counter hash (no mask
+    tensor); the normalizer uses the undropped row sum, matching eager
+    dropout(softmax(S)) @ V."""
+    seed = _draw_seed(q.device) if dropout_p > 0 else 0
     return _FlashAttention.apply(q.contiguous(), k.contiguous(),
-                                 v.contiguous(), scale)
+                                 v.contiguous(), scale, causal, klens,
+                                 float(dropout_p), seed)
+
+
+def mask_to_klens(mask: torch.Tensor, sk: int) -> Optional[torch.Tensor]:
+    """Convert a True=masked pad mask into per-batch key lengths when it is
+    a pure suffix-padding pattern ([b,1,1,sk] or [b,sk]); else None."""
+    if mask.dtype != torch.bool:
+        mask = mask != 0
+    if mask.dim() == 4:
+        if mask.shape[2] != 1 or mask.shape[1] != 1 or mask.shape[3] != sk:
+            return None
+        m2 = mask[:, 0, 0, :]
+    elif mask.dim() == 2 and mask.shape[1] == sk:
+        m2 = mask
+    else:
+        return None
+    klens = (~m2).sum(dim=1, dtype=torch.int32)
+    ar = torch.arange(sk, device=mask.device, dtype=torch.int32)
+    if not torch.equal(m2, ar.unsqueeze(0) >= klens.unsqueeze(1)):
+        return None  # not suffix padding
+    return klens.contiguous()
 
 
 def flash_attn_supported(q, k, v, causal, mask, dropout_p) -> bool:
-    return (causal and mask is None and dropout_p == 0.0
-            and q.dtype == torch.bfloat16 and q.shape[-1] == 128
-            and q.shape[-2] == k.shape[-2] and q.shape[-2] % 64 == 0)
+    if q.dtype != torch.bfloat16:
+        return False
+    if q.shape[-1] not in _FLASH_DIMS:
+        return False
+    sq, sk = q.shape[-2], k.shape[-2]
+    if causal:
+        return mask is None and sq == sk and sq % 64 == 0
+    # bidirectional: sq tiles at 16; sk arbitrary; mask must convert to
+    # suffix-pad klens (checked by the caller via mask_to_klens)
+    return sq % 16 == 0 and sq >= 16

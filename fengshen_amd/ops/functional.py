@@ -386,9 +386,20 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         scale = 1.0 / math.sqrt(hn)
     ext = get_ext() if use_hip(q) else None
     if ext is not None and hasattr(ext, "flash_attn_fwd"):
-        from fengshen_amd.ops.flash import flash_attention, flash_attn_supported
-        if flash_attn_supported(q, k, v, causal, mask, dropout_p):
-            return flash_attention(q, k, v, scale)
+        from fengshen_amd.ops.flash import (
+            flash_attention, flash_attn_supported, mask_to_klens)
+        drop = dropout_p if training else 0.0
+        if flash_attn_supported(q, k, v, causal, mask, drop):
+            if causal and mask is None:
+                return flash_attention(q, k, v, scale, causal=True,
+                                       dropout_p=drop)
+            if not causal:
+                klens = None
+                if mask is not None:
+                    klens = mask_to_klens(mask, sk)
+                if mask is None or klens is not None:
+                    return flash_attention(q, k, v, scale, causal=False,
+                                           klens=klens, dropout_p=drop)
     # bmm path: scores in [b*np, sq, sk]
     q2 = q.reshape(b * np_, sq, hn)
     k2 = k.reshape(b * np_, sk, hn)
