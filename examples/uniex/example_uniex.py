@@ -41,18 +41,15 @@ def main():
 
     entity_types = ["人名", "地名"]
     text = "李明今天去了北京。"
-    input_ids = torch.tensor([tokenizer.encode(text)])
-    prompts = [tokenizer.encode(t) for t in entity_types]
-    L = max(len(p) for p in prompts)
-    label_prompt_ids = torch.tensor(
-        [p + [tokenizer.pad_token_id] * (L - len(p)) for p in prompts])
-
-    results = model.extract(input_ids, label_prompt_ids, threshold=0.0,
-                            max_spans=4)
-    for span in results[0]:
-        s, e = span["span"]
-        print(f"type={entity_types[span['type']]} span={text[s - 1:e]} "
-              f"score={span['score']:.3f}")
+    from fengshen_amd.models.uniex.modeling_uniex import UniEXExtractor
+    ex = UniEXExtractor(model, tokenizer, max_length=64)
+    for mode in (True, False):  # fast and full extract
+        results = ex.extract([text], entity_types, threshold=0.0, fast=mode)
+        for span in results[0][:4]:
+            s, e = span["span"]
+            print(f"mode={'fast' if mode else 'full'} "
+                  f"type={entity_types[span['type']]} span={text[s:e + 1]} "
+                  f"score={span['score']:.3f}")
 
 
 if __name__ == "__main__":

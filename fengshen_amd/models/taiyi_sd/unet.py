@@ -24,6 +24,24 @@ from fengshen_amd.ops import functional as F_ops
 from fengshen_amd.parallel.random import checkpoint as activation_checkpoint
 
 
+def _gn(norm: nn.GroupNorm, x: torch.Tensor) -> torch.Tensor:
+    """GroupNorm in fp32 (stats + affine) regardless of param dtype."""
+    return nn.functional.group_norm(
+        x.float(), norm.num_groups,
+        norm.weight.float() if norm.weight is not None else None,
+        norm.bias.float() if norm.bias is not None else None,
+        norm.eps).to(x.dtype)
+
+
+def _ln(norm: nn.LayerNorm, x: torch.Tensor) -> torch.Tensor:
+    """LayerNorm in fp32 regardless of param dtype."""
+    return nn.functional.layer_norm(
+        x.float(), norm.normalized_shape,
+        norm.weight.float() if norm.weight is not None else None,
+        norm.bias.float() if norm.bias is not None else None,
+        norm.eps).to(x.dtype)
+
+
 class UNetConfig(PretrainedConfig):
     model_type = "fengshen_sd_unet"
 
@@ -116,12 +134,12 @@ class ResnetBlock2D(nn.Module):
                               if in_ch != out_ch else None)
 
     def forward(self, x, temb):
-        h = self.conv1(nn.functional.silu(
-            self.norm1(x.float())).to(x.dtype))
+        h = self.conv1(nn.functional.silu(_gn(self.norm1, x).float())
+                       .to(x.dtype))
         h = h + self.time_emb_proj(
             nn.functional.silu(temb))[:, :, None, None].to(h.dtype)
-        h = self.conv2(nn.functional.silu(
-            self.norm2(h.float())).to(h.dtype))
+        h = self.conv2(nn.functional.silu(_gn(self.norm2, h).float())
+                       .to(h.dtype))
         skip = x if self.conv_shortcut is None else self.conv_shortcut(x)
         return h + skip
 
@@ -193,9 +211,9 @@ class BasicTransformerBlock(nn.Module):
         self.ff = FeedForward(dim)
 
     def forward(self, x, context):
-        x = x + self.attn1(self.norm1(x.float()).to(x.dtype))
-        x = x + self.attn2(self.norm2(x.float()).to(x.dtype), context)
-        x = x + self.ff(self.norm3(x.float()).to(x.dtype))
+        x = x + self.attn1(_ln(self.norm1, x))
+        x = x + self.attn2(_ln(self.norm2, x), context)
+        x = x + self.ff(_ln(self.norm3, x))
         return x
 
 
@@ -213,7 +231,7 @@ class Transformer2DModel(nn.Module):
     def forward(self, x, context):
         b, c, hh, ww = x.shape
         res = x
-        h = self.proj_in(self.norm(x.float()).to(x.dtype))
+        h = self.proj_in(_gn(self.norm, x))
         tokens = h.flatten(2).transpose(1, 2)  # [b, hw, c]
         for block in self.transformer_blocks:
             tokens = block(tokens, context)
@@ -406,7 +424,7 @@ class UNet2DConditionModel(PreTrainedModel):
             h = block(h, skips, temb, encoder_hidden_states,
                       ckpt=self._ckpt)
         h = nn.functional.silu(
-            self.conv_norm_out(h.float())).to(h.dtype)
+            _gn(self.conv_norm_out, h).float()).to(h.dtype)
         return self.conv_out(h)
 
 

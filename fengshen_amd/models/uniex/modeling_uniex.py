@@ -1,14 +1,23 @@
-"""UniEX: unified information extraction (span + type matching).
+"""UniEX: unified information extraction via triaffine span-type scoring.
 
-Behavioral parity: reference models/uniex/modeling_uniex.py (fast/full
-extract modes) — text and task-label prompts encoded together; span scorer
-[b, s, s] picks entity spans, type scorer matches each span against the
-label prompts' CLS representations.
+Behavioral parity with reference models/uniex/modeling_uniex.py:
+- MLP start/end/cls projections + Triaffine scorer (:858-882): span logits
+  [b, x, y, n_label] = einsum(start, W, end) x cls, where cls vectors are
+  the hidden states at the in-sequence label-prompt token positions
+  (label_token_idx);
+- training path (:924-970): label 0 is the "index" head scored over the
+  FULL sequence, labels 1.. are type heads scored over text tokens only
+  (gathered by text_token_idx, span_gather :902-922); loss =
+  1e5 * (BCE(index) + BCE(types)) with additive span_labels_mask;
+- full extract (:972-984): sigmoid triaffine over every (start, end, type)
+  for gathered text tokens;
+- fast extract (:986-1025): stage 1 scores spans with the index head only,
+  stage 2 types just the surviving spans.
 """
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Optional
+from typing import List, Optional
 
 import torch
 import torch.nn as nn
@@ -21,85 +30,228 @@ from fengshen_amd.models.megatron_bert.modeling_megatron_bert import (
     MegatronBertModel,
     MegatronBertPreTrainedModel,
 )
-from fengshen_amd.models.tagging_models.bert_for_tagging import Biaffine
+
+
+class MLPLayer(nn.Module):
+    """Linear + GELU (ref :875-882)."""
+
+    def __init__(self, input_size: int, output_size: int):
+        super().__init__()
+        self.mlp = nn.Sequential(nn.Linear(input_size, output_size),
+                                 nn.GELU())
+
+    def forward(self, hidden_state):
+        return self.mlp(hidden_state)
+
+
+class Triaffine(nn.Module):
+    """span_logits[b,x,y,z] = (start_x W end_y) . cls_z (ref :858-872)."""
+
+    def __init__(self, triaffine_hidden_size: int):
+        super().__init__()
+        self.weight = nn.Parameter(torch.zeros(
+            triaffine_hidden_size, triaffine_hidden_size,
+            triaffine_hidden_size))
+        nn.init.normal_(self.weight, mean=0, std=0.1)
+
+    def forward(self, start_logits, end_logits, cls_logits):
+        span = torch.einsum("bxi,ioj,byj->bxyo",
+                            start_logits.float(), self.weight.float(),
+                            end_logits.float())
+        return torch.einsum("bxyo,bzo->bxyz", span, cls_logits.float())
 
 
 @dataclass
 class UniEXOutput(ModelOutput):
     loss: Optional[torch.Tensor] = None
     span_logits: Optional[torch.Tensor] = None
-    type_logits: Optional[torch.Tensor] = None
+    span_labels: Optional[torch.Tensor] = None
+
+
+def span_gather(span_labels: torch.Tensor,
+                text_token_idx: torch.Tensor) -> torch.Tensor:
+    """Subset a [b, s, s, L] span grid to the text tokens only ->
+    [b, t, t, L] (ref :902-922)."""
+    batch_size, seq_len, _, num_labels = span_labels.shape
+    _, text_len = text_token_idx.shape
+    e = torch.arange(seq_len, device=span_labels.device) * seq_len
+    e = e.unsqueeze(0).unsqueeze(-1).repeat(batch_size, 1, text_len)
+    e = e.gather(1, text_token_idx.unsqueeze(-1).repeat(1, 1, text_len))
+    idx = text_token_idx.unsqueeze(1).repeat(1, text_len, 1) + e
+    idx = idx.reshape(-1, text_len * text_len)
+    flat = span_labels.reshape(-1, seq_len * seq_len, num_labels)
+    out = flat.gather(1, idx.unsqueeze(-1).repeat(1, 1, num_labels))
+    return out.reshape(-1, text_len, text_len, num_labels)
 
 
 class UniEXModel(MegatronBertPreTrainedModel):
+    """Ref UniEXBertModel (:885-1025) on the MI355X-native BERT stack."""
+
     config_class = UniEXConfig
 
-    def __init__(self, config: UniEXConfig, biaffine_size: int = 128):
+    def __init__(self, config: UniEXConfig,
+                 triaffine_hidden_size: int = 128):
         super().__init__(config)
         self.bert = MegatronBertModel(config, add_pooling_layer=False)
-        self.start_proj = nn.Sequential(
-            nn.Linear(config.hidden_size, biaffine_size), nn.GELU())
-        self.end_proj = nn.Sequential(
-            nn.Linear(config.hidden_size, biaffine_size), nn.GELU())
-        self.span_biaffine = Biaffine(biaffine_size, 1)
-        self.span_rep = nn.Linear(2 * config.hidden_size, config.hidden_size)
+        ths = getattr(config, "triaffine_hidden_size",
+                      triaffine_hidden_size)
+        self.mlp_start = MLPLayer(config.hidden_size, ths)
+        self.mlp_end = MLPLayer(config.hidden_size, ths)
+        self.mlp_cls = MLPLayer(config.hidden_size, ths)
+        self.triaffine = Triaffine(ths)
+        self.loss_sigmoid = nn.BCEWithLogitsLoss()
         self.post_init()
 
     def forward(self, input_ids, attention_mask=None, token_type_ids=None,
-                label_prompt_ids=None, span_labels=None, type_labels=None,
-                candidate_spans=None, **_kw):
-        """label_prompt_ids [n_types, prompt_len]: one prompt per type;
-        span_labels [b, s, s] binary; candidate_spans [b, n_cand, 2] with
-        type_labels [b, n_cand] for type matching."""
-        h = self.bert(input_ids, attention_mask,
-                      token_type_ids).last_hidden_state
-        span_logits = self.span_biaffine(
-            self.start_proj(h), self.end_proj(h)).squeeze(-1)
-        loss = None
-        type_logits = None
-        if label_prompt_ids is not None:
-            th = self.bert(label_prompt_ids).last_hidden_state[:, 0]  # [T, h]
-            if candidate_spans is not None:
-                b, n_cand, _ = candidate_spans.shape
-                starts = candidate_spans[..., 0]
-                ends = candidate_spans[..., 1]
-                hs = torch.gather(
-                    h, 1, starts[:, :, None].expand(-1, -1, h.shape[-1]))
-                he = torch.gather(
-                    h, 1, ends[:, :, None].expand(-1, -1, h.shape[-1]))
-                rep = self.span_rep(torch.cat([hs, he], dim=-1))
-                type_logits = rep.float() @ th.float().t()  # [b, n_cand, T]
-        if span_labels is not None:
-            loss = nn.functional.binary_cross_entropy_with_logits(
-                span_logits.float(), span_labels.float())
-            if type_logits is not None and type_labels is not None:
-                loss = loss + nn.functional.cross_entropy(
-                    type_logits.view(-1, type_logits.shape[-1]),
-                    type_labels.view(-1), ignore_index=-100)
-        return UniEXOutput(loss=loss, span_logits=span_logits,
-                           type_logits=type_logits)
+                position_ids=None, span_labels=None, span_labels_mask=None,
+                label_token_idx=None, text_token_idx=None,
+                fast_ex_mode=False, threshold=0.5, **_kw):
+        h = self.bert(input_ids, attention_mask, token_type_ids,
+                      position_ids).last_hidden_state
+        batch_size, seq_len, hidden_size = h.shape
+
+        if span_labels_mask is not None:
+            # ---- training (ref :944-969) --------------------------------
+            start_logits = self.mlp_start(h)
+            end_logits = self.mlp_end(h)
+            cls_h = h.gather(
+                1, label_token_idx.unsqueeze(-1).repeat(1, 1, hidden_size))
+            cls_logits = self.mlp_cls(cls_h)
+
+            # index head (label 0) over the full sequence
+            idx_logits = self.triaffine(start_logits, end_logits,
+                                        cls_logits[:, [0], :])
+            idx_logits = idx_logits + span_labels_mask[:, :, :, [0]]
+            index_loss = self.loss_sigmoid(idx_logits,
+                                           span_labels[:, :, :, [0]].float())
+
+            # type heads (labels 1..) over text tokens only
+            ths = start_logits.shape[-1]
+            st_t = start_logits.gather(
+                1, text_token_idx.unsqueeze(-1).repeat(1, 1, ths))
+            en_t = end_logits.gather(
+                1, text_token_idx.unsqueeze(-1).repeat(1, 1, ths))
+            span_logits = self.triaffine(st_t, en_t, cls_logits[:, 1:, :])
+            span_labels_t = span_gather(span_labels[:, :, :, 1:].float(),
+                                        text_token_idx)
+            span_mask_t = span_gather(span_labels_mask[:, :, :, 1:],
+                                      text_token_idx)
+            span_logits = span_logits + span_mask_t
+            span_loss = self.loss_sigmoid(span_logits, span_labels_t)
+            all_loss = 100000 * span_loss + 100000 * index_loss
+            return UniEXOutput(loss=all_loss, span_logits=span_logits,
+                               span_labels=span_labels_t)
+
+        if not fast_ex_mode:
+            # ---- full extract (ref :972-984) ----------------------------
+            text_h = h.gather(
+                1, text_token_idx.unsqueeze(-1).repeat(1, 1, hidden_size))
+            start_logits = self.mlp_start(text_h)
+            end_logits = self.mlp_end(text_h)
+            cls_h = h.gather(
+                1, label_token_idx.unsqueeze(-1).repeat(1, 1, hidden_size))
+            cls_logits = self.mlp_cls(cls_h)
+            span_logits = torch.sigmoid(
+                self.triaffine(start_logits, end_logits, cls_logits))
+            return UniEXOutput(span_logits=span_logits)
+
+        # ---- fast extract (ref :986-1025): index spans then type --------
+        text_h = h.gather(
+            1, text_token_idx.unsqueeze(-1).repeat(1, 1, hidden_size))
+        start_logits = self.mlp_start(text_h)
+        end_logits = self.mlp_end(text_h)
+        cls_h = h.gather(
+            1, label_token_idx.unsqueeze(-1).repeat(1, 1, hidden_size))
+        cls_logits = self.mlp_cls(cls_h)
+        index_logits = cls_logits[:, :1, :]
+        type_logits = cls_logits[:, 1:, :]
+
+        span_index = torch.sigmoid(
+            self.triaffine(start_logits, end_logits,
+                           index_logits)).squeeze(-1)  # [b, t, t]
+        results = []
+        for bi in range(batch_size):
+            hits = (span_index[bi].triu() > threshold) \
+                .nonzero(as_tuple=False)
+            spans = []
+            if hits.numel():
+                st = start_logits[bi:bi + 1, hits[:, 0]]
+                en = end_logits[bi:bi + 1, hits[:, 1]]
+                # type each surviving span: score[z] for start/end pair
+                tl = self.triaffine(st, en, type_logits[bi:bi + 1])
+                # diagonal (start_i, end_i) pairs only
+                n = hits.shape[0]
+                diag = tl[0, torch.arange(n), torch.arange(n)]  # [n, T]
+                probs = torch.sigmoid(diag)
+                types = probs.argmax(-1)
+                for j in range(n):
+                    spans.append({
+                        "span": (int(hits[j, 0]), int(hits[j, 1])),
+                        "type": int(types[j]),
+                        "score": float(span_index[bi, hits[j, 0],
+                                                  hits[j, 1]]),
+                        "type_score": float(probs[j, types[j]]),
+                    })
+            results.append(spans)
+        return results
+
+
+class UniEXExtractor:
+    """Convenience extraction wrapper used by the IE pipeline: builds
+    label-prompt + text sequences and decodes span/type structures."""
+
+    def __init__(self, model: UniEXModel, tokenizer, max_length: int = 128):
+        self.model = model
+        self.tokenizer = tokenizer
+        self.max_length = max_length
 
     @torch.no_grad()
-    def extract(self, input_ids, label_prompt_ids, attention_mask=None,
-                threshold: float = 0.5, max_spans: int = 32):
-        """fast extract: top spans then type match."""
-        out = self.forward(input_ids, attention_mask)
-        probs = out.span_logits.sigmoid()
-        b, s, _ = probs.shape
+    def extract(self, texts: List[str], entity_types: List[str],
+                threshold: float = 0.5, fast: bool = True):
+        tk = self.tokenizer
+        dev = next(self.model.parameters()).device
+        batch_ids, lab_idx, txt_idx = [], [], []
+        for text in texts:
+            ids = [tk.cls_token_id]
+            li = [0]
+            for et in entity_types:
+                li.append(len(ids))
+                ids += tk.encode(et, add_special_tokens=False)
+                ids.append(tk.sep_token_id)
+            ti = list(range(len(ids), len(ids)
+                            + len(tk.encode(text,
+                                            add_special_tokens=False))))
+            ids += tk.encode(text, add_special_tokens=False)
+            ids.append(tk.sep_token_id)
+            batch_ids.append(ids[:self.max_length])
+            lab_idx.append(li)
+            txt_idx.append([i for i in ti if i < self.max_length])
+        smax = max(len(x) for x in batch_ids)
+        tmax = max(len(x) for x in txt_idx)
+        ids_t = torch.zeros(len(texts), smax, dtype=torch.long)
+        for i, x in enumerate(batch_ids):
+            ids_t[i, :len(x)] = torch.tensor(x)
+        lab_t = torch.tensor(lab_idx, dtype=torch.long)
+        txt_t = torch.zeros(len(texts), tmax, dtype=torch.long)
+        for i, x in enumerate(txt_idx):
+            txt_t[i, :len(x)] = torch.tensor(x)
+        out = self.model(ids_t.to(dev), label_token_idx=lab_t.to(dev),
+                         text_token_idx=txt_t.to(dev),
+                         fast_ex_mode=fast, threshold=threshold)
+        if fast:
+            return out
+        # full mode: decode [b, t, t, 1+T] grid; label 0 is the index head
         results = []
-        th = self.bert(label_prompt_ids).last_hidden_state[:, 0]
-        h = self.bert(input_ids, attention_mask).last_hidden_state
-        for bi in range(b):
-            upper = probs[bi].triu()
-            flat = upper.flatten()
-            vals, idxs = flat.topk(min(max_spans, flat.numel()))
+        span_logits = out.span_logits
+        for bi in range(len(texts)):
+            hits = (span_logits[bi, :, :, 0].triu()
+                    > threshold).nonzero(as_tuple=False)
             spans = []
-            for v, ix in zip(vals, idxs):
-                if v < threshold:
-                    break
-                st, en = int(ix // s), int(ix % s)
-                rep = self.span_rep(torch.cat([h[bi, st], h[bi, en]], dim=-1))
-                t = int((rep.float() @ th.float().t()).argmax())
-                spans.append({"span": (st, en), "type": t, "score": float(v)})
+            for st, en in hits.tolist():
+                tscores = span_logits[bi, st, en, 1:]
+                t = int(tscores.argmax())
+                spans.append({"span": (st, en), "type": t,
+                              "score": float(span_logits[bi, st, en, 0]),
+                              "type_score": float(tscores[t])})
             results.append(spans)
         return results

@@ -154,25 +154,45 @@ def test_tcbert_prompt_classification():
     out.loss.backward()
 
 
-def test_uniex_span_and_type():
-    from fengshen_amd.models.uniex.modeling_uniex import UniEXModel
+def test_uniex_triaffine_training_and_extract():
+    """Reference UniEXBertModel semantics (:885-1025): triaffine span
+    scoring with an index head (label 0) over the full sequence and type
+    heads over gathered text tokens; full + fast extract modes."""
+    from fengshen_amd.models.uniex.modeling_uniex import (
+        UniEXModel, span_gather)
     from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
         bert_tiny_config)
     torch.manual_seed(0)
-    m = UniEXModel(bert_tiny_config())
-    b, s = 2, 16
+    m = UniEXModel(bert_tiny_config(), triaffine_hidden_size=32)
+    b, s, n_lab = 2, 16, 3  # label 0 = index head, 1..2 = types
     ids = torch.randint(3, 256, (b, s))
-    prompts = torch.randint(3, 256, (3, 6))  # 3 types
-    span_labels = torch.zeros(b, s, s)
-    span_labels[:, 2, 4] = 1
-    cand = torch.tensor([[[2, 4], [5, 7]]] * b)
-    type_labels = torch.tensor([[0, 2]] * b)
-    out = m(ids, label_prompt_ids=prompts, span_labels=span_labels,
-            candidate_spans=cand, type_labels=type_labels)
+    label_token_idx = torch.tensor([[1, 3, 5]] * b)
+    text_token_idx = torch.tensor([[7 + i for i in range(8)]] * b)
+    span_labels = torch.zeros(b, s, s, n_lab)
+    span_labels[:, 8, 10, 0] = 1   # index head hit
+    span_labels[:, 8, 10, 1] = 1   # type 1
+    span_mask = torch.zeros(b, s, s, n_lab) - 10000.0
+    span_mask[:, 7:15, 7:15, :] = 0.0
+    out = m(ids, span_labels=span_labels, span_labels_mask=span_mask,
+            label_token_idx=label_token_idx, text_token_idx=text_token_idx)
     assert out.loss.isfinite()
+    assert out.span_logits.shape == (b, 8, 8, n_lab - 1)
     out.loss.backward()
-    res = m.extract(ids, prompts, threshold=0.9)
-    assert len(res) == b
+    # span_gather subsets the grid correctly
+    sub = span_gather(span_labels[:, :, :, 1:], text_token_idx)
+    assert sub.shape == (b, 8, 8, n_lab - 1)
+    assert sub[0, 1, 3, 0] == 1  # (8,10) -> text-relative (1,3)
+    # full extract
+    full = m(ids, label_token_idx=label_token_idx,
+             text_token_idx=text_token_idx, fast_ex_mode=False)
+    assert full.span_logits.shape == (b, 8, 8, n_lab)
+    # fast extract returns span/type dicts
+    fast = m(ids, label_token_idx=label_token_idx,
+             text_token_idx=text_token_idx, fast_ex_mode=True,
+             threshold=0.0)
+    assert len(fast) == b
+    if fast[0]:
+        assert {"span", "type", "score"} <= set(fast[0][0])
 
 
 def test_tcbert_pipeline():
