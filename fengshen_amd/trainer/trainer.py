@@ -519,7 +519,21 @@ class Trainer:
         return dict(self._metrics)
 
     @torch.no_grad()
-    def predict(self, model, dataloaders, datamodule=None):
+    def predict(self, model, dataloaders=None, datamodule=None):
+        if dataloaders is None and datamodule is not None:
+            datamodule.setup("predict")
+            for name in ("predict_dataloader", "test_dataloader",
+                         "val_dataloader"):
+                fn = getattr(datamodule, name, None)
+                if fn is not None:
+                    try:
+                        dataloaders = fn()
+                    except Exception:
+                        dataloaders = None
+                    if dataloaders is not None:
+                        break
+        if dataloaders is None:
+            raise ValueError("predict needs dataloaders or a datamodule")
         self.strategy.setup_environment(self)
         self.module = model
         model.trainer = self

@@ -35,3 +35,62 @@ def test_training_example_one_step():
     out = _run("examples/clue_sim/finetune_clue_sim.py",
                "--max_steps", "1", "--strategy", "ddp")
     assert "train_loss" in out
+
+
+@pytest.mark.parametrize("rel,extra", [
+    ("examples/zen1_finetune/fengshen_sequence_level_ft_task.py",
+     ("--max_steps", "2", "--precision", "fp32")),
+    ("examples/zen1_finetune/fengshen_token_level_ft_task.py",
+     ("--max_steps", "2", "--precision", "fp32")),
+    ("examples/summary/seq2seq_summary.py",
+     ("--max_steps", "2", "--precision", "fp32")),
+    ("examples/clue1.1/run_clue_unimc.py",
+     ("--max_steps", "2", "--precision", "fp32")),
+    ("examples/unimc/finetune_unimc.py",
+     ("--max_steps", "2", "--precision", "fp32")),
+])
+def test_new_example_smokes(rel, extra, tmp_path):
+    _run(rel, *extra, "--default_root_dir", str(tmp_path))
+
+
+def test_clue_converters_roundtrip(tmp_path):
+    import json
+    import subprocess
+    src = tmp_path / "tnews.json"
+    src.write_text(json.dumps(
+        {"sentence": "球队赢得比赛", "label_desc": "news_sports",
+         "id": 7}, ensure_ascii=False) + "\n")
+    dst = tmp_path / "uni.jsonl"
+    out = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "examples/clue1.1/clue2unidata.py"),
+         "--task", "tnews", "--input", str(src), "--output", str(dst)],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr[-500:]
+    rec = json.loads(dst.read_text())
+    assert rec["answer"] == "体育" and rec["choice"][rec["label"]] == "体育"
+
+
+def test_launcher_scripts_exist_and_reference_real_files():
+    import glob
+    import re
+    launchers = glob.glob(os.path.join(ROOT, "examples/*/run.sh"))
+    assert len(launchers) >= 10
+    for sh in launchers:
+        body = open(sh).read()
+        m = re.search(r"exec python (?:-m \S+ .*?)?(\S+\.py)", body)
+        assert m, sh
+        assert os.path.exists(os.path.join(os.path.dirname(sh),
+                                           m.group(1))), (sh, m.group(1))
+
+
+def test_rouge_score():
+    from fengshen_amd.metric.rouge import RougeScore, rouge_l, rouge_n
+    r = rouge_n(list("股市大涨"), list("股市大涨"), 1)
+    assert r["fmeasure"] == 1.0
+    r2 = rouge_l(list("股市涨"), list("股市大涨"))
+    assert abs(r2["recall"] - 3 / 4) < 1e-9
+    rs = RougeScore()
+    rs.update(["股 市 涨"], ["股 市 大 涨"])
+    out = rs.compute()
+    assert 0 < out["rougeL_fmeasure"] < 1
+    assert out["rouge1_precision"] == 1.0
