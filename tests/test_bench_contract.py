@@ -28,6 +28,9 @@ def _run_bench(nproc, extra):
 @pytest.mark.parametrize("nproc,extra,parallelism", [
     (2, ["--zero_stage", "3"], "zero3_dp2"),
     (2, ["--zero_stage", "2", "--tp", "2"], "zero2_dp1_tp2"),
+    # the driver's N=4 scaling point and the BASELINE config-4 topology
+    (4, ["--zero_stage", "3"], "zero3_dp4"),
+    (4, ["--zero_stage", "3", "--tp", "2"], "zero3_dp2_tp2"),
 ])
 def test_bench_multirank_json(nproc, extra, parallelism):
     res = _run_bench(nproc, extra)
