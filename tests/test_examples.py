@@ -94,3 +94,32 @@ def test_rouge_score():
     out = rs.compute()
     assert 0 < out["rougeL_fmeasure"] < 1
     assert out["rouge1_precision"] == 1.0
+
+
+@pytest.mark.parametrize("rel", [
+    "examples/pretrain_erlangshen_bert/pretrain_erlangshen.py",
+    "examples/ziya_llama/finetune_ziya_llama.py",
+    "examples/classification/finetune_classification.py",
+    "examples/sequence_tagging/finetune_tagging.py",
+    "examples/finetune_bart_qg/finetune_bart_qg.py",
+    "examples/mt5_summary/finetune_summary.py",
+    "examples/translate/finetune_deltalm.py",
+    "examples/wenzhong_qa/finetune_medicalQA.py",
+    "examples/hubert/pretrain_hubert.py",
+    "examples/deepVAE/pretrain_deep_vae.py",
+    "examples/pegasus/pretrain_pegasus.py",
+    "examples/clip_finetune/clip_finetune_flickr.py",
+    "examples/pretrain_taiyi_clip/pretrain_clip.py",
+    "examples/finetune_taiyi_stable_diffusion/finetune.py",
+    "examples/pretrain_deberta_v2/pretrain_deberta.py",
+    "examples/pretrain_bert/pretrain_bert_mmap.py",
+    "examples/pretrain_t5/pretrain_randeng_t5.py",
+    "examples/pretrain_randeng_bart/pretrain_bart.py",
+    "examples/zen2_finetune/finetune_zen_classification.py",
+    "examples/qa_t5/finetune_t5_qa.py",
+])
+def test_training_example_smokes(rel, tmp_path):
+    """Every training app runs 2 steps on synthetic data (the reference's
+    example-as-integration-test pattern)."""
+    _run(rel, "--max_steps", "2", "--precision", "fp32",
+         "--default_root_dir", str(tmp_path))
