@@ -474,25 +474,66 @@ void flash_attn_bwd_dq_kernel(const bf16_t* __restrict__ Q,
     n_kv_tiles = (klen + FB_T - 1) / FB_T;
   }
 
+  // async-STAGE (T14, like the forward): tile t+1 K/V global loads are
+  // issued DURING tile t's MFMA so HBM latency hides under compute; only
+  // the LDS writes sit between the barriers.
+  constexpr int BELEMS = FB_T * DP;
+  constexpr int BSWEEPS = (BELEMS + FA_WAVES * 64 * 8 - 1)
+                          / (FA_WAVES * 64 * 8);
+  bf16x8 kst[BSWEEPS], vst[BSWEEPS];
+  {
+    const int tid = threadIdx.x;
+    const int kt0_base = (CAUSAL ? 0 : 0) * FB_T;
+#pragma unroll
+    for (int sw = 0; sw < BSWEEPS; ++sw) {
+      const int i = tid * 8 + sw * (FA_WAVES * 64 * 8);
+      if (i < BELEMS) {
+        const int kr = i / DP;
+        const int kc = i % DP;
+        const long krg = min(kt0_base + kr, sk - 1);
+        kst[sw] = fa_load8<D>(Kp + krg * D, kc);
+        vst[sw] = fa_load8<D>(Vp + krg * D, kc);
+      }
+    }
+  }
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
-    const int k_base = kt * FB_T;
     __syncthreads();
     {
       const int tid = threadIdx.x;
-      for (int i = tid * 8; i < FB_T * DP; i += FA_WAVES * 64 * 8) {
-        const int kr = i / DP;
-        const int kc = i % DP;
-        const long krg = min(k_base + kr, sk - 1);
-        bf16x8 kk = fa_load8<D>(Kp + krg * D, kc);
-        *reinterpret_cast<bf16x8*>(k_raw + kr * SWB + kswz(kr, kc * 2)) = kk;
-        bf16x8 vv = fa_load8<D>(Vp + krg * D, kc);
-        *reinterpret_cast<bf16x8*>(v_raw + kr * SWB + kswz(kr, kc * 2)) = vv;
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          kt_lds[kc + j][kr ^ ((kc + j) & 0x18)] = kk[j];
+      for (int sw = 0; sw < BSWEEPS; ++sw) {
+        const int i = tid * 8 + sw * (FA_WAVES * 64 * 8);
+        if (i < BELEMS) {
+          const int kr = i / DP;
+          const int kc = i % DP;
+          bf16x8 kk = kst[sw];
+          *reinterpret_cast<bf16x8*>(
+              k_raw + kr * SWB + kswz(kr, kc * 2)) = kk;
+          *reinterpret_cast<bf16x8*>(
+              v_raw + kr * SWB + kswz(kr, kc * 2)) = vst[sw];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            kt_lds[kc + j][kr ^ ((kc + j) & 0x18)] = kk[j];
+        }
       }
     }
     __syncthreads();
+    const int k_base = kt * FB_T;
+    if (kt + 1 < n_kv_tiles) {
+      const int nb = (kt + 1) * FB_T;
+      const int tid = threadIdx.x;
+#pragma unroll
+      for (int sw = 0; sw < BSWEEPS; ++sw) {
+        const int i = tid * 8 + sw * (FA_WAVES * 64 * 8);
+        if (i < BELEMS) {
+          const int kr = i / DP;
+          const int kc = i % DP;
+          const long krg = min(nb + kr, sk - 1);
+          kst[sw] = fa_load8<D>(Kp + krg * D, kc);
+          vst[sw] = fa_load8<D>(Vp + krg * D, kc);
+        }
+      }
+    }
 
     f32x4 s_acc[2], dp_acc[2];
 #pragma unroll
@@ -649,27 +690,66 @@ void flash_attn_bwd_dkv_kernel(const bf16_t* __restrict__ Q,
   const int first_qt = CAUSAL ? kv0_blk / FB_T : 0;
   const int n_q_tiles = (sq + FB_T - 1) / FB_T;
 
+  constexpr int BELEMS = FB_T * DP;
+  constexpr int BSWEEPS = (BELEMS + FA_WAVES * 64 * 8 - 1)
+                          / (FA_WAVES * 64 * 8);
+  bf16x8 qst[BSWEEPS], dst[BSWEEPS];
+  {
+    const int tid = threadIdx.x;
+    const int q0_base = first_qt * FB_T;
+#pragma unroll
+    for (int sw = 0; sw < BSWEEPS; ++sw) {
+      const int i = tid * 8 + sw * (FA_WAVES * 64 * 8);
+      if (i < BELEMS) {
+        const int qr = i / DP;
+        const int qc = i % DP;
+        const long qrg = min(q0_base + qr, sq - 1);
+        qst[sw] = fa_load8<D>(Qp + qrg * D, qc);
+        dst[sw] = fa_load8<D>(dOp + qrg * D, qc);
+      }
+    }
+  }
   for (int qt = first_qt; qt < n_q_tiles; ++qt) {
     const int q_base = qt * FB_T;
     __syncthreads();
     {
       const int tid = threadIdx.x;
-      for (int i = tid * 8; i < FB_T * DP; i += FA_WAVES * 64 * 8) {
-        const int qr = i / DP;
-        const int qc = i % DP;
-        const long qrg = min(q_base + qr, sq - 1);
-        bf16x8 qq = fa_load8<D>(Qp + qrg * D, qc);
-        *reinterpret_cast<bf16x8*>(q_raw + qr * SWB + kswz(qr, qc * 2)) = qq;
-        bf16x8 dd = fa_load8<D>(dOp + qrg * D, qc);
-        *reinterpret_cast<bf16x8*>(do_raw + qr * SWB + kswz(qr, qc * 2)) = dd;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          qt_lds[qc + j][qr ^ ((qc + j) & 0x18)] = qq[j];
-          dot_lds[qc + j][qr ^ ((qc + j) & 0x18)] = dd[j];
+      for (int sw = 0; sw < BSWEEPS; ++sw) {
+        const int i = tid * 8 + sw * (FA_WAVES * 64 * 8);
+        if (i < BELEMS) {
+          const int qr = i / DP;
+          const int qc = i % DP;
+          bf16x8 qq = qst[sw];
+          bf16x8 dd = dst[sw];
+          *reinterpret_cast<bf16x8*>(
+              q_raw + qr * SWB + kswz(qr, qc * 2)) = qq;
+          *reinterpret_cast<bf16x8*>(
+              do_raw + qr * SWB + kswz(qr, qc * 2)) = dd;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            qt_lds[qc + j][qr ^ ((qc + j) & 0x18)] = qq[j];
+            dot_lds[qc + j][qr ^ ((qc + j) & 0x18)] = dd[j];
+          }
         }
       }
     }
     __syncthreads();
+    if (qt + 1 < n_q_tiles) {
+      const int nb = (qt + 1) * FB_T;
+      const int tid = threadIdx.x;
+#pragma unroll
+      for (int sw = 0; sw < BSWEEPS; ++sw) {
+        const int i = tid * 8 + sw * (FA_WAVES * 64 * 8);
+        if (i < BELEMS) {
+          const int qr = i / DP;
+          const int qc = i % DP;
+          const long qrg = min(nb + qr, sq - 1);
+          qst[sw] = fa_load8<D>(Qp + qrg * D, qc);
+          dst[sw] = fa_load8<D>(dOp + qrg * D, qc);
+        }
+      }
+    }
 
     f32x4 st_acc[2], dpt_acc[2];
 #pragma unroll
