@@ -27,6 +27,8 @@ def main():
     p.add_argument("--new_tokens", type=int, default=64)
     p.add_argument("--batch", type=int, default=1)
     p.add_argument("--quant", choices=["bf16", "int8"], default="bf16")
+    p.add_argument("--graph", action="store_true",
+                   help="hipGraph-captured decode loop (serving fast path)")
     args = p.parse_args()
     assert torch.cuda.is_available()
 
@@ -52,12 +54,23 @@ def main():
     vocab = model.config.vocab_size
     ids = torch.randint(3, vocab, (args.batch, args.prompt_len),
                         device="cuda")
-    # warmup
-    model.generate(ids, max_new_tokens=4, do_sample=False)
-    torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    out = model.generate(ids, max_new_tokens=args.new_tokens, do_sample=False)
-    torch.cuda.synchronize()
+    if args.graph:
+        from fengshen_amd.serving.graphed_decode import GraphedDecoder
+        dec = GraphedDecoder(model, batch=args.batch,
+                             max_len=args.prompt_len + args.new_tokens + 8,
+                             max_new_tokens=args.new_tokens)
+        dec.generate(ids, max_new_tokens=4)  # warmup + capture
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        out = dec.generate(ids, max_new_tokens=args.new_tokens)
+        torch.cuda.synchronize()
+    else:
+        model.generate(ids, max_new_tokens=4, do_sample=False)  # warmup
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        out = model.generate(ids, max_new_tokens=args.new_tokens,
+                             do_sample=False)
+        torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     n_new = out.shape[1] - args.prompt_len
     print(json.dumps({
@@ -65,6 +78,7 @@ def main():
         "value": round(dt / n_new * 1000, 2),
         "unit": "ms/token",
         "quant": args.quant,
+        "graph": bool(args.graph),
         "batch": args.batch,
         "prompt_len": args.prompt_len,
         "new_tokens": n_new,
