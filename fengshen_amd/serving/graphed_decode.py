@@ -131,7 +131,8 @@ class GraphedDecoder:
         out = self.model(input_ids=prompt_ids.to(self.dev),
                          past_key_values=cache, use_cache=True)
         for i in range(len(self.k_cache)):
-            k, v = cache[i]
+            layer_cache = cache.layers[i]  # transformers >= 5 DynamicCache
+            k, v = layer_cache.keys, layer_cache.values
             self.k_cache[i][:, :, :plen].copy_(k)
             self.v_cache[i][:, :, :plen].copy_(v)
             self.k_cache[i][:, :, plen:].zero_()

@@ -15,3 +15,123 @@ def test_chinese_char_tokenize():
     # idempotent-ish: already separated text keeps tokens
     again = chinese_char_tokenize(out)
     assert "中" in again.split()
+
+
+# ---------------------------------------------------------------------------
+# TF checkpoint importer (ref convert_tf_checkpoint_to_pytorch.py; no TF in
+# the image — native TensorBundle reader).  The test writes a
+# spec-conformant LevelDB table + data shard by hand and reads it back.
+# ---------------------------------------------------------------------------
+def _varint(n):
+    out = b""
+    while True:
+        b = n & 0x7F
+        n >>= 7
+        if n:
+            out += bytes([b | 0x80])
+        else:
+            return out + bytes([b])
+
+
+def _pb_tag(field, wire):
+    return _varint((field << 3) | wire)
+
+
+def _bundle_entry(dtype, shape, offset, size):
+    shape_msg = b""
+    for d in shape:
+        dim = _pb_tag(1, 0) + _varint(d)
+        shape_msg += _pb_tag(2, 2) + _varint(len(dim)) + dim
+    msg = _pb_tag(1, 0) + _varint(dtype)
+    msg += _pb_tag(2, 2) + _varint(len(shape_msg)) + shape_msg
+    msg += _pb_tag(4, 0) + _varint(offset)
+    msg += _pb_tag(5, 0) + _varint(size)
+    return msg
+
+
+def _leveldb_block(entries):
+    """One block, no prefix sharing, single restart at 0."""
+    import struct
+    body = b""
+    for k, v in entries:
+        body += _varint(0) + _varint(len(k)) + _varint(len(v)) + k + v
+    body += struct.pack("<I", 0) + struct.pack("<I", 1)
+    return body
+
+
+def _write_tf_checkpoint(tmp_path, tensors):
+    """tensors: {name: np.ndarray (float32)}"""
+    import struct
+    import numpy as np
+    data = b""
+    entries = []
+    for name in sorted(tensors):
+        arr = np.ascontiguousarray(tensors[name], dtype=np.float32)
+        entries.append((name.encode(), _bundle_entry(
+            1, arr.shape, len(data), arr.nbytes)))
+        data += arr.tobytes()
+    (tmp_path / "model.ckpt.data-00000-of-00001").write_bytes(data)
+
+    blk = _leveldb_block(entries)
+    out = blk + b"\x00" + struct.pack("<I", 0)  # type + crc (crc unchecked)
+    data_handle = _varint(0) + _varint(len(blk))
+    # index block: one entry pointing at the data block
+    idx = _leveldb_block([(b"\xff", data_handle)])
+    idx_off = len(out)
+    out += idx + b"\x00" + struct.pack("<I", 0)
+    meta_off = len(out)
+    meta = _leveldb_block([])
+    out += meta + b"\x00" + struct.pack("<I", 0)
+    footer = (_varint(meta_off) + _varint(len(meta))
+              + _varint(idx_off) + _varint(len(idx)))
+    footer += b"\x00" * (40 - len(footer))
+    out += footer + struct.pack("<Q", 0xDB4775248B80FB57)
+    (tmp_path / "model.ckpt.index").write_bytes(out)
+    return str(tmp_path / "model.ckpt")
+
+
+def test_tf_checkpoint_reader(tmp_path):
+    import numpy as np
+    from fengshen_amd.utils.tf_checkpoint import TFCheckpointReader
+    rng = np.random.RandomState(0)
+    tensors = {
+        "bert/encoder/layer_0/attention/self/query/kernel":
+            rng.randn(8, 8).astype(np.float32),
+        "bert/embeddings/word_embeddings":
+            rng.randn(16, 8).astype(np.float32),
+        "global_step": np.array([7.0], dtype=np.float32),
+    }
+    prefix = _write_tf_checkpoint(tmp_path, tensors)
+    r = TFCheckpointReader(prefix)
+    assert set(r.variable_names()) == set(tensors)
+    for name, arr in tensors.items():
+        got = r.load_variable(name)
+        assert got.shape == arr.shape
+        assert np.allclose(got, arr)
+
+
+def test_tf_bert_name_mapping(tmp_path):
+    import numpy as np
+    from fengshen_amd.utils.tf_checkpoint import (
+        convert_tf_bert_to_state_dict)
+    rng = np.random.RandomState(1)
+    kern = rng.randn(4, 6).astype(np.float32)
+    tensors = {
+        "bert/encoder/layer_0/attention/self/query/kernel": kern,
+        "bert/encoder/layer_0/attention/output/LayerNorm/gamma":
+            rng.randn(4).astype(np.float32),
+        "bert/embeddings/word_embeddings":
+            rng.randn(16, 4).astype(np.float32),
+        "cls/predictions/output_bias": rng.randn(16).astype(np.float32),
+        "bert/adam_m/skip_me": rng.randn(2).astype(np.float32),
+    }
+    prefix = _write_tf_checkpoint(tmp_path, tensors)
+    sd = convert_tf_bert_to_state_dict(prefix)
+    assert "bert.encoder.layer.0.attention.self.query.weight" in sd
+    # TF kernels are [in, out]; torch wants [out, in]
+    assert np.allclose(
+        sd["bert.encoder.layer.0.attention.self.query.weight"], kern.T)
+    assert "bert.encoder.layer.0.attention.output.LayerNorm.weight" in sd
+    assert "bert.embeddings.word_embeddings.weight" in sd
+    assert "cls.predictions.bias" in sd
+    assert not any("adam_m" in k for k in sd)
