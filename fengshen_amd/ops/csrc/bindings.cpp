@@ -56,6 +56,8 @@ void fs_flash_attn_bwd(const void*, const void*, const void*, const void*,
                        float, unsigned long long, hipStream_t);
 void fs_w8_gemv(const void*, const float*, const void*, void*, int, int, int,
                 hipStream_t);
+void fs_flash_attn_fwd_v3(const void*, const void*, const void*, void*,
+                          float*, int, int, int, float, hipStream_t);
 }
 
 // ---------------------------------------------------------------------------
@@ -347,6 +349,19 @@ static at::Tensor w8_gemv(at::Tensor q8, at::Tensor scale, at::Tensor x) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("w8_gemv", &w8_gemv);
   mod.def("flash_attn_fwd", &flash_attn_fwd);
+  mod.def("flash_attn_fwd_v3", [](at::Tensor q, at::Tensor k, at::Tensor v,
+                                  double scale) {
+    TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+    TORCH_CHECK(q.scalar_type() == at::kBFloat16 && q.size(3) == 128);
+    const int b = q.size(0), h = q.size(1), s = q.size(2);
+    TORCH_CHECK(s % 64 == 0 && s >= 64);
+    auto o = at::empty_like(q);
+    auto lse = at::empty({b, h, s}, q.options().dtype(at::kFloat));
+    fs_flash_attn_fwd_v3(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                         o.data_ptr(), lse.data_ptr<float>(), b, h, s,
+                         (float)scale, cur_stream());
+    return std::vector<at::Tensor>{o, lse};
+  });
   mod.def("flash_attn_bwd", &flash_attn_bwd);
   mod.def("rms_norm_fwd", &rms_norm_fwd);
   mod.def("rms_norm_bwd", &rms_norm_bwd);

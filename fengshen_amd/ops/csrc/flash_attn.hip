@@ -971,3 +971,247 @@ extern "C" void fs_flash_attn_bwd(const void* q, const void* k, const void* v,
   }
 #undef BWD_CASE
 }
+
+// ===========================================================================
+// FORWARD v3 — 8-wave swapped-QK^T schedule (guide §B 8-warp ladder)
+// ===========================================================================
+// Causal, D=128 (the flagship LLaMA shape).  Differences vs the general
+// kernel above:
+//   - wave owns 32 q rows (QBLK = 256/block): K/V LDS traffic is amortized
+//     over 2x more query rows;
+//   - mfma_f32_32x32x16_bf16 with SWAPPED operands (A = K-tile, B = Q):
+//     each lane holds a full 32-wide P-row slice for ONE q row in
+//     registers, so the row max/sum are 31 in-lane ops + one cross-half
+//     shuffle — no 4-round shuffle-reduce trees, and softmax state m/l is
+//     a per-lane scalar;
+//   - P crosses to the PV A-operand through a per-wave LDS exchange
+//     (bf16, 8B stores / 16B reads).
+// Layouts verified on HW by scripts/mfma32_probe.hip.
+
+#define V3_QBLK 256          // 8 waves x 32 q rows
+#define V3_KVBLK 64
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+__global__ __launch_bounds__(FA_WAVES * 64)
+void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
+                              const bf16_t* __restrict__ K,
+                              const bf16_t* __restrict__ V,
+                              bf16_t* __restrict__ O,
+                              float* __restrict__ LSE,
+                              int b, int h, int s, float scale) {
+  constexpr int D = 128;
+  constexpr int SWB = 256;  // fa_swb(128)
+
+  __shared__ char k_raw[V3_KVBLK * SWB];                // swizzled K rows
+  __shared__ short vt_lds[D][V3_KVBLK + FA_VPAD];       // V transposed
+  __shared__ short p_x[FA_WAVES][32][V3_KVBLK + FA_VPAD];  // P exchange
+  __shared__ float a_x[FA_WAVES][32];                   // alpha / l bcast
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int hi = lane >> 5;      // lane half
+  const int ln = lane & 31;
+  const long bh = (long)blockIdx.z * h + blockIdx.y;
+  const bf16_t* Qp = Q + bh * s * D;
+  const bf16_t* Kp = K + bh * s * D;
+  const bf16_t* Vp = V + bh * s * D;
+  bf16_t* Op = O + bh * s * D;
+
+  const int q0w = blockIdx.x * V3_QBLK + wave * 32;
+  const bool q_active = q0w < s;
+  const int q0c = q_active ? q0w : s - 32;
+  const int my_q = q0c + ln;      // softmax row this lane owns
+
+  // ---- Q -> B-fragments (8 K=16 chunks), pre-scaled ---------------------
+  bf16x8 q_frag[8];
+#pragma unroll
+  for (int c = 0; c < 8; ++c) {
+    bf16x8 raw = *reinterpret_cast<const bf16x8*>(
+        Qp + (long)my_q * D + c * 16 + hi * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      unsigned short u = (unsigned short)raw[j];
+      float f = __uint_as_float(((unsigned int)u) << 16) * scale;
+      raw[j] = fa_bf16bits(f);
+    }
+    q_frag[c] = raw;
+  }
+
+  float m_run = -INFINITY, l_run = 0.f;   // per-lane: row my_q
+  f32x16 o_acc[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) o_acc[t][r] = 0.f;
+  }
+
+  const int q_hi_row = min(blockIdx.x * V3_QBLK + V3_QBLK, s) - 1;
+  const int n_kv_tiles = (q_hi_row / V3_KVBLK) + 1;
+
+  // async-STAGE staging registers (64*128 elems / 512 threads / 8 = 2)
+  const int tid = threadIdx.x;
+  bf16x8 k_reg[2], v_reg[2];
+#pragma unroll
+  for (int sweep = 0; sweep < 2; ++sweep) {
+    const int i = tid * 8 + sweep * 4096;
+    const int kr = i / D;
+    const int kc = i % D;
+    k_reg[sweep] = *reinterpret_cast<const bf16x8*>(Kp + (long)kr * D + kc);
+    v_reg[sweep] = *reinterpret_cast<const bf16x8*>(Vp + (long)kr * D + kc);
+  }
+
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int k_base = kt * V3_KVBLK;
+    __syncthreads();
+#pragma unroll
+    for (int sweep = 0; sweep < 2; ++sweep) {
+      const int i = tid * 8 + sweep * 4096;
+      const int kr = i / D;
+      const int kc = i % D;
+      *reinterpret_cast<bf16x8*>(k_raw + kr * SWB + kswz(kr, kc * 2)) =
+          k_reg[sweep];
+      bf16x8 vv = v_reg[sweep];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        vt_lds[kc + j][kr ^ ((kc + j) & 0x38)] = vv[j];
+    }
+    __syncthreads();
+    if (kt + 1 < n_kv_tiles) {
+      const int nb = (kt + 1) * V3_KVBLK;
+#pragma unroll
+      for (int sweep = 0; sweep < 2; ++sweep) {
+        const int i = tid * 8 + sweep * 4096;
+        const int kr = i / D;
+        const int kc = i % D;
+        k_reg[sweep] = *reinterpret_cast<const bf16x8*>(
+            Kp + (long)(nb + kr) * D + kc);
+        v_reg[sweep] = *reinterpret_cast<const bf16x8*>(
+            Vp + (long)(nb + kr) * D + kc);
+      }
+    }
+
+    // ---- S^T = K_tile @ (sQ)^T : two 32x32 kv-tiles ---------------------
+    f32x16 st[2];
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) st[nt][r] = 0.f;
+    }
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+      const int krow = nt * 32 + ln;
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            k_raw + krow * SWB + kswz(krow, (c * 16 + hi * 8) * 2));
+        st[nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, q_frag[c],
+                                                         st[nt], 0, 0, 0);
+      }
+    }
+    // lane now holds S^T[kv][q = my row]: kv = k_base + nt*32 + crow(r,hi)
+    // with crow(r,hi) = (r&3) + 8*(r>>2) + 4*hi
+
+    // ---- causal mask + in-register online softmax ----------------------
+    float p[32];
+    float tmax = -INFINITY;
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kcol = k_base + nt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float v = st[nt][r];
+        if (kcol > my_q) v = -INFINITY;
+        p[nt * 16 + r] = v;
+        tmax = fmaxf(tmax, v);
+      }
+    }
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));  // combine lane halves
+    const float m_new = fmaxf(m_run, tmax);
+    const float alpha = (m_run == -INFINITY) ? 0.f : __expf(m_run - m_new);
+    const bool rescale = (m_new != m_run);
+    m_run = m_new;
+    float rs = 0.f;
+#pragma unroll
+    for (int i = 0; i < 32; ++i) {
+      float e = (p[i] == -INFINITY) ? 0.f : __expf(p[i] - m_new);
+      p[i] = e;
+      rs += e;
+    }
+    rs += __shfl_xor(rs, 32, 64);
+    l_run = l_run * alpha + rs;
+
+    // broadcast alpha to the lanes holding this row's O accumulators
+    if (hi == 0) a_x[wave][ln] = alpha;
+    __builtin_amdgcn_s_waitcnt(0);
+    if (__any(rescale)) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float al = a_x[wave][(r & 3) + 8 * (r >> 2) + 4 * hi];
+#pragma unroll
+        for (int t = 0; t < 4; ++t) o_acc[t][r] *= al;
+      }
+    }
+
+    // ---- P -> per-wave LDS exchange (8B runs of 4 kv) -------------------
+#pragma unroll
+    for (int nt = 0; nt < 2; ++nt) {
+#pragma unroll
+      for (int g = 0; g < 4; ++g) {
+        short pk[4] __attribute__((aligned(8)));
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          pk[j] = fa_bf16bits(p[nt * 16 + g * 4 + j]);
+        *reinterpret_cast<long*>(
+            &p_x[wave][ln][nt * 32 + 8 * g + 4 * hi]) =
+            *reinterpret_cast<const long*>(pk);
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+
+    // ---- PV: O[q][d] += P[q][kv] @ V[kv][d] -----------------------------
+#pragma unroll
+    for (int c2 = 0; c2 < 4; ++c2) {  // kv chunks of 16
+      bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+          &p_x[wave][ln][c2 * 16 + hi * 8]);
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {   // d cols, 32 each
+        const int dcol = t * 32 + ln;
+        bf16x8 vb = *reinterpret_cast<const bf16x8*>(
+            &vt_lds[dcol][(c2 * 16 + hi * 8) ^ (dcol & 0x38)]);
+        o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, vb, o_acc[t],
+                                                           0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue ---------------------------------------------------------
+  if (!q_active) return;
+  // broadcast 1/l to o-accumulator lanes via the alpha buffer
+  if (hi == 0) a_x[wave][ln] = (l_run > 0.f) ? 1.f / l_run : 0.f;
+  __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int qrow = q0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
+    const float inv_l = a_x[wave][(r & 3) + 8 * (r >> 2) + 4 * hi];
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      Op[(long)qrow * D + t * 32 + ln] =
+          __float2bfloat16(o_acc[t][r] * inv_l);
+    }
+  }
+  if (hi == 0 && LSE) {
+    LSE[bh * s + q0w + ln] = m_run + logf(fmaxf(l_run, 1e-30f));
+  }
+}
+
+extern "C" void fs_flash_attn_fwd_v3(const void* q, const void* k,
+                                     const void* v, void* o, float* lse,
+                                     int b, int h, int s, float scale,
+                                     hipStream_t stream) {
+  dim3 grid((s + V3_QBLK - 1) / V3_QBLK, h, b);
+  dim3 block(FA_WAVES * 64);
+  hipLaunchKernelGGL(flash_attn_fwd_v3_kernel, grid, block, 0, stream,
+                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
+                     (bf16_t*)o, lse, b, h, s, scale);
+}
