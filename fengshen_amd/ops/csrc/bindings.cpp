@@ -58,6 +58,9 @@ void fs_w8_gemv(const void*, const float*, const void*, void*, int, int, int,
                 hipStream_t);
 void fs_flash_attn_fwd_v3(const void*, const void*, const void*, void*,
                           float*, int, int, int, float, hipStream_t);
+void fs_flash_attn_bwd_v3(const void*, const void*, const void*, const void*,
+                          const void*, const float*, void*, void*, void*,
+                          float*, int, int, int, float, hipStream_t);
 }
 
 // ---------------------------------------------------------------------------
@@ -349,6 +352,22 @@ static at::Tensor w8_gemv(at::Tensor q8, at::Tensor scale, at::Tensor x) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("w8_gemv", &w8_gemv);
   mod.def("flash_attn_fwd", &flash_attn_fwd);
+  mod.def("flash_attn_bwd_v3", [](at::Tensor q, at::Tensor k, at::Tensor v,
+                                  at::Tensor o, at::Tensor dout,
+                                  at::Tensor lse, double scale) {
+    const int b = q.size(0), h = q.size(1), s = q.size(2);
+    TORCH_CHECK(q.size(3) == 128 && s % 64 == 0);
+    auto dq = at::empty_like(q);
+    auto dk = at::empty_like(k);
+    auto dv = at::empty_like(v);
+    auto delta = at::empty({b, h, s}, q.options().dtype(at::kFloat));
+    fs_flash_attn_bwd_v3(q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                         o.data_ptr(), dout.contiguous().data_ptr(),
+                         lse.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
+                         dv.data_ptr(), delta.data_ptr<float>(), b, h, s,
+                         (float)scale, cur_stream());
+    return std::vector<at::Tensor>{dq, dk, dv};
+  });
   mod.def("flash_attn_fwd_v3", [](at::Tensor q, at::Tensor k, at::Tensor v,
                                   double scale) {
     TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());

@@ -1215,3 +1215,364 @@ extern "C" void fs_flash_attn_fwd_v3(const void* q, const void* k,
                      (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
                      (bf16_t*)o, lse, b, h, s, scale);
 }
+
+// ===========================================================================
+// BACKWARD v3 — swapped 32x32 schedule (causal, D=128)
+// ===========================================================================
+// Same design move as forward v3: orient every S-like MFMA so the softmax
+// row variable (q for dq, q-column for dkv) is lane-local — lse/delta
+// become per-lane scalar loads, masks are per-register compares, and the
+// only cross-lane traffic is the compact P/dS LDS exchange.
+
+#define B3_FB 32  // streamed tile width
+
+// q-major: wave owns 32 q rows (block 256), streams KV in 32-tiles.
+__global__ __launch_bounds__(FA_WAVES * 64)
+void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
+                                 const bf16_t* __restrict__ K,
+                                 const bf16_t* __restrict__ V,
+                                 const bf16_t* __restrict__ dO,
+                                 const float* __restrict__ LSE,
+                                 const float* __restrict__ Delta,
+                                 bf16_t* __restrict__ dQ,
+                                 int b, int h, int s, float scale) {
+  constexpr int D = 128;
+  constexpr int SWB = 256;
+
+  __shared__ char k_raw[B3_FB * SWB];                  // K rows, swizzled
+  __shared__ char v_raw[B3_FB * SWB];                  // V rows, swizzled
+  __shared__ short kt_lds[D][B3_FB + FA_VPAD];         // K^T (kv-swizzled)
+  __shared__ short p_x[FA_WAVES][32][B3_FB + FA_VPAD];
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int hi = lane >> 5;
+  const int ln = lane & 31;
+  const long bh = (long)blockIdx.z * h + blockIdx.y;
+  const bf16_t* Qp = Q + bh * s * D;
+  const bf16_t* Kp = K + bh * s * D;
+  const bf16_t* Vp = V + bh * s * D;
+  const bf16_t* dOp = dO + bh * s * D;
+  bf16_t* dQp = dQ + bh * s * D;
+  const float* lse = LSE + bh * s;
+  const float* dlt = Delta + bh * s;
+
+  const int q0w = blockIdx.x * V3_QBLK + wave * 32;
+  const bool q_active = q0w < s;
+  const int q0c = q_active ? q0w : s - 32;
+  const int my_q = q0c + ln;
+
+  // Q (pre-scaled) and dO rows in registers (B-fragments)
+  bf16x8 q_frag[8], do_frag[8];
+#pragma unroll
+  for (int c = 0; c < 8; ++c) {
+    bf16x8 raw = *reinterpret_cast<const bf16x8*>(
+        Qp + (long)my_q * D + c * 16 + hi * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      unsigned short u = (unsigned short)raw[j];
+      raw[j] = fa_bf16bits(
+          __uint_as_float(((unsigned int)u) << 16) * scale);
+    }
+    q_frag[c] = raw;
+    do_frag[c] = *reinterpret_cast<const bf16x8*>(
+        dOp + (long)my_q * D + c * 16 + hi * 8);
+  }
+  const float lse_r = lse[my_q];
+  const float dlt_r = dlt[my_q];
+
+  f32x16 dq_acc[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dq_acc[t][r] = 0.f;
+  }
+
+  const int q_hi_row = min(blockIdx.x * V3_QBLK + V3_QBLK, s) - 1;
+  const int n_kv_tiles = (q_hi_row / B3_FB) + 1;
+
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int k_base = kt * B3_FB;
+    __syncthreads();
+    {
+      const int tid = threadIdx.x;
+      for (int i = tid * 8; i < B3_FB * D; i += FA_WAVES * 64 * 8) {
+        const int kr = i / D;
+        const int kc = i % D;
+        bf16x8 kk = *reinterpret_cast<const bf16x8*>(
+            Kp + (long)(k_base + kr) * D + kc);
+        *reinterpret_cast<bf16x8*>(k_raw + kr * SWB + kswz(kr, kc * 2)) = kk;
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
+            Vp + (long)(k_base + kr) * D + kc);
+        *reinterpret_cast<bf16x8*>(v_raw + kr * SWB + kswz(kr, kc * 2)) = vv;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          kt_lds[kc + j][kr ^ ((kc + j) & 0x18)] = kk[j];
+      }
+    }
+    __syncthreads();
+
+    // S^T = K (sQ)^T and dP^T = V dO^T : one 32x32 kv-tile each
+    f32x16 st, dpt;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      st[r] = 0.f;
+      dpt[r] = 0.f;
+    }
+    {
+      const int krow = ln;
+#pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+            k_raw + krow * SWB + kswz(krow, (c * 16 + hi * 8) * 2));
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, q_frag[c], st,
+                                                     0, 0, 0);
+        bf16x8 va = *reinterpret_cast<const bf16x8*>(
+            v_raw + krow * SWB + kswz(krow, (c * 16 + hi * 8) * 2));
+        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, do_frag[c], dpt,
+                                                      0, 0, 0);
+      }
+    }
+    // lane holds, for its OWN q row: kv = k_base + crow(r,hi)
+
+    // dS = P * (dP - delta) * scale, in-register; exchange by q row
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      short pk[4] __attribute__((aligned(8)));
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int r = g * 4 + j;
+        const int kcol = k_base + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float pv = (kcol > my_q) ? 0.f : __expf(st[r] - lse_r);
+        pk[j] = fa_bf16bits(pv * (dpt[r] - dlt_r) * scale);
+      }
+      *reinterpret_cast<long*>(&p_x[wave][ln][8 * g + 4 * hi]) =
+          *reinterpret_cast<const long*>(pk);
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+
+    // dQ += dS @ K : A = dS [32q x 16kv], B = K^T
+#pragma unroll
+    for (int c2 = 0; c2 < 2; ++c2) {
+      bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+          &p_x[wave][ln][c2 * 16 + hi * 8]);
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const int dcol = t * 32 + ln;
+        bf16x8 kb = *reinterpret_cast<const bf16x8*>(
+            &kt_lds[dcol][(c2 * 16 + hi * 8) ^ (dcol & 0x18)]);
+        dq_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, kb,
+                                                            dq_acc[t],
+                                                            0, 0, 0);
+      }
+    }
+  }
+
+  if (!q_active) return;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int qrow = q0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
+#pragma unroll
+    for (int t = 0; t < 4; ++t)
+      dQp[(long)qrow * D + t * 32 + ln] = __float2bfloat16(dq_acc[t][r]);
+  }
+}
+
+// KV-major: wave owns 32 kv rows (block 256), streams Q/dO in 32-tiles.
+// VGPR budget forces a T16-style time-share: the wave's own K rows load
+// into a register block for the St MFMAs, then V overwrites the same
+// block for the dPt MFMAs (rows are L2-hot; the loads hide under MFMA).
+// The softmax scale folds into the exp (st*scale - lse) so K stays raw.
+__global__ __launch_bounds__(FA_WAVES * 64)
+void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
+                                  const bf16_t* __restrict__ K,
+                                  const bf16_t* __restrict__ V,
+                                  const bf16_t* __restrict__ dO,
+                                  const float* __restrict__ LSE,
+                                  const float* __restrict__ Delta,
+                                  bf16_t* __restrict__ dK,
+                                  bf16_t* __restrict__ dV,
+                                  int b, int h, int s, float scale) {
+  constexpr int D = 128;
+  constexpr int SWB = 256;
+
+  __shared__ char q_raw[B3_FB * SWB];                   // Q rows, swizzled
+  __shared__ char do_raw[B3_FB * SWB];                  // dO rows, swizzled
+  __shared__ short qt_lds[D][B3_FB + FA_VPAD];          // Q^T
+  __shared__ short dot_lds[D][B3_FB + FA_VPAD];         // dO^T
+  __shared__ short p_x[FA_WAVES][32][B3_FB + FA_VPAD];  // by kv row
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int hi = lane >> 5;
+  const int ln = lane & 31;
+  const long bh = (long)blockIdx.z * h + blockIdx.y;
+  const bf16_t* Qp = Q + bh * s * D;
+  const bf16_t* Kp = K + bh * s * D;
+  const bf16_t* Vp = V + bh * s * D;
+  const bf16_t* dOp = dO + bh * s * D;
+  bf16_t* dKp = dK + bh * s * D;
+  bf16_t* dVp = dV + bh * s * D;
+  const float* lse = LSE + bh * s;
+  const float* dlt = Delta + bh * s;
+
+  const int kv0_blk = blockIdx.x * V3_QBLK;
+  const int kv0w = kv0_blk + wave * 32;
+  const bool kv_active = kv0w < s;
+  const int kv0c = kv_active ? kv0w : s - 32;
+  const int my_kv = kv0c + ln;
+
+  // dv/dk in the 32x32 layout: 4 d-tiles x 16 f32 each
+  f32x16 dv_acc[4], dk_acc[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      dv_acc[t][r] = 0.f;
+      dk_acc[t][r] = 0.f;
+    }
+  }
+
+  const int first_qt = kv0_blk / B3_FB;
+  const int n_q_tiles = s / B3_FB;
+
+  for (int qt = first_qt; qt < n_q_tiles; ++qt) {
+    const int q_base = qt * B3_FB;
+    __syncthreads();
+    {
+      const int tid = threadIdx.x;
+      for (int i = tid * 8; i < B3_FB * D; i += FA_WAVES * 64 * 8) {
+        const int qr = i / D;
+        const int qc = i % D;
+        bf16x8 qq = *reinterpret_cast<const bf16x8*>(
+            Qp + (long)(q_base + qr) * D + qc);
+        *reinterpret_cast<bf16x8*>(q_raw + qr * SWB + kswz(qr, qc * 2)) = qq;
+        bf16x8 dd = *reinterpret_cast<const bf16x8*>(
+            dOp + (long)(q_base + qr) * D + qc);
+        *reinterpret_cast<bf16x8*>(do_raw + qr * SWB + kswz(qr, qc * 2)) =
+            dd;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          qt_lds[qc + j][qr ^ ((qc + j) & 0x18)] = qq[j];
+          dot_lds[qc + j][qr ^ ((qc + j) & 0x18)] = dd[j];
+        }
+      }
+    }
+    __syncthreads();
+
+    // St = K Q^T (raw K; scale folded into the exp below)
+    f32x16 st;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) st[r] = 0.f;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      bf16x8 ka = *reinterpret_cast<const bf16x8*>(
+          Kp + (long)my_kv * D + c * 16 + hi * 8);
+      bf16x8 qb = *reinterpret_cast<const bf16x8*>(
+          q_raw + ln * SWB + kswz(ln, (c * 16 + hi * 8) * 2));
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qb, st, 0, 0, 0);
+    }
+    // dPt = V dO^T (same register pattern, V overwrites K's slots)
+    f32x16 dpt;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dpt[r] = 0.f;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      bf16x8 va = *reinterpret_cast<const bf16x8*>(
+          Vp + (long)my_kv * D + c * 16 + hi * 8);
+      bf16x8 db = *reinterpret_cast<const bf16x8*>(
+          do_raw + ln * SWB + kswz(ln, (c * 16 + hi * 8) * 2));
+      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(va, db, dpt, 0, 0, 0);
+    }
+
+    const int qcol = q_base + ln;
+    const float lse_c = lse[qcol];
+    const float dlt_c = dlt[qcol];
+
+    // Pt by kv row into the exchange buffer (2B scatter stores)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kvrow_loc = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      float pv = (qcol < kv0c + kvrow_loc)
+                     ? 0.f : __expf(st[r] * scale - lse_c);
+      st[r] = pv;  // reuse st as Pt storage
+      p_x[wave][kvrow_loc][ln] = fa_bf16bits(pv);
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+
+    // dV += Pt dO : A = Pt [32kv x 16q-chunk] own-row, B = dO^T
+#pragma unroll
+    for (int c2 = 0; c2 < 2; ++c2) {
+      bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+          &p_x[wave][ln][c2 * 16 + hi * 8]);
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const int dcol = t * 32 + ln;
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+            &dot_lds[dcol][(c2 * 16 + hi * 8) ^ (dcol & 0x18)]);
+        dv_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            pa, bfrag, dv_acc[t], 0, 0, 0);
+      }
+    }
+
+    // dSt = Pt * (dPt - delta) * scale; exchange and accumulate dK
+    __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kvrow_loc = (r & 3) + 8 * (r >> 2) + 4 * hi;
+      p_x[wave][kvrow_loc][ln] =
+          fa_bf16bits(st[r] * (dpt[r] - dlt_c) * scale);
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+#pragma unroll
+    for (int c2 = 0; c2 < 2; ++c2) {
+      bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+          &p_x[wave][ln][c2 * 16 + hi * 8]);
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const int dcol = t * 32 + ln;
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+            &qt_lds[dcol][(c2 * 16 + hi * 8) ^ (dcol & 0x18)]);
+        dk_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            pa, bfrag, dk_acc[t], 0, 0, 0);
+      }
+    }
+  }
+
+  if (!kv_active) return;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int kvrow = kv0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      dKp[(long)kvrow * D + t * 32 + ln] = __float2bfloat16(dk_acc[t][r]);
+      dVp[(long)kvrow * D + t * 32 + ln] = __float2bfloat16(dv_acc[t][r]);
+    }
+  }
+}
+
+extern "C" void fs_flash_attn_bwd_v3(const void* q, const void* k,
+                                     const void* v, const void* o,
+                                     const void* dout, const float* lse,
+                                     void* dq, void* dk, void* dv,
+                                     float* delta_ws, int b, int h, int s,
+                                     float scale, hipStream_t stream) {
+  const long rows = (long)b * h * s;
+  {
+    long blocks = (rows + 3) / 4;
+    if (blocks > 4096) blocks = 4096;
+    hipLaunchKernelGGL((flash_delta_kernel<128>), dim3((unsigned)blocks),
+                       dim3(256), 0, stream, (const bf16_t*)dout,
+                       (const bf16_t*)o, delta_ws, rows);
+  }
+  dim3 grid((s + V3_QBLK - 1) / V3_QBLK, h, b);
+  dim3 block(FA_WAVES * 64);
+  hipLaunchKernelGGL(flash_attn_bwd_dq_v3_kernel, grid, block, 0, stream,
+                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
+                     (const bf16_t*)dout, lse, delta_ws, (bf16_t*)dq,
+                     b, h, s, scale);
+  hipLaunchKernelGGL(flash_attn_bwd_dkv_v3_kernel, grid, block, 0, stream,
+                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
+                     (const bf16_t*)dout, lse, delta_ws, (bf16_t*)dk,
+                     (bf16_t*)dv, b, h, s, scale);
+}
