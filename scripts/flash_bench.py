@@ -58,3 +58,13 @@ for _ in range(N):
 torch.cuda.synchronize()
 dt3 = (time.perf_counter() - t0) / N
 print(f"flash fwd v3: {dt3*1000:.2f} ms, {flops/dt3/1e12:.1f} TF")
+
+# v3 fwd+bwd timing (routing picks v3 for causal d=128)
+torch.cuda.synchronize()
+t0 = time.perf_counter()
+for _ in range(N2):
+    out = flash_attention(q, k, v, scale)
+    torch.autograd.backward(out, gy)
+torch.cuda.synchronize()
+dt4 = (time.perf_counter() - t0) / N2
+print(f"flash fwd+bwd (v3 routed): {dt4*1000:.2f} ms, {flops*3.5/dt4/1e12:.1f} TF-equiv")
