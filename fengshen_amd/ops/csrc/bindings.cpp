@@ -57,10 +57,12 @@ void fs_flash_attn_bwd(const void*, const void*, const void*, const void*,
 void fs_w8_gemv(const void*, const float*, const void*, void*, int, int, int,
                 hipStream_t);
 void fs_flash_attn_fwd_v3(const void*, const void*, const void*, void*,
-                          float*, int, int, int, float, hipStream_t);
+                          float*, const int*, int, int, int, int, int, float,
+                          float, unsigned long long, hipStream_t);
 void fs_flash_attn_bwd_v3(const void*, const void*, const void*, const void*,
                           const void*, const float*, void*, void*, void*,
-                          float*, int, int, int, float, hipStream_t);
+                          float*, const int*, int, int, int, int, int, float,
+                          float, unsigned long long, hipStream_t);
 }
 
 // ---------------------------------------------------------------------------
@@ -354,9 +356,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("flash_attn_fwd", &flash_attn_fwd);
   mod.def("flash_attn_bwd_v3", [](at::Tensor q, at::Tensor k, at::Tensor v,
                                   at::Tensor o, at::Tensor dout,
-                                  at::Tensor lse, double scale) {
-    const int b = q.size(0), h = q.size(1), s = q.size(2);
-    TORCH_CHECK(q.size(3) == 128 && s % 64 == 0);
+                                  at::Tensor lse, double scale, bool causal,
+                                  c10::optional<at::Tensor> klens,
+                                  double drop_p, int64_t seed) {
+    const int b = q.size(0), h = q.size(1), s = q.size(2), d = q.size(3);
+    TORCH_CHECK((d == 128 || d == 64) && s % 64 == 0);
+    const int* klp = nullptr;
+    if (klens.has_value()) klp = klens->data_ptr<int>();
     auto dq = at::empty_like(q);
     auto dk = at::empty_like(k);
     auto dv = at::empty_like(v);
@@ -364,21 +370,32 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     fs_flash_attn_bwd_v3(q.data_ptr(), k.data_ptr(), v.data_ptr(),
                          o.data_ptr(), dout.contiguous().data_ptr(),
                          lse.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
-                         dv.data_ptr(), delta.data_ptr<float>(), b, h, s,
-                         (float)scale, cur_stream());
+                         dv.data_ptr(), delta.data_ptr<float>(), klp, b, h,
+                         s, d, causal ? 1 : 0, (float)scale, (float)drop_p,
+                         (unsigned long long)seed, cur_stream());
     return std::vector<at::Tensor>{dq, dk, dv};
   });
   mod.def("flash_attn_fwd_v3", [](at::Tensor q, at::Tensor k, at::Tensor v,
-                                  double scale) {
+                                  double scale, bool causal,
+                                  c10::optional<at::Tensor> klens,
+                                  double drop_p, int64_t seed) {
     TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
-    TORCH_CHECK(q.scalar_type() == at::kBFloat16 && q.size(3) == 128);
-    const int b = q.size(0), h = q.size(1), s = q.size(2);
-    TORCH_CHECK(s % 64 == 0 && s >= 64);
+    TORCH_CHECK(q.scalar_type() == at::kBFloat16);
+    const int b = q.size(0), h = q.size(1), s = q.size(2), d = q.size(3);
+    TORCH_CHECK((d == 128 || d == 64) && s % 64 == 0 && s >= 64);
+    TORCH_CHECK(k.size(2) == s, "v3 is self-attention (sq == sk)");
+    const int* klp = nullptr;
+    if (klens.has_value()) {
+      TORCH_CHECK(klens->scalar_type() == at::kInt
+                  && klens->numel() == b && klens->is_contiguous());
+      klp = klens->data_ptr<int>();
+    }
     auto o = at::empty_like(q);
     auto lse = at::empty({b, h, s}, q.options().dtype(at::kFloat));
     fs_flash_attn_fwd_v3(q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                         o.data_ptr(), lse.data_ptr<float>(), b, h, s,
-                         (float)scale, cur_stream());
+                         o.data_ptr(), lse.data_ptr<float>(), klp, b, h, s,
+                         d, causal ? 1 : 0, (float)scale, (float)drop_p,
+                         (unsigned long long)seed, cur_stream());
     return std::vector<at::Tensor>{o, lse};
   });
   mod.def("flash_attn_bwd", &flash_attn_bwd);

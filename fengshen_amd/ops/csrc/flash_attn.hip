@@ -993,15 +993,20 @@ extern "C" void fs_flash_attn_bwd(const void* q, const void* k, const void* v,
 
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
+template <int D, bool CAUSAL>
 __global__ __launch_bounds__(FA_WAVES * 64)
 void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
                               const bf16_t* __restrict__ K,
                               const bf16_t* __restrict__ V,
                               bf16_t* __restrict__ O,
                               float* __restrict__ LSE,
-                              int b, int h, int s, float scale) {
-  constexpr int D = 128;
-  constexpr int SWB = 256;  // fa_swb(128)
+                              const int* __restrict__ klens,
+                              int b, int h, int s, float scale,
+                              unsigned int drop_thresh, float keep_scale,
+                              unsigned long long drop_seed) {
+  constexpr int NC = D / 16;     // MFMA K=16 chunks
+  constexpr int NTO = D / 32;    // 32-wide output d-tiles
+  constexpr int SWB = fa_swb(D);
 
   __shared__ char k_raw[V3_KVBLK * SWB];                // swizzled K rows
   __shared__ short vt_lds[D][V3_KVBLK + FA_VPAD];       // V transposed
@@ -1018,15 +1023,17 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
   const bf16_t* Vp = V + bh * s * D;
   bf16_t* Op = O + bh * s * D;
 
+  const int klen = CAUSAL ? s : min(klens ? klens[blockIdx.z] : s, s);
+
   const int q0w = blockIdx.x * V3_QBLK + wave * 32;
   const bool q_active = q0w < s;
   const int q0c = q_active ? q0w : s - 32;
   const int my_q = q0c + ln;      // softmax row this lane owns
 
-  // ---- Q -> B-fragments (8 K=16 chunks), pre-scaled ---------------------
-  bf16x8 q_frag[8];
+  // ---- Q -> B-fragments (K=16 chunks), pre-scaled -----------------------
+  bf16x8 q_frag[NC];
 #pragma unroll
-  for (int c = 0; c < 8; ++c) {
+  for (int c = 0; c < NC; ++c) {
     bf16x8 raw = *reinterpret_cast<const bf16x8*>(
         Qp + (long)my_q * D + c * 16 + hi * 8);
 #pragma unroll
@@ -1039,55 +1046,67 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
   }
 
   float m_run = -INFINITY, l_run = 0.f;   // per-lane: row my_q
-  f32x16 o_acc[4];
+  f32x16 o_acc[NTO];
 #pragma unroll
-  for (int t = 0; t < 4; ++t) {
+  for (int t = 0; t < NTO; ++t) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) o_acc[t][r] = 0.f;
   }
 
-  const int q_hi_row = min(blockIdx.x * V3_QBLK + V3_QBLK, s) - 1;
-  const int n_kv_tiles = (q_hi_row / V3_KVBLK) + 1;
+  int n_kv_tiles;
+  if (CAUSAL) {
+    const int q_hi_row = min(blockIdx.x * V3_QBLK + V3_QBLK, s) - 1;
+    n_kv_tiles = (q_hi_row / V3_KVBLK) + 1;
+  } else {
+    n_kv_tiles = (klen + V3_KVBLK - 1) / V3_KVBLK;
+    if (n_kv_tiles < 1) n_kv_tiles = 1;
+  }
 
-  // async-STAGE staging registers (64*128 elems / 512 threads / 8 = 2)
+  // async-STAGE staging registers
   const int tid = threadIdx.x;
-  bf16x8 k_reg[2], v_reg[2];
+  constexpr int ELEMS = V3_KVBLK * D;
+  constexpr int SWEEPS = (ELEMS + FA_WAVES * 64 * 8 - 1) / (FA_WAVES * 64 * 8);
+  bf16x8 k_reg[SWEEPS], v_reg[SWEEPS];
 #pragma unroll
-  for (int sweep = 0; sweep < 2; ++sweep) {
-    const int i = tid * 8 + sweep * 4096;
-    const int kr = i / D;
-    const int kc = i % D;
-    k_reg[sweep] = *reinterpret_cast<const bf16x8*>(Kp + (long)kr * D + kc);
-    v_reg[sweep] = *reinterpret_cast<const bf16x8*>(Vp + (long)kr * D + kc);
+  for (int sweep = 0; sweep < SWEEPS; ++sweep) {
+    const int i = tid * 8 + sweep * (FA_WAVES * 64 * 8);
+    if (i < ELEMS) {
+      const long kr = min(i / D, s - 1);
+      const int kc = i % D;
+      k_reg[sweep] = *reinterpret_cast<const bf16x8*>(Kp + kr * D + kc);
+      v_reg[sweep] = *reinterpret_cast<const bf16x8*>(Vp + kr * D + kc);
+    }
   }
 
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
     const int k_base = kt * V3_KVBLK;
     __syncthreads();
 #pragma unroll
-    for (int sweep = 0; sweep < 2; ++sweep) {
-      const int i = tid * 8 + sweep * 4096;
-      const int kr = i / D;
-      const int kc = i % D;
-      *reinterpret_cast<bf16x8*>(k_raw + kr * SWB + kswz(kr, kc * 2)) =
-          k_reg[sweep];
-      bf16x8 vv = v_reg[sweep];
+    for (int sweep = 0; sweep < SWEEPS; ++sweep) {
+      const int i = tid * 8 + sweep * (FA_WAVES * 64 * 8);
+      if (i < ELEMS) {
+        const int kr = i / D;
+        const int kc = i % D;
+        *reinterpret_cast<bf16x8*>(k_raw + kr * SWB + kswz(kr, kc * 2)) =
+            k_reg[sweep];
+        bf16x8 vv = v_reg[sweep];
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        vt_lds[kc + j][kr ^ ((kc + j) & 0x38)] = vv[j];
+        for (int j = 0; j < 8; ++j)
+          vt_lds[kc + j][kr ^ ((kc + j) & 0x38)] = vv[j];
+      }
     }
     __syncthreads();
     if (kt + 1 < n_kv_tiles) {
       const int nb = (kt + 1) * V3_KVBLK;
 #pragma unroll
-      for (int sweep = 0; sweep < 2; ++sweep) {
-        const int i = tid * 8 + sweep * 4096;
-        const int kr = i / D;
-        const int kc = i % D;
-        k_reg[sweep] = *reinterpret_cast<const bf16x8*>(
-            Kp + (long)(nb + kr) * D + kc);
-        v_reg[sweep] = *reinterpret_cast<const bf16x8*>(
-            Vp + (long)(nb + kr) * D + kc);
+      for (int sweep = 0; sweep < SWEEPS; ++sweep) {
+        const int i = tid * 8 + sweep * (FA_WAVES * 64 * 8);
+        if (i < ELEMS) {
+          const long kr = min((long)(nb + i / D), (long)s - 1);
+          const int kc = i % D;
+          k_reg[sweep] = *reinterpret_cast<const bf16x8*>(Kp + kr * D + kc);
+          v_reg[sweep] = *reinterpret_cast<const bf16x8*>(Vp + kr * D + kc);
+        }
       }
     }
 
@@ -1102,7 +1121,7 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
     for (int nt = 0; nt < 2; ++nt) {
       const int krow = nt * 32 + ln;
 #pragma unroll
-      for (int c = 0; c < 8; ++c) {
+      for (int c = 0; c < NC; ++c) {
         bf16x8 a = *reinterpret_cast<const bf16x8*>(
             k_raw + krow * SWB + kswz(krow, (c * 16 + hi * 8) * 2));
         st[nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, q_frag[c],
@@ -1121,7 +1140,7 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
       for (int r = 0; r < 16; ++r) {
         const int kcol = k_base + nt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
         float v = st[nt][r];
-        if (kcol > my_q) v = -INFINITY;
+        if (CAUSAL ? (kcol > my_q) : (kcol >= klen)) v = -INFINITY;
         p[nt * 16 + r] = v;
         tmax = fmaxf(tmax, v);
       }
@@ -1141,6 +1160,23 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
     rs += __shfl_xor(rs, 32, 64);
     l_run = l_run * alpha + rs;
 
+    // attention dropout on P (normalizer keeps the undropped sum)
+    if (drop_thresh) {
+      const unsigned long long qi =
+          (bh * (unsigned long long)s + my_q) * (unsigned long long)s;
+#pragma unroll
+      for (int nt = 0; nt < 2; ++nt) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kcol =
+              k_base + nt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float& pv = p[nt * 16 + r];
+          pv = (fa_hash(drop_seed, qi + kcol) < drop_thresh)
+                   ? 0.f : pv * keep_scale;
+        }
+      }
+    }
+
     // broadcast alpha to the lanes holding this row's O accumulators
     if (hi == 0) a_x[wave][ln] = alpha;
     __builtin_amdgcn_s_waitcnt(0);
@@ -1149,7 +1185,7 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
       for (int r = 0; r < 16; ++r) {
         const float al = a_x[wave][(r & 3) + 8 * (r >> 2) + 4 * hi];
 #pragma unroll
-        for (int t = 0; t < 4; ++t) o_acc[t][r] *= al;
+        for (int t = 0; t < NTO; ++t) o_acc[t][r] *= al;
       }
     }
 
@@ -1175,7 +1211,7 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(
           &p_x[wave][ln][c2 * 16 + hi * 8]);
 #pragma unroll
-      for (int t = 0; t < 4; ++t) {   // d cols, 32 each
+      for (int t = 0; t < NTO; ++t) {   // d cols, 32 each
         const int dcol = t * 32 + ln;
         bf16x8 vb = *reinterpret_cast<const bf16x8*>(
             &vt_lds[dcol][(c2 * 16 + hi * 8) ^ (dcol & 0x38)]);
@@ -1195,7 +1231,7 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
     const int qrow = q0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
     const float inv_l = a_x[wave][(r & 3) + 8 * (r >> 2) + 4 * hi];
 #pragma unroll
-    for (int t = 0; t < 4; ++t) {
+    for (int t = 0; t < NTO; ++t) {
       Op[(long)qrow * D + t * 32 + ln] =
           __float2bfloat16(o_acc[t][r] * inv_l);
     }
@@ -1207,13 +1243,23 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
 
 extern "C" void fs_flash_attn_fwd_v3(const void* q, const void* k,
                                      const void* v, void* o, float* lse,
-                                     int b, int h, int s, float scale,
+                                     const int* klens, int b, int h, int s,
+                                     int d, int causal, float scale,
+                                     float drop_p, unsigned long long seed,
                                      hipStream_t stream) {
   dim3 grid((s + V3_QBLK - 1) / V3_QBLK, h, b);
   dim3 block(FA_WAVES * 64);
-  hipLaunchKernelGGL(flash_attn_fwd_v3_kernel, grid, block, 0, stream,
-                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
-                     (bf16_t*)o, lse, b, h, s, scale);
+  const unsigned int dth =
+      (drop_p > 0.f) ? (unsigned int)(drop_p * 4294967296.0) : 0u;
+  const float ksc = (drop_p > 0.f) ? 1.f / (1.f - drop_p) : 1.f;
+#define V3F(DD, CC) hipLaunchKernelGGL((flash_attn_fwd_v3_kernel<DD, CC>), \
+    grid, block, 0, stream, (const bf16_t*)q, (const bf16_t*)k, \
+    (const bf16_t*)v, (bf16_t*)o, lse, klens, b, h, s, scale, dth, ksc, seed)
+  if (d == 128 && causal) V3F(128, true);
+  else if (d == 128) V3F(128, false);
+  else if (d == 64 && causal) V3F(64, true);
+  else V3F(64, false);
+#undef V3F
 }
 
 // ===========================================================================
@@ -1227,6 +1273,7 @@ extern "C" void fs_flash_attn_fwd_v3(const void* q, const void* k,
 #define B3_FB 32  // streamed tile width
 
 // q-major: wave owns 32 q rows (block 256), streams KV in 32-tiles.
+template <int D, bool CAUSAL>
 __global__ __launch_bounds__(FA_WAVES * 64)
 void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
                                  const bf16_t* __restrict__ K,
@@ -1235,9 +1282,13 @@ void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
                                  const float* __restrict__ LSE,
                                  const float* __restrict__ Delta,
                                  bf16_t* __restrict__ dQ,
-                                 int b, int h, int s, float scale) {
-  constexpr int D = 128;
-  constexpr int SWB = 256;
+                                 const int* __restrict__ klens,
+                                 int b, int h, int s, float scale,
+                                 unsigned int drop_thresh, float keep_scale,
+                                 unsigned long long drop_seed) {
+  constexpr int NC = D / 16;
+  constexpr int NTO = D / 32;
+  constexpr int SWB = fa_swb(D);
 
   __shared__ char k_raw[B3_FB * SWB];                  // K rows, swizzled
   __shared__ char v_raw[B3_FB * SWB];                  // V rows, swizzled
@@ -1257,15 +1308,16 @@ void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
   const float* lse = LSE + bh * s;
   const float* dlt = Delta + bh * s;
 
+  const int klen = CAUSAL ? s : min(klens ? klens[blockIdx.z] : s, s);
   const int q0w = blockIdx.x * V3_QBLK + wave * 32;
   const bool q_active = q0w < s;
   const int q0c = q_active ? q0w : s - 32;
   const int my_q = q0c + ln;
 
   // Q (pre-scaled) and dO rows in registers (B-fragments)
-  bf16x8 q_frag[8], do_frag[8];
+  bf16x8 q_frag[NC], do_frag[NC];
 #pragma unroll
-  for (int c = 0; c < 8; ++c) {
+  for (int c = 0; c < NC; ++c) {
     bf16x8 raw = *reinterpret_cast<const bf16x8*>(
         Qp + (long)my_q * D + c * 16 + hi * 8);
 #pragma unroll
@@ -1281,15 +1333,20 @@ void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
   const float lse_r = lse[my_q];
   const float dlt_r = dlt[my_q];
 
-  f32x16 dq_acc[4];
+  f32x16 dq_acc[NTO];
 #pragma unroll
-  for (int t = 0; t < 4; ++t) {
+  for (int t = 0; t < NTO; ++t) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) dq_acc[t][r] = 0.f;
   }
 
-  const int q_hi_row = min(blockIdx.x * V3_QBLK + V3_QBLK, s) - 1;
-  const int n_kv_tiles = (q_hi_row / B3_FB) + 1;
+  int n_kv_tiles;
+  if (CAUSAL) {
+    const int q_hi_row = min(blockIdx.x * V3_QBLK + V3_QBLK, s) - 1;
+    n_kv_tiles = (q_hi_row / B3_FB) + 1;
+  } else {
+    n_kv_tiles = (klen + B3_FB - 1) / B3_FB;
+  }
 
   for (int kt = 0; kt < n_kv_tiles; ++kt) {
     const int k_base = kt * B3_FB;
@@ -1299,11 +1356,10 @@ void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
       for (int i = tid * 8; i < B3_FB * D; i += FA_WAVES * 64 * 8) {
         const int kr = i / D;
         const int kc = i % D;
-        bf16x8 kk = *reinterpret_cast<const bf16x8*>(
-            Kp + (long)(k_base + kr) * D + kc);
+        const long krg = min(k_base + kr, s - 1);
+        bf16x8 kk = *reinterpret_cast<const bf16x8*>(Kp + krg * D + kc);
         *reinterpret_cast<bf16x8*>(k_raw + kr * SWB + kswz(kr, kc * 2)) = kk;
-        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
-            Vp + (long)(k_base + kr) * D + kc);
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(Vp + krg * D + kc);
         *reinterpret_cast<bf16x8*>(v_raw + kr * SWB + kswz(kr, kc * 2)) = vv;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
@@ -1322,7 +1378,7 @@ void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
     {
       const int krow = ln;
 #pragma unroll
-      for (int c = 0; c < 8; ++c) {
+      for (int c = 0; c < NC; ++c) {
         bf16x8 ka = *reinterpret_cast<const bf16x8*>(
             k_raw + krow * SWB + kswz(krow, (c * 16 + hi * 8) * 2));
         st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, q_frag[c], st,
@@ -1343,8 +1399,17 @@ void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
       for (int j = 0; j < 4; ++j) {
         const int r = g * 4 + j;
         const int kcol = k_base + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        float pv = (kcol > my_q) ? 0.f : __expf(st[r] - lse_r);
-        pk[j] = fa_bf16bits(pv * (dpt[r] - dlt_r) * scale);
+        const bool masked = CAUSAL ? (kcol > my_q) : (kcol >= klen);
+        float pv = masked ? 0.f : __expf(st[r] - lse_r);
+        float dpv = dpt[r];
+        if (drop_thresh) {
+          const unsigned long long idx =
+              (bh * (unsigned long long)s + my_q) *
+                  (unsigned long long)s + kcol;
+          dpv = (fa_hash(drop_seed, idx) < drop_thresh)
+                    ? 0.f : dpv * keep_scale;
+        }
+        pk[j] = fa_bf16bits(pv * (dpv - dlt_r) * scale);
       }
       *reinterpret_cast<long*>(&p_x[wave][ln][8 * g + 4 * hi]) =
           *reinterpret_cast<const long*>(pk);
@@ -1357,7 +1422,7 @@ void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(
           &p_x[wave][ln][c2 * 16 + hi * 8]);
 #pragma unroll
-      for (int t = 0; t < 4; ++t) {
+      for (int t = 0; t < NTO; ++t) {
         const int dcol = t * 32 + ln;
         bf16x8 kb = *reinterpret_cast<const bf16x8*>(
             &kt_lds[dcol][(c2 * 16 + hi * 8) ^ (dcol & 0x18)]);
@@ -1373,7 +1438,7 @@ void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
   for (int r = 0; r < 16; ++r) {
     const int qrow = q0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
 #pragma unroll
-    for (int t = 0; t < 4; ++t)
+    for (int t = 0; t < NTO; ++t)
       dQp[(long)qrow * D + t * 32 + ln] = __float2bfloat16(dq_acc[t][r]);
   }
 }
@@ -1383,6 +1448,7 @@ void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
 // into a register block for the St MFMAs, then V overwrites the same
 // block for the dPt MFMAs (rows are L2-hot; the loads hide under MFMA).
 // The softmax scale folds into the exp (st*scale - lse) so K stays raw.
+template <int D, bool CAUSAL>
 __global__ __launch_bounds__(FA_WAVES * 64)
 void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
                                   const bf16_t* __restrict__ K,
@@ -1392,9 +1458,13 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
                                   const float* __restrict__ Delta,
                                   bf16_t* __restrict__ dK,
                                   bf16_t* __restrict__ dV,
-                                  int b, int h, int s, float scale) {
-  constexpr int D = 128;
-  constexpr int SWB = 256;
+                                  const int* __restrict__ klens,
+                                  int b, int h, int s, float scale,
+                                  unsigned int drop_thresh, float keep_scale,
+                                  unsigned long long drop_seed) {
+  constexpr int NC = D / 16;
+  constexpr int NTO = D / 32;
+  constexpr int SWB = fa_swb(D);
 
   __shared__ char q_raw[B3_FB * SWB];                   // Q rows, swizzled
   __shared__ char do_raw[B3_FB * SWB];                  // dO rows, swizzled
@@ -1416,16 +1486,17 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
   const float* lse = LSE + bh * s;
   const float* dlt = Delta + bh * s;
 
+  const int klen = CAUSAL ? s : min(klens ? klens[blockIdx.z] : s, s);
   const int kv0_blk = blockIdx.x * V3_QBLK;
   const int kv0w = kv0_blk + wave * 32;
   const bool kv_active = kv0w < s;
   const int kv0c = kv_active ? kv0w : s - 32;
   const int my_kv = kv0c + ln;
 
-  // dv/dk in the 32x32 layout: 4 d-tiles x 16 f32 each
-  f32x16 dv_acc[4], dk_acc[4];
+  // dv/dk in the 32x32 layout: NTO d-tiles x 16 f32 each
+  f32x16 dv_acc[NTO], dk_acc[NTO];
 #pragma unroll
-  for (int t = 0; t < 4; ++t) {
+  for (int t = 0; t < NTO; ++t) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       dv_acc[t][r] = 0.f;
@@ -1433,7 +1504,7 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
     }
   }
 
-  const int first_qt = kv0_blk / B3_FB;
+  const int first_qt = CAUSAL ? kv0_blk / B3_FB : 0;
   const int n_q_tiles = s / B3_FB;
 
   for (int qt = first_qt; qt < n_q_tiles; ++qt) {
@@ -1465,7 +1536,7 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
     for (int r = 0; r < 16; ++r) st[r] = 0.f;
 #pragma unroll
-    for (int c = 0; c < 8; ++c) {
+    for (int c = 0; c < NC; ++c) {
       bf16x8 ka = *reinterpret_cast<const bf16x8*>(
           Kp + (long)my_kv * D + c * 16 + hi * 8);
       bf16x8 qb = *reinterpret_cast<const bf16x8*>(
@@ -1477,7 +1548,7 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
     for (int r = 0; r < 16; ++r) dpt[r] = 0.f;
 #pragma unroll
-    for (int c = 0; c < 8; ++c) {
+    for (int c = 0; c < NC; ++c) {
       bf16x8 va = *reinterpret_cast<const bf16x8*>(
           Vp + (long)my_kv * D + c * 16 + hi * 8);
       bf16x8 db = *reinterpret_cast<const bf16x8*>(
@@ -1493,10 +1564,19 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kvrow_loc = (r & 3) + 8 * (r >> 2) + 4 * hi;
-      float pv = (qcol < kv0c + kvrow_loc)
-                     ? 0.f : __expf(st[r] * scale - lse_c);
-      st[r] = pv;  // reuse st as Pt storage
-      p_x[wave][kvrow_loc][ln] = fa_bf16bits(pv);
+      const int kvrow = kv0c + kvrow_loc;
+      const bool masked = CAUSAL ? (qcol < kvrow) : (kvrow >= klen);
+      float pv = masked ? 0.f : __expf(st[r] * scale - lse_c);
+      float dm = 1.f;
+      if (drop_thresh) {
+        const unsigned long long idx =
+            (bh * (unsigned long long)s + qcol) *
+                (unsigned long long)s + kvrow;
+        dm = (fa_hash(drop_seed, idx) < drop_thresh) ? 0.f : keep_scale;
+      }
+      st[r] = pv;           // undropped P for the dSt product
+      dpt[r] *= dm;         // dropped dP
+      p_x[wave][kvrow_loc][ln] = fa_bf16bits(pv * dm);  // dropped P -> dV
     }
     __builtin_amdgcn_s_waitcnt(0);
 
@@ -1506,7 +1586,7 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(
           &p_x[wave][ln][c2 * 16 + hi * 8]);
 #pragma unroll
-      for (int t = 0; t < 4; ++t) {
+      for (int t = 0; t < NTO; ++t) {
         const int dcol = t * 32 + ln;
         bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
             &dot_lds[dcol][(c2 * 16 + hi * 8) ^ (dcol & 0x18)]);
@@ -1529,7 +1609,7 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(
           &p_x[wave][ln][c2 * 16 + hi * 8]);
 #pragma unroll
-      for (int t = 0; t < 4; ++t) {
+      for (int t = 0; t < NTO; ++t) {
         const int dcol = t * 32 + ln;
         bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
             &qt_lds[dcol][(c2 * 16 + hi * 8) ^ (dcol & 0x18)]);
@@ -1544,35 +1624,63 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
   for (int r = 0; r < 16; ++r) {
     const int kvrow = kv0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
 #pragma unroll
-    for (int t = 0; t < 4; ++t) {
+    for (int t = 0; t < NTO; ++t) {
       dKp[(long)kvrow * D + t * 32 + ln] = __float2bfloat16(dk_acc[t][r]);
       dVp[(long)kvrow * D + t * 32 + ln] = __float2bfloat16(dv_acc[t][r]);
     }
   }
 }
 
-extern "C" void fs_flash_attn_bwd_v3(const void* q, const void* k,
-                                     const void* v, const void* o,
-                                     const void* dout, const float* lse,
-                                     void* dq, void* dk, void* dv,
-                                     float* delta_ws, int b, int h, int s,
-                                     float scale, hipStream_t stream) {
+template <int D, bool CAUSAL>
+static void launch_bwd_v3(const void* q, const void* k, const void* v,
+                          const void* o, const void* dout, const float* lse,
+                          void* dq, void* dk, void* dv, float* delta_ws,
+                          const int* klens, int b, int h, int s, float scale,
+                          unsigned int dth, float ksc,
+                          unsigned long long seed, hipStream_t stream) {
   const long rows = (long)b * h * s;
   {
     long blocks = (rows + 3) / 4;
     if (blocks > 4096) blocks = 4096;
-    hipLaunchKernelGGL((flash_delta_kernel<128>), dim3((unsigned)blocks),
+    hipLaunchKernelGGL((flash_delta_kernel<D>), dim3((unsigned)blocks),
                        dim3(256), 0, stream, (const bf16_t*)dout,
                        (const bf16_t*)o, delta_ws, rows);
   }
   dim3 grid((s + V3_QBLK - 1) / V3_QBLK, h, b);
   dim3 block(FA_WAVES * 64);
-  hipLaunchKernelGGL(flash_attn_bwd_dq_v3_kernel, grid, block, 0, stream,
-                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
-                     (const bf16_t*)dout, lse, delta_ws, (bf16_t*)dq,
-                     b, h, s, scale);
-  hipLaunchKernelGGL(flash_attn_bwd_dkv_v3_kernel, grid, block, 0, stream,
-                     (const bf16_t*)q, (const bf16_t*)k, (const bf16_t*)v,
-                     (const bf16_t*)dout, lse, delta_ws, (bf16_t*)dk,
-                     (bf16_t*)dv, b, h, s, scale);
+  hipLaunchKernelGGL((flash_attn_bwd_dq_v3_kernel<D, CAUSAL>), grid, block,
+                     0, stream, (const bf16_t*)q, (const bf16_t*)k,
+                     (const bf16_t*)v, (const bf16_t*)dout, lse, delta_ws,
+                     (bf16_t*)dq, klens, b, h, s, scale, dth, ksc, seed);
+  hipLaunchKernelGGL((flash_attn_bwd_dkv_v3_kernel<D, CAUSAL>), grid, block,
+                     0, stream, (const bf16_t*)q, (const bf16_t*)k,
+                     (const bf16_t*)v, (const bf16_t*)dout, lse, delta_ws,
+                     (bf16_t*)dk, (bf16_t*)dv, klens, b, h, s, scale, dth,
+                     ksc, seed);
+}
+
+extern "C" void fs_flash_attn_bwd_v3(const void* q, const void* k,
+                                     const void* v, const void* o,
+                                     const void* dout, const float* lse,
+                                     void* dq, void* dk, void* dv,
+                                     float* delta_ws, const int* klens,
+                                     int b, int h, int s, int d, int causal,
+                                     float scale, float drop_p,
+                                     unsigned long long seed,
+                                     hipStream_t stream) {
+  const unsigned int dth =
+      (drop_p > 0.f) ? (unsigned int)(drop_p * 4294967296.0) : 0u;
+  const float ksc = (drop_p > 0.f) ? 1.f / (1.f - drop_p) : 1.f;
+  if (d == 128 && causal)
+    launch_bwd_v3<128, true>(q, k, v, o, dout, lse, dq, dk, dv, delta_ws,
+                             klens, b, h, s, scale, dth, ksc, seed, stream);
+  else if (d == 128)
+    launch_bwd_v3<128, false>(q, k, v, o, dout, lse, dq, dk, dv, delta_ws,
+                              klens, b, h, s, scale, dth, ksc, seed, stream);
+  else if (d == 64 && causal)
+    launch_bwd_v3<64, true>(q, k, v, o, dout, lse, dq, dk, dv, delta_ws,
+                            klens, b, h, s, scale, dth, ksc, seed, stream);
+  else
+    launch_bwd_v3<64, false>(q, k, v, o, dout, lse, dq, dk, dv, delta_ws,
+                             klens, b, h, s, scale, dth, ksc, seed, stream);
 }

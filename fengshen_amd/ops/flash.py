@@ -20,15 +20,23 @@ from fengshen_amd.ops import get_ext
 _FLASH_DIMS = (40, 64, 80, 96, 128, 160)
 
 
+def _v3_eligible(ext, q, k, causal, klens):
+    """v3 (swapped-QK^T 32x32 schedule, ~1.8x): self-attention at
+    d in {64, 128} with s % 64 == 0; causal or klens-masked bidirectional;
+    dropout fused in both directions."""
+    return (hasattr(ext, "flash_attn_fwd_v3")
+            and q.shape[-1] in (64, 128)
+            and q.shape[-2] == k.shape[-2]
+            and q.shape[-2] % 64 == 0 and q.shape[-2] >= 64)
+
+
 class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale, causal, klens, dropout_p, seed):
         ext = get_ext()
-        if (causal and klens is None and dropout_p == 0.0
-                and q.shape[-1] == 128 and hasattr(ext, "flash_attn_fwd_v3")):
-            # v3: swapped-QK^T 32x32 schedule, 1.8x the general kernel at
-            # the LLaMA shape; same LSE contract, shared backward
-            o, lse = ext.flash_attn_fwd_v3(q, k, v, scale)
+        if _v3_eligible(ext, q, k, causal, klens):
+            o, lse = ext.flash_attn_fwd_v3(q, k, v, scale, causal, klens,
+                                           dropout_p, seed)
         else:
             o, lse = ext.flash_attn_fwd(q, k, v, scale, causal, klens,
                                         dropout_p, seed)
@@ -44,11 +52,10 @@ class _FlashAttention(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         ext = get_ext()
-        if (ctx.causal and ctx.klens is None and ctx.dropout_p == 0.0
-                and q.shape[-1] == 128
-                and hasattr(ext, "flash_attn_bwd_v3")):
+        if _v3_eligible(ext, q, k, ctx.causal, ctx.klens):
             dq, dk, dv = ext.flash_attn_bwd_v3(
-                q, k, v, o, do.contiguous(), lse, ctx.scale)
+                q, k, v, o, do.contiguous(), lse, ctx.scale, ctx.causal,
+                ctx.klens, ctx.dropout_p, ctx.seed)
         else:
             dq, dk, dv = ext.flash_attn_bwd(
                 q, k, v, o, do.contiguous(), lse, ctx.scale, ctx.causal,
