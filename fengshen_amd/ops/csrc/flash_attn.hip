@@ -993,7 +993,7 @@ extern "C" void fs_flash_attn_bwd(const void* q, const void* k, const void* v,
 
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool HAS_DROP>
 __global__ __launch_bounds__(FA_WAVES * 64)
 void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
                               const bf16_t* __restrict__ K,
@@ -1161,7 +1161,7 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
     l_run = l_run * alpha + rs;
 
     // attention dropout on P (normalizer keeps the undropped sum)
-    if (drop_thresh) {
+    if (HAS_DROP && drop_thresh) {
       const unsigned long long qi =
           (bh * (unsigned long long)s + my_q) * (unsigned long long)s;
 #pragma unroll
@@ -1252,13 +1252,21 @@ extern "C" void fs_flash_attn_fwd_v3(const void* q, const void* k,
   const unsigned int dth =
       (drop_p > 0.f) ? (unsigned int)(drop_p * 4294967296.0) : 0u;
   const float ksc = (drop_p > 0.f) ? 1.f / (1.f - drop_p) : 1.f;
-#define V3F(DD, CC) hipLaunchKernelGGL((flash_attn_fwd_v3_kernel<DD, CC>), \
+#define V3F(DD, CC, DR) hipLaunchKernelGGL( \
+    (flash_attn_fwd_v3_kernel<DD, CC, DR>), \
     grid, block, 0, stream, (const bf16_t*)q, (const bf16_t*)k, \
     (const bf16_t*)v, (bf16_t*)o, lse, klens, b, h, s, scale, dth, ksc, seed)
-  if (d == 128 && causal) V3F(128, true);
-  else if (d == 128) V3F(128, false);
-  else if (d == 64 && causal) V3F(64, true);
-  else V3F(64, false);
+  if (dth) {
+    if (d == 128 && causal) V3F(128, true, true);
+    else if (d == 128) V3F(128, false, true);
+    else if (d == 64 && causal) V3F(64, true, true);
+    else V3F(64, false, true);
+  } else {
+    if (d == 128 && causal) V3F(128, true, false);
+    else if (d == 128) V3F(128, false, false);
+    else if (d == 64 && causal) V3F(64, true, false);
+    else V3F(64, false, false);
+  }
 #undef V3F
 }
 
@@ -1273,7 +1281,7 @@ extern "C" void fs_flash_attn_fwd_v3(const void* q, const void* k,
 #define B3_FB 32  // streamed tile width
 
 // q-major: wave owns 32 q rows (block 256), streams KV in 32-tiles.
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool HAS_DROP>
 __global__ __launch_bounds__(FA_WAVES * 64)
 void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
                                  const bf16_t* __restrict__ K,
@@ -1402,7 +1410,7 @@ void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
         const bool masked = CAUSAL ? (kcol > my_q) : (kcol >= klen);
         float pv = masked ? 0.f : __expf(st[r] - lse_r);
         float dpv = dpt[r];
-        if (drop_thresh) {
+        if (HAS_DROP && drop_thresh) {
           const unsigned long long idx =
               (bh * (unsigned long long)s + my_q) *
                   (unsigned long long)s + kcol;
@@ -1448,7 +1456,7 @@ void flash_attn_bwd_dq_v3_kernel(const bf16_t* __restrict__ Q,
 // into a register block for the St MFMAs, then V overwrites the same
 // block for the dPt MFMAs (rows are L2-hot; the loads hide under MFMA).
 // The softmax scale folds into the exp (st*scale - lse) so K stays raw.
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool HAS_DROP>
 __global__ __launch_bounds__(FA_WAVES * 64)
 void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
                                   const bf16_t* __restrict__ K,
@@ -1568,7 +1576,7 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
       const bool masked = CAUSAL ? (qcol < kvrow) : (kvrow >= klen);
       float pv = masked ? 0.f : __expf(st[r] * scale - lse_c);
       float dm = 1.f;
-      if (drop_thresh) {
+      if (HAS_DROP && drop_thresh) {
         const unsigned long long idx =
             (bh * (unsigned long long)s + qcol) *
                 (unsigned long long)s + kvrow;
@@ -1631,7 +1639,7 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
   }
 }
 
-template <int D, bool CAUSAL>
+template <int D, bool CAUSAL, bool HAS_DROP>
 static void launch_bwd_v3(const void* q, const void* k, const void* v,
                           const void* o, const void* dout, const float* lse,
                           void* dq, void* dk, void* dv, float* delta_ws,
@@ -1648,15 +1656,16 @@ static void launch_bwd_v3(const void* q, const void* k, const void* v,
   }
   dim3 grid((s + V3_QBLK - 1) / V3_QBLK, h, b);
   dim3 block(FA_WAVES * 64);
-  hipLaunchKernelGGL((flash_attn_bwd_dq_v3_kernel<D, CAUSAL>), grid, block,
-                     0, stream, (const bf16_t*)q, (const bf16_t*)k,
-                     (const bf16_t*)v, (const bf16_t*)dout, lse, delta_ws,
-                     (bf16_t*)dq, klens, b, h, s, scale, dth, ksc, seed);
-  hipLaunchKernelGGL((flash_attn_bwd_dkv_v3_kernel<D, CAUSAL>), grid, block,
-                     0, stream, (const bf16_t*)q, (const bf16_t*)k,
-                     (const bf16_t*)v, (const bf16_t*)dout, lse, delta_ws,
-                     (bf16_t*)dk, (bf16_t*)dv, klens, b, h, s, scale, dth,
+  hipLaunchKernelGGL((flash_attn_bwd_dq_v3_kernel<D, CAUSAL, HAS_DROP>),
+                     grid, block, 0, stream, (const bf16_t*)q,
+                     (const bf16_t*)k, (const bf16_t*)v, (const bf16_t*)dout,
+                     lse, delta_ws, (bf16_t*)dq, klens, b, h, s, scale, dth,
                      ksc, seed);
+  hipLaunchKernelGGL((flash_attn_bwd_dkv_v3_kernel<D, CAUSAL, HAS_DROP>),
+                     grid, block, 0, stream, (const bf16_t*)q,
+                     (const bf16_t*)k, (const bf16_t*)v, (const bf16_t*)dout,
+                     lse, delta_ws, (bf16_t*)dk, (bf16_t*)dv, klens, b, h, s,
+                     scale, dth, ksc, seed);
 }
 
 extern "C" void fs_flash_attn_bwd_v3(const void* q, const void* k,
@@ -1671,16 +1680,19 @@ extern "C" void fs_flash_attn_bwd_v3(const void* q, const void* k,
   const unsigned int dth =
       (drop_p > 0.f) ? (unsigned int)(drop_p * 4294967296.0) : 0u;
   const float ksc = (drop_p > 0.f) ? 1.f / (1.f - drop_p) : 1.f;
-  if (d == 128 && causal)
-    launch_bwd_v3<128, true>(q, k, v, o, dout, lse, dq, dk, dv, delta_ws,
-                             klens, b, h, s, scale, dth, ksc, seed, stream);
-  else if (d == 128)
-    launch_bwd_v3<128, false>(q, k, v, o, dout, lse, dq, dk, dv, delta_ws,
-                              klens, b, h, s, scale, dth, ksc, seed, stream);
-  else if (d == 64 && causal)
-    launch_bwd_v3<64, true>(q, k, v, o, dout, lse, dq, dk, dv, delta_ws,
-                            klens, b, h, s, scale, dth, ksc, seed, stream);
-  else
-    launch_bwd_v3<64, false>(q, k, v, o, dout, lse, dq, dk, dv, delta_ws,
-                             klens, b, h, s, scale, dth, ksc, seed, stream);
+#define BWD3(DD, CC, DR) launch_bwd_v3<DD, CC, DR>( \
+      q, k, v, o, dout, lse, dq, dk, dv, delta_ws, klens, b, h, s, scale, \
+      dth, ksc, seed, stream)
+  if (dth) {
+    if (d == 128 && causal) BWD3(128, true, true);
+    else if (d == 128) BWD3(128, false, true);
+    else if (d == 64 && causal) BWD3(64, true, true);
+    else BWD3(64, false, true);
+  } else {
+    if (d == 128 && causal) BWD3(128, true, false);
+    else if (d == 128) BWD3(128, false, false);
+    else if (d == 64 && causal) BWD3(64, true, false);
+    else BWD3(64, false, false);
+  }
+#undef BWD3
 }

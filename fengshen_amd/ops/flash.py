@@ -20,21 +20,27 @@ from fengshen_amd.ops import get_ext
 _FLASH_DIMS = (40, 64, 80, 96, 128, 160)
 
 
-def _v3_eligible(ext, q, k, causal, klens):
+def _v3_eligible(ext, q, k, causal, klens, dropout_p):
     """v3 (swapped-QK^T 32x32 schedule, ~1.8x): self-attention at
-    d in {64, 128} with s % 64 == 0; causal or klens-masked bidirectional;
-    dropout fused in both directions."""
-    return (hasattr(ext, "flash_attn_fwd_v3")
+    d in {64, 128} with s % 64 == 0; causal or klens-masked bidirectional.
+    Dropout is fused at d=64 (BERT); at d=128 the hash registers push the
+    kernel past 256 VGPRs into scratch, so dropout stays on the general
+    kernels there."""
+    if not (hasattr(ext, "flash_attn_fwd_v3")
             and q.shape[-1] in (64, 128)
             and q.shape[-2] == k.shape[-2]
-            and q.shape[-2] % 64 == 0 and q.shape[-2] >= 64)
+            and q.shape[-2] % 64 == 0 and q.shape[-2] >= 64):
+        return False
+    if dropout_p > 0 and q.shape[-1] != 64:
+        return False
+    return True
 
 
 class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale, causal, klens, dropout_p, seed):
         ext = get_ext()
-        if _v3_eligible(ext, q, k, causal, klens):
+        if _v3_eligible(ext, q, k, causal, klens, dropout_p):
             o, lse = ext.flash_attn_fwd_v3(q, k, v, scale, causal, klens,
                                            dropout_p, seed)
         else:
@@ -52,7 +58,7 @@ class _FlashAttention(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         ext = get_ext()
-        if _v3_eligible(ext, q, k, ctx.causal, ctx.klens):
+        if _v3_eligible(ext, q, k, ctx.causal, ctx.klens, ctx.dropout_p):
             dq, dk, dv = ext.flash_attn_bwd_v3(
                 q, k, v, o, do.contiguous(), lse, ctx.scale, ctx.causal,
                 ctx.klens, ctx.dropout_p, ctx.seed)

@@ -449,7 +449,8 @@ def test_flash_fwd_v3_matches_oracle(s):
     k = _rand(b, h, s, d, seed=1)
     v = _rand(b, h, s, d, seed=2)
     scale = 1.0 / math.sqrt(d)
-    o, lse = get_ext().flash_attn_fwd_v3(q, k, v, scale)
+    o, lse = get_ext().flash_attn_fwd_v3(q, k, v, scale, True, None,
+                                         0.0, 0)
     _close(o, _flash_oracle(q, k, v, scale, causal=True))
     # LSE must match the general kernel's (used by the shared backward)
     o2, lse2 = get_ext().flash_attn_fwd(q, k, v, scale, True, None, 0.0, 0)
