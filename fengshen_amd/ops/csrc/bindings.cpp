@@ -360,7 +360,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
                                   c10::optional<at::Tensor> klens,
                                   double drop_p, int64_t seed) {
     const int b = q.size(0), h = q.size(1), s = q.size(2), d = q.size(3);
-    TORCH_CHECK((d == 128 || d == 64) && s % 64 == 0);
+    TORCH_CHECK((d == 128 || d == 96 || d == 64) && s % 64 == 0);
     const int* klp = nullptr;
     if (klens.has_value()) klp = klens->data_ptr<int>();
     auto dq = at::empty_like(q);
@@ -382,7 +382,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
     TORCH_CHECK(q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
     TORCH_CHECK(q.scalar_type() == at::kBFloat16);
     const int b = q.size(0), h = q.size(1), s = q.size(2), d = q.size(3);
-    TORCH_CHECK((d == 128 || d == 64) && s % 64 == 0 && s >= 64);
+    TORCH_CHECK((d == 128 || d == 96 || d == 64) && s % 64 == 0 && s >= 64);
     TORCH_CHECK(k.size(2) == s, "v3 is self-attention (sq == sk)");
     const int* klp = nullptr;
     if (klens.has_value()) {

@@ -1259,11 +1259,15 @@ extern "C" void fs_flash_attn_fwd_v3(const void* q, const void* k,
   if (dth) {
     if (d == 128 && causal) V3F(128, true, true);
     else if (d == 128) V3F(128, false, true);
+    else if (d == 96 && causal) V3F(96, true, true);
+    else if (d == 96) V3F(96, false, true);
     else if (d == 64 && causal) V3F(64, true, true);
     else V3F(64, false, true);
   } else {
     if (d == 128 && causal) V3F(128, true, false);
     else if (d == 128) V3F(128, false, false);
+    else if (d == 96 && causal) V3F(96, true, false);
+    else if (d == 96) V3F(96, false, false);
     else if (d == 64 && causal) V3F(64, true, false);
     else V3F(64, false, false);
   }
@@ -1568,7 +1572,8 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
     const float lse_c = lse[qcol];
     const float dlt_c = dlt[qcol];
 
-    // Pt by kv row into the exchange buffer (2B scatter stores)
+    // Pt AND dSt by kv row, one exchange pass: st/dpt registers die
+    // here (keeping them through the dV MFMAs spilled 128 B/lane)
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kvrow_loc = (r & 3) + 8 * (r >> 2) + 4 * hi;
@@ -1582,9 +1587,9 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
                 (unsigned long long)s + kvrow;
         dm = (fa_hash(drop_seed, idx) < drop_thresh) ? 0.f : keep_scale;
       }
-      st[r] = pv;           // undropped P for the dSt product
-      dpt[r] *= dm;         // dropped dP
       p_x[wave][kvrow_loc][ln] = fa_bf16bits(pv * dm);  // dropped P -> dV
+      ds_x[wave][kvrow_loc][ln] =
+          fa_bf16bits(pv * (dm * dpt[r] - dlt_c) * scale);  // dSt -> dK
     }
     __builtin_amdgcn_s_waitcnt(0);
 
@@ -1603,19 +1608,11 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
       }
     }
 
-    // dSt = Pt * (dPt - delta) * scale; exchange and accumulate dK
-    __builtin_amdgcn_s_waitcnt(0);
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int kvrow_loc = (r & 3) + 8 * (r >> 2) + 4 * hi;
-      p_x[wave][kvrow_loc][ln] =
-          fa_bf16bits(st[r] * (dpt[r] - dlt_c) * scale);
-    }
-    __builtin_amdgcn_s_waitcnt(0);
+    // dK += dSt Q (dSt already exchanged above)
 #pragma unroll
     for (int c2 = 0; c2 < 2; ++c2) {
       bf16x8 pa = *reinterpret_cast<const bf16x8*>(
-          &p_x[wave][ln][c2 * 16 + hi * 8]);
+          &ds_x[wave][ln][c2 * 16 + hi * 8]);
 #pragma unroll
       for (int t = 0; t < NTO; ++t) {
         const int dcol = t * 32 + ln;
@@ -1686,11 +1683,15 @@ extern "C" void fs_flash_attn_bwd_v3(const void* q, const void* k,
   if (dth) {
     if (d == 128 && causal) BWD3(128, true, true);
     else if (d == 128) BWD3(128, false, true);
+    else if (d == 96 && causal) BWD3(96, true, true);
+    else if (d == 96) BWD3(96, false, true);
     else if (d == 64 && causal) BWD3(64, true, true);
     else BWD3(64, false, true);
   } else {
     if (d == 128 && causal) BWD3(128, true, false);
     else if (d == 128) BWD3(128, false, false);
+    else if (d == 96 && causal) BWD3(96, true, false);
+    else if (d == 96) BWD3(96, false, false);
     else if (d == 64 && causal) BWD3(64, true, false);
     else BWD3(64, false, false);
   }

@@ -47,14 +47,14 @@ print(f"flash fwd+bwd: {dt*1000:.2f} ms, {fb_flops/dt/1e12:.1f} TF-equiv")
 # v3 fwd timing (same shape)
 from fengshen_amd.ops import get_ext
 qd, kd, vd = q.detach(), k.detach(), v.detach()
-o3, lse3 = get_ext().flash_attn_fwd_v3(qd, kd, vd, scale)
+o3, lse3 = get_ext().flash_attn_fwd_v3(qd, kd, vd, scale, True, None, 0.0, 0)
 o1 = flash_attention(qd, kd, vd, scale)
 err3 = (o3.float() - o1.float()).abs().max() / o1.float().abs().max()
 print("v3 vs v2 rel err:", err3.item())
 torch.cuda.synchronize()
 t0 = time.perf_counter()
 for _ in range(N):
-    get_ext().flash_attn_fwd_v3(qd, kd, vd, scale)
+    get_ext().flash_attn_fwd_v3(qd, kd, vd, scale, True, None, 0.0, 0)
 torch.cuda.synchronize()
 dt3 = (time.perf_counter() - t0) / N
 print(f"flash fwd v3: {dt3*1000:.2f} ms, {flops/dt3/1e12:.1f} TF")
