@@ -1482,7 +1482,8 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
   __shared__ char do_raw[B3_FB * SWB];                  // dO rows, swizzled
   __shared__ short qt_lds[D][B3_FB + FA_VPAD];          // Q^T
   __shared__ short dot_lds[D][B3_FB + FA_VPAD];         // dO^T
-  __shared__ short p_x[FA_WAVES][32][B3_FB + FA_VPAD];  // by kv row
+  __shared__ short p_x[FA_WAVES][32][B3_FB + FA_VPAD];   // Pt by kv row
+  __shared__ short ds_x[FA_WAVES][32][B3_FB + FA_VPAD];  // dSt by kv row
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
