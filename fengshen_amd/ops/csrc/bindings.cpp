@@ -56,6 +56,10 @@ void fs_flash_attn_bwd(const void*, const void*, const void*, const void*,
                        float, unsigned long long, hipStream_t);
 void fs_w8_gemv(const void*, const float*, const void*, void*, int, int, int,
                 hipStream_t);
+void fs_vocab_ce_fwd(const void*, const long*, float*, float*, float*,
+                     long, int, int, int, hipStream_t);
+void fs_vocab_ce_bwd(const void*, const long*, const float*, const float*,
+                     const float*, void*, long, int, int, int, hipStream_t);
 void fs_flash_attn_fwd_v3(const void*, const void*, const void*, void*,
                           float*, const int*, int, int, int, int, int, float,
                           float, unsigned long long, hipStream_t);
@@ -354,6 +358,37 @@ static at::Tensor w8_gemv(at::Tensor q8, at::Tensor scale, at::Tensor x) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("w8_gemv", &w8_gemv);
   mod.def("flash_attn_fwd", &flash_attn_fwd);
+  mod.def("vocab_ce_fwd", [](at::Tensor logits2d, at::Tensor targets,
+                             int64_t vstart, int64_t vend) {
+    TORCH_CHECK(logits2d.is_contiguous() && logits2d.dim() == 2
+                && logits2d.scalar_type() == at::kBFloat16);
+    TORCH_CHECK(targets.scalar_type() == at::kLong
+                && targets.is_contiguous());
+    const long n = logits2d.size(0);
+    const int w = logits2d.size(1);
+    TORCH_CHECK(w % 8 == 0, "vocab shard must be divisible by 8");
+    auto opts = logits2d.options().dtype(at::kFloat);
+    auto m = at::empty({n}, opts);
+    auto z = at::empty({n}, opts);
+    auto pred = at::empty({n}, opts);
+    fs_vocab_ce_fwd(logits2d.data_ptr(), targets.data_ptr<long>(),
+                    m.data_ptr<float>(), z.data_ptr<float>(),
+                    pred.data_ptr<float>(), n, w, (int)vstart, (int)vend,
+                    cur_stream());
+    return std::vector<at::Tensor>{m, z, pred};
+  });
+  mod.def("vocab_ce_bwd", [](at::Tensor logits2d, at::Tensor targets,
+                             at::Tensor m, at::Tensor z, at::Tensor gout,
+                             int64_t vstart, int64_t vend) {
+    const long n = logits2d.size(0);
+    const int w = logits2d.size(1);
+    auto dl = at::empty_like(logits2d);
+    fs_vocab_ce_bwd(logits2d.data_ptr(), targets.data_ptr<long>(),
+                    m.data_ptr<float>(), z.data_ptr<float>(),
+                    gout.contiguous().data_ptr<float>(), dl.data_ptr(), n, w,
+                    (int)vstart, (int)vend, cur_stream());
+    return dl;
+  });
   mod.def("flash_attn_bwd_v3", [](at::Tensor q, at::Tensor k, at::Tensor v,
                                   at::Tensor o, at::Tensor dout,
                                   at::Tensor lse, double scale, bool causal,

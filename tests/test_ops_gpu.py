@@ -478,3 +478,39 @@ def test_flash_bwd_v3_matches_oracle(s):
     _close(q.grad, q2.grad, 3e-2)
     _close(k.grad, k2.grad, 3e-2)
     _close(v.grad, v2.grad, 3e-2)
+
+
+def test_fused_vocab_ce_matches_composite():
+    """Fused CE (per-row stats, bwd recompute) vs the composite
+    vocab-parallel CE and torch.nn CE, incl. ignored (-100) rows."""
+    from fengshen_amd.parallel.cross_entropy import (
+        _FusedVocabParallelCrossEntropy, _VocabParallelCrossEntropy,
+        _hip_ext)
+    torch.manual_seed(0)
+    b, s, v = 2, 64, 1024
+    logits = _rand(b, s, v, seed=4).requires_grad_(True)
+    target = torch.randint(0, v, (b, s), device="cuda")
+    target[0, :5] = -100
+    ext = _hip_ext(logits)
+    assert ext is not None
+    loss_f = _FusedVocabParallelCrossEntropy.apply(logits, target, ext)
+    valid = (target != -100)
+    total_f = (loss_f * valid).sum() / valid.sum()
+    total_f.backward()
+    g_fused = logits.grad.clone()
+
+    logits2 = logits.detach().clone().requires_grad_(True)
+    loss_c = _VocabParallelCrossEntropy.apply(logits2,
+                                              target.clamp(min=0))
+    total_c = (loss_c * valid).sum() / valid.sum()
+    total_c.backward()
+    # clamp(min=0) gives ignored rows a fake target: compare valid only
+    assert torch.allclose((loss_f * valid), (loss_c * valid), atol=2e-2,
+                          rtol=1e-2)
+    _close(g_fused[valid], logits2.grad[valid], 3e-2)
+
+    # absolute check vs torch CE
+    ref = torch.nn.functional.cross_entropy(
+        logits.detach().float().reshape(-1, v), target.reshape(-1),
+        ignore_index=-100, reduction="none").view(b, s)
+    assert torch.allclose(loss_f * valid, ref * valid, atol=2e-2, rtol=1e-2)
