@@ -68,3 +68,20 @@ for _ in range(N2):
 torch.cuda.synchronize()
 dt4 = (time.perf_counter() - t0) / N2
 print(f"flash fwd+bwd (v3 routed): {dt4*1000:.2f} ms, {flops*3.5/dt4/1e12:.1f} TF-equiv")
+
+# long-context scaling (the reference fused softmax capped at sk<=2048)
+for s_long in (4096, 8192):
+    bl = 2
+    ql = torch.randn(bl, 40, s_long, 128, device="cuda").to(torch.bfloat16)
+    kl = torch.randn(bl, 40, s_long, 128, device="cuda").to(torch.bfloat16)
+    vl = torch.randn(bl, 40, s_long, 128, device="cuda").to(torch.bfloat16)
+    for _ in range(2):
+        flash_attention(ql, kl, vl, scale)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(5):
+        flash_attention(ql, kl, vl, scale)
+    torch.cuda.synchronize()
+    dtl = (time.perf_counter() - t0) / 5
+    fl = 2 * 2 * bl * 40 * s_long * s_long * 128 * 0.5
+    print(f"flash fwd s={s_long}: {dtl*1000:.2f} ms, {fl/dtl/1e12:.1f} TF")
