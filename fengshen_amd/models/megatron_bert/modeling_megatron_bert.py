@@ -110,7 +110,8 @@ class MegatronBertModel(MegatronBertPreTrainedModel):
         self.embeddings.word_embeddings = v
 
     def forward(self, input_ids, attention_mask=None, token_type_ids=None,
-                position_ids=None, **_kw):
+                position_ids=None, output_hidden_states: bool = False,
+                **_kw):
         h = self.embeddings(input_ids, token_type_ids, position_ids)
         mask = None
         if attention_mask is not None:
@@ -121,6 +122,7 @@ class MegatronBertModel(MegatronBertPreTrainedModel):
             else:
                 # HF 1=keep [b,s] -> internal True=masked [b,1,1,s]
                 mask = (attention_mask == 0)[:, None, None, :]
+        all_h = (h,) if output_hidden_states else None
         skip = self.gradient_checkpointing_skip_interval
         for i, layer in enumerate(self.encoder):
             ckpt = self.gradient_checkpointing and self.training
@@ -131,10 +133,15 @@ class MegatronBertModel(MegatronBertPreTrainedModel):
                     lambda x, m, lyr=layer: lyr(x, attention_mask=m), h, mask)
             else:
                 h = layer(h, attention_mask=mask)
+            if output_hidden_states:
+                all_h = all_h + (h,)
         h = self.ln_f(h)
         pooled = None
         if self.pooler is not None:
             pooled = torch.tanh(self.pooler(h[:, 0]))
+        if output_hidden_states:
+            # HF convention: embeddings output + every layer output
+            return BaseModelOutput(last_hidden_state=h, hidden_states=all_h)
         return BaseModelOutput(last_hidden_state=h,
                                hidden_states=(pooled,) if pooled is not None else None)
 

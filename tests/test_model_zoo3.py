@@ -178,3 +178,55 @@ def test_paraphrase_and_reasoning_generate():
                               end_token_id=119)
     assert len(outs) == 1
     assert en_to_zh("a,b.") == "a，b。"
+
+
+# ---------------------------------------------------------------------------
+# deepVAE (Della) reference mechanisms (ref models/deepVAE/deep_vae.py)
+# ---------------------------------------------------------------------------
+def test_deepvae_della_mechanisms():
+    from fengshen_amd.models.deep_vae.modeling_deep_vae import (
+        DeepVAEModel, deep_vae_tiny_config)
+    torch.manual_seed(0)
+    m = DeepVAEModel(deep_vae_tiny_config())
+    ids = torch.randint(3, 250, (2, 12))
+    out = m(ids, labels=ids)
+    assert out.loss.isfinite()
+    # learned prior: KL is Gaussian-vs-Gaussian per layer
+    assert len(out.layer_kl) == m.layer_num
+    out.loss.backward()
+    # prior nets receive gradient (learned prior, not N(0,1)); layer 0's
+    # prior input is the zero vector (ref comment :91), so check layer 1
+    assert m.prior_nets[1].weight.grad is not None
+    assert m.prior_nets[1].weight.grad.abs().sum() > 0
+    # recursion net ties z across layers
+    assert m.latent_nets[0].W_hh.weight.grad is not None
+    gen = m.inference(ids, max_length=5, top_p=0.9, sample=True)
+    assert gen.shape[0] == 2 and gen.shape[1] <= 6
+
+
+def test_deepvae_cvae_mode():
+    from fengshen_amd.models.deep_vae.modeling_deep_vae import (
+        DeepVAEModel, deep_vae_tiny_config)
+    torch.manual_seed(0)
+    m = DeepVAEModel(deep_vae_tiny_config(cvae=True))
+    ids = torch.randint(3, 250, (2, 8))
+    cond = torch.randint(3, 250, (2, 4))
+    out = m(ids, labels=ids, cond_inputs=cond)
+    assert out.loss.isfinite()
+    out.loss.backward()
+    # CVAE inference conditions on the prefix and continues from it
+    gen = m.inference(cond, max_length=4, top_p=0.9)
+    assert gen.shape[1] >= cond.shape[1]
+
+
+def test_bert_output_hidden_states():
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.models.megatron_bert.modeling_megatron_bert import (
+        MegatronBertModel)
+    m = MegatronBertModel(bert_tiny_config(), add_pooling_layer=False)
+    ids = torch.randint(3, 250, (2, 10))
+    out = m(ids, output_hidden_states=True)
+    # embeddings + one per layer
+    assert len(out.hidden_states) == m.config.num_hidden_layers + 1
+    assert out.hidden_states[-1].shape == (2, 10, m.config.hidden_size)
