@@ -31,6 +31,8 @@ def _run_bench(nproc, extra):
     # the driver's N=4 scaling point and the BASELINE config-4 topology
     (4, ["--zero_stage", "3"], "zero3_dp4"),
     (4, ["--zero_stage", "3", "--tp", "2"], "zero3_dp2_tp2"),
+    # the driver's N=8 headline point
+    (8, ["--zero_stage", "3"], "zero3_dp8"),
 ])
 def test_bench_multirank_json(nproc, extra, parallelism):
     res = _run_bench(nproc, extra)
