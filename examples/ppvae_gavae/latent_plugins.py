@@ -50,9 +50,9 @@ def main():
         out.loss.backward()
         opt.step()
     print(f"ppvae final loss {out.loss.item():.4f} "
-          f"(rec {out.rec_loss.item():.4f}, kl {out.kl_loss.item():.4f})")
+          f"(kl {out.kl_loss.item():.4f})")
     # sample the bottleneck prior -> latent -> decode one step
-    z = ppvae.decoder(torch.randn(2, ppvae.config.bottleneck_dim))
+    z = ppvae.gen_latent(2)
     bos = torch.full((2, 1), 5, dtype=torch.long)
     h = base.decode(z, bos)
     print("ppvae-decoded first-step hidden norm:",

@@ -117,6 +117,28 @@ class DAVAEModel(PreTrainedModel):
         return DAVAEOutput(loss=loss, rec_loss=rec, kl_loss=kl, logits=logits)
 
     @torch.no_grad()
+    def sample_from_latent(self, z: torch.Tensor, seq_len: int,
+                           bos_id: int = 5):
+        """Greedy decode text ids from given latent codes (ref DAVAEModel
+        text_from_latent_code_batch — used by the PPVAE/GAVAE plug-ins)."""
+        n = z.shape[0]
+        device = z.device
+        ids = torch.full((n, 1), bos_id, dtype=torch.long, device=device)
+        for _ in range(seq_len - 1):
+            h = self.decode(z, ids)
+            logits = parallel_lm_logits(h[:, -1:], self.decoder.wte.weight,
+                                        parallel_output=False)
+            ids = torch.cat([ids, logits[:, -1].argmax(-1, keepdim=True)],
+                            dim=1)
+        return ids
+
+    @torch.no_grad()
+    def latent_code_from_text_batch(self, input_ids, attention_mask=None):
+        """Posterior mean latents for a batch (ref naming)."""
+        mu, _logvar = self.encode(input_ids, attention_mask)
+        return mu
+
+    @torch.no_grad()
     def sample(self, n: int, seq_len: int, device=None, bos_id: int = 5):
         """unconditional generation from the prior."""
         device = device or next(self.parameters()).device

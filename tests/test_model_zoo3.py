@@ -7,7 +7,7 @@ def test_ppvae():
     from fengshen_amd.models.ppvae.modeling_ppvae import (
         PPVAEModel, PPVAEConfig)
     torch.manual_seed(0)
-    m = PPVAEModel(PPVAEConfig(latent_dim=32))
+    m = PPVAEModel(PPVAEConfig(latent_dim=32, bottle_dim=8))
     lat = torch.randn(8, 32)
     out = m(lat)
     assert out.loss.isfinite()
@@ -230,3 +230,48 @@ def test_bert_output_hidden_states():
     # embeddings + one per layer
     assert len(out.hidden_states) == m.config.num_hidden_layers + 1
     assert out.hidden_states[-1].shape == (2, 10, m.config.hidden_size)
+
+
+def test_ppvae_conditional_train_plugin():
+    """Reference train_plugin (:94-160): pos/neg conditional training
+    with detached-negative threshold, dynamic beta, early stopping;
+    gen_latent decodes bottleneck samples to big-VAE latents."""
+    from fengshen_amd.models.ppvae.modeling_ppvae import (
+        PPVAEConfig, PPVAEModel)
+    torch.manual_seed(0)
+    cfg = PPVAEConfig(latent_dim=32, bottle_dim=8, total_epoch=6,
+                      batch_size=8, gamma=0.5, get_dymanic_beta=True,
+                      beta_total_step=10)
+    m = PPVAEModel(cfg)
+    pos = torch.randn(32, 32) + 2.0   # conditional cluster
+    neg = torch.randn(16, 32) - 2.0
+    log = []
+    m.train_plugin(pos, neg_latents=neg, log=log)
+    assert len(log) >= 1
+    m.pluginvae.eval()
+    # plug-in reconstructs positives better than negatives after training
+    with torch.no_grad():
+        pos_rec, _ = m.pluginvae(pos)
+        neg_rec, _ = m.pluginvae(neg)
+        pos_err = ((pos_rec - pos) ** 2).mean()
+        neg_err = ((neg_rec - neg) ** 2).mean()
+    assert pos_err < neg_err
+    z = m.gen_latent(4)
+    assert z.shape == (4, 32)
+
+
+def test_ppvae_generate_through_davae():
+    from fengshen_amd.models.davae.modeling_davae import (
+        DAVAEModel, davae_tiny_config)
+    from fengshen_amd.models.ppvae.modeling_ppvae import (
+        PPVAEConfig, PPVAEModel)
+    torch.manual_seed(0)
+    vae = DAVAEModel(davae_tiny_config()).eval()
+    m = PPVAEModel(PPVAEConfig(latent_dim=vae.config.latent_dim,
+                               bottle_dim=8), vae_model=vae)
+    ids = m.generate(2, seq_len=6)
+    assert ids.shape == (2, 6)
+    # latent helpers used by the plug-in recipe
+    text_ids = torch.randint(3, 250, (2, 10))
+    lat = vae.latent_code_from_text_batch(text_ids)
+    assert lat.shape == (2, vae.config.latent_dim)
