@@ -23,6 +23,11 @@ class RequestData(BaseModel):
     extra: Optional[dict] = None
 
 
+class GenerateRequest(BaseModel):
+    input_text: str
+    max_new_tokens: int = 64
+
+
 class APIConfig(BaseModel):
     pipeline_type: str = "text_classification"
     model: Optional[str] = None
@@ -72,6 +77,17 @@ def build_app(config: APIConfig, pipeline=None):
         logger.info("request: %.80s", req.input_text)
         result = pipeline(req.input_text)
         return {"result": result}
+
+    if hasattr(pipeline, "generate"):
+        # generation endpoint (e.g. a model wrapped with the hipGraph
+        # GraphedDecoder for launch-free decode; ref alt serving path
+        # examples/mt5_summary/fastapi_mt5_summary.py:28-40)
+        @app.post("/generate")
+        def generate(req: GenerateRequest) -> Any:
+            logger.info("generate: %.80s", req.input_text)
+            out = pipeline.generate(req.input_text,
+                                    max_new_tokens=req.max_new_tokens)
+            return {"result": out}
 
     return app
 

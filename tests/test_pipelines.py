@@ -103,3 +103,24 @@ def test_cli_help_paths():
     from fengshen_amd.cli.fengshen_pipeline import main
     assert main([]) == 1
     assert main(["nonexistent_task", "train"]) == 1
+
+
+def test_serving_generate_endpoint():
+    """/generate appears when the pipeline exposes .generate (the
+    hipGraph-decode serving path)."""
+    from fengshen_amd.serving.main import APIConfig, build_app
+
+    class GenPipe:
+        def __call__(self, text):
+            return {"echo": text}
+
+        def generate(self, text, max_new_tokens=8):
+            return text + "!" * min(max_new_tokens, 3)
+
+    app = build_app(APIConfig(pipeline_type="demo"), pipeline=GenPipe())
+    from fastapi.testclient import TestClient
+    c = TestClient(app)
+    r = c.post("/generate", json={"input_text": "你好",
+                                  "max_new_tokens": 2})
+    assert r.status_code == 200
+    assert r.json()["result"].startswith("你好")
