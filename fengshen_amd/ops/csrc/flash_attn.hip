@@ -1548,7 +1548,7 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
     f32x16 st;
 #pragma unroll
     for (int r = 0; r < 16; ++r) st[r] = 0.f;
-#pragma unroll 4
+#pragma unroll
     for (int c = 0; c < NC; ++c) {
       bf16x8 ka = *reinterpret_cast<const bf16x8*>(
           Kp + (long)my_kv * D + c * 16 + hi * 8);
@@ -1560,7 +1560,7 @@ void flash_attn_bwd_dkv_v3_kernel(const bf16_t* __restrict__ Q,
     f32x16 dpt;
 #pragma unroll
     for (int r = 0; r < 16; ++r) dpt[r] = 0.f;
-#pragma unroll 4
+#pragma unroll
     for (int c = 0; c < NC; ++c) {
       bf16x8 va = *reinterpret_cast<const bf16x8*>(
           Vp + (long)my_kv * D + c * 16 + hi * 8);
