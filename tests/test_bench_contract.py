@@ -13,9 +13,10 @@ ROOT = os.path.abspath(os.path.join(os.path.dirname(__file__), ".."))
 def _run_bench(nproc, extra):
     env = dict(os.environ, FENGSHEN_BENCH_CPU="1",
                FENGSHEN_AMD_FORCE_EAGER="1")
+    port = str(29000 + (os.getpid() % 900) + nproc)
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            f"--nproc-per-node={nproc}", "--master-addr", "127.0.0.1",
-           "--master-port", "29513", os.path.join(ROOT, "bench.py"),
+           "--master-port", port, os.path.join(ROOT, "bench.py"),
            "--model", "llama-tiny", "--micro_batch", "2", "--seq_len", "64",
            "--steps", "2", "--warmup", "1"] + extra
     out = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
