@@ -125,6 +125,9 @@ class GraphedDecoder:
         """Eager prefill through the model's normal cache path, then copy
         the per-layer K/V into the static buffers."""
         from transformers.cache_utils import DynamicCache
+        assert prompt_ids.shape[0] == self.batch, (
+            f"decoder captured for batch {self.batch}, "
+            f"got {prompt_ids.shape[0]}")
         plen = prompt_ids.shape[1]
         assert plen + self.max_new <= self.max_len, "window too small"
         cache = DynamicCache()
