@@ -1010,8 +1010,9 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
 
   __shared__ char k_raw[V3_KVBLK * SWB];                // swizzled K rows
   __shared__ short vt_lds[D][V3_KVBLK + FA_VPAD];       // V transposed
-  __shared__ short p_x[FA_WAVES][32][V3_KVBLK + FA_VPAD];  // P exchange
-  __shared__ float a_x[FA_WAVES][32];                   // alpha / l bcast
+  // (P stays fully in registers via cvt-pack + permlane32_swap — T12;
+  // alpha/1-l broadcasts ride lane shuffles: no LDS exchange, no
+  // wave-wide lgkmcnt drains in the softmax/PV region)
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -1177,39 +1178,54 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
       }
     }
 
-    // broadcast alpha to the lanes holding this row's O accumulators
-    if (hi == 0) a_x[wave][ln] = alpha;
-    __builtin_amdgcn_s_waitcnt(0);
+    // rescale: alpha for o-row crow(r,hi) fetched by lane shuffle
+    // (alpha is per-q-row, identical in lanes l and l+32)
     if (__any(rescale)) {
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const float al = a_x[wave][(r & 3) + 8 * (r >> 2) + 4 * hi];
+        const float al = __shfl(alpha, (r & 3) + 8 * (r >> 2) + 4 * hi, 64);
 #pragma unroll
         for (int t = 0; t < NTO; ++t) o_acc[t][r] *= al;
       }
     }
 
-    // ---- P -> per-wave LDS exchange (8B runs of 4 kv) -------------------
+    // ---- P -> A-fragments in registers (T12) ----------------------------
+    // Lane layout: p holds, for its own q row, kv octs G = 4*nt + g with
+    // 4-runs at kv = 8G + 4*hi + j.  A-frag for PV chunk c2 needs kv
+    // 16c2 + 8*hi' + j: word pair k from oct 2c2 (hi=0 half) and oct
+    // 2c2+1 (hi=1 half).  One permlane32_swap per word yields BOTH
+    // halves' targets: out0 = words 0/1, out1 = words 2/3, uniformly.
+    unsigned int pw[8][2];  // [oct][word]: packed bf16 pairs of own run
 #pragma unroll
-    for (int nt = 0; nt < 2; ++nt) {
+    for (int G = 0; G < 8; ++G) {
+      const int base = (G >> 2) * 16 + (G & 3) * 4;
 #pragma unroll
-      for (int g = 0; g < 4; ++g) {
-        short pk[4] __attribute__((aligned(8)));
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          pk[j] = fa_bf16bits(p[nt * 16 + g * 4 + j]);
-        *reinterpret_cast<long*>(
-            &p_x[wave][ln][nt * 32 + 8 * g + 4 * hi]) =
-            *reinterpret_cast<const long*>(pk);
+      for (int k = 0; k < 2; ++k) {
+        const unsigned int lo =
+            (unsigned short)fa_bf16bits(p[base + 2 * k]);
+        const unsigned int hi_b =
+            (unsigned short)fa_bf16bits(p[base + 2 * k + 1]);
+        pw[G][k] = (hi_b << 16) | lo;
       }
     }
-    __builtin_amdgcn_s_waitcnt(0);
 
     // ---- PV: O[q][d] += P[q][kv] @ V[kv][d] -----------------------------
 #pragma unroll
     for (int c2 = 0; c2 < 4; ++c2) {  // kv chunks of 16
-      bf16x8 pa = *reinterpret_cast<const bf16x8*>(
-          &p_x[wave][ln][c2 * 16 + hi * 8]);
+      unsigned int paw[4];
+#pragma unroll
+      for (int k = 0; k < 2; ++k) {
+        auto pr = __builtin_amdgcn_permlane32_swap(
+            (int)pw[2 * c2][k], (int)pw[2 * c2 + 1][k], false, false);
+        paw[k] = (unsigned int)pr[0];
+        paw[k + 2] = (unsigned int)pr[1];
+      }
+      bf16x8 pa;
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        pa[2 * k] = (short)(paw[k] & 0xFFFF);
+        pa[2 * k + 1] = (short)(paw[k] >> 16);
+      }
 #pragma unroll
       for (int t = 0; t < NTO; ++t) {   // d cols, 32 each
         const int dcol = t * 32 + ln;
@@ -1223,13 +1239,12 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
 
   // ---- epilogue ---------------------------------------------------------
   if (!q_active) return;
-  // broadcast 1/l to o-accumulator lanes via the alpha buffer
-  if (hi == 0) a_x[wave][ln] = (l_run > 0.f) ? 1.f / l_run : 0.f;
-  __builtin_amdgcn_s_waitcnt(0);
+  const float my_inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int qrow = q0w + (r & 3) + 8 * (r >> 2) + 4 * hi;
-    const float inv_l = a_x[wave][(r & 3) + 8 * (r >> 2) + 4 * hi];
+    const float inv_l =
+        __shfl(my_inv_l, (r & 3) + 8 * (r >> 2) + 4 * hi, 64);
 #pragma unroll
     for (int t = 0; t < NTO; ++t) {
       Op[(long)qrow * D + t * 32 + ln] =
