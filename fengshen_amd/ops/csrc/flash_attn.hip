@@ -1008,9 +1008,11 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
   constexpr int NTO = D / 32;    // 32-wide output d-tiles
   constexpr int SWB = fa_swb(D);
 
-  __shared__ char k_raw[V3_KVBLK * SWB];                // swizzled K rows
-  __shared__ short vt_lds[D][V3_KVBLK + FA_VPAD];       // V transposed
-  // (P stays fully in registers via cvt-pack + permlane32_swap — T12;
+  __shared__ char k_raw[2][V3_KVBLK * SWB];             // swizzled K rows
+  __shared__ short vt_lds[2][D][V3_KVBLK + FA_VPAD];    // V transposed
+  // (double-buffered: tile t+1 stages into buf^1 while buf holds tile t,
+  // so the loop needs ONE barrier per tile instead of two.
+  // P stays fully in registers via cvt-pack + permlane32_swap — T12;
   // alpha/1-l broadcasts ride lane shuffles: no LDS exchange, no
   // wave-wide lgkmcnt drains in the softmax/PV region)
 
@@ -1079,24 +1081,27 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
     }
   }
 
-  for (int kt = 0; kt < n_kv_tiles; ++kt) {
-    const int k_base = kt * V3_KVBLK;
-    __syncthreads();
+  // stage tile 0 into buffer 0
+  {
 #pragma unroll
     for (int sweep = 0; sweep < SWEEPS; ++sweep) {
       const int i = tid * 8 + sweep * (FA_WAVES * 64 * 8);
       if (i < ELEMS) {
         const int kr = i / D;
         const int kc = i % D;
-        *reinterpret_cast<bf16x8*>(k_raw + kr * SWB + kswz(kr, kc * 2)) =
-            k_reg[sweep];
+        *reinterpret_cast<bf16x8*>(
+            k_raw[0] + kr * SWB + kswz(kr, kc * 2)) = k_reg[sweep];
         bf16x8 vv = v_reg[sweep];
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          vt_lds[kc + j][kr ^ ((kc + j) & 0x38)] = vv[j];
+          vt_lds[0][kc + j][kr ^ ((kc + j) & 0x38)] = vv[j];
       }
     }
-    __syncthreads();
+  }
+  int cur = 0;
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int k_base = kt * V3_KVBLK;
+    __syncthreads();  // publishes buf[cur]; everyone done with buf[cur^1]
     if (kt + 1 < n_kv_tiles) {
       const int nb = (kt + 1) * V3_KVBLK;
 #pragma unroll
@@ -1124,7 +1129,7 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
         bf16x8 a = *reinterpret_cast<const bf16x8*>(
-            k_raw + krow * SWB + kswz(krow, (c * 16 + hi * 8) * 2));
+            k_raw[cur] + krow * SWB + kswz(krow, (c * 16 + hi * 8) * 2));
         st[nt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, q_frag[c],
                                                          st[nt], 0, 0, 0);
       }
@@ -1230,10 +1235,30 @@ void flash_attn_fwd_v3_kernel(const bf16_t* __restrict__ Q,
       for (int t = 0; t < NTO; ++t) {   // d cols, 32 each
         const int dcol = t * 32 + ln;
         bf16x8 vb = *reinterpret_cast<const bf16x8*>(
-            &vt_lds[dcol][(c2 * 16 + hi * 8) ^ (dcol & 0x38)]);
+            &vt_lds[cur][dcol][(c2 * 16 + hi * 8) ^ (dcol & 0x38)]);
         o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, vb, o_acc[t],
                                                            0, 0, 0);
       }
+    }
+
+    // stage tile kt+1 into the spare buffer (loop-top barrier publishes)
+    if (kt + 1 < n_kv_tiles) {
+      const int nxt = cur ^ 1;
+#pragma unroll
+      for (int sweep = 0; sweep < SWEEPS; ++sweep) {
+        const int i = tid * 8 + sweep * (FA_WAVES * 64 * 8);
+        if (i < ELEMS) {
+          const int kr = i / D;
+          const int kc = i % D;
+          *reinterpret_cast<bf16x8*>(
+              k_raw[nxt] + kr * SWB + kswz(kr, kc * 2)) = k_reg[sweep];
+          bf16x8 vv = v_reg[sweep];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            vt_lds[nxt][kc + j][kr ^ ((kc + j) & 0x38)] = vv[j];
+        }
+      }
+      cur = nxt;
     }
   }
 
