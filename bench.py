@@ -297,8 +297,29 @@ def main():
         opt.step()
         return out.loss
 
-    for _ in range(args.warmup):
-        loss = step()
+    # Warmup doubles as an OOM self-check for the auto ckpt-skip estimate:
+    # the heuristic can overshoot on topologies we could not measure
+    # (e.g. dp8 shards optimizer state 8x, freeing HBM the activation
+    # model then over-claims).  Warmup steps are untimed, so on OOM we
+    # tighten the skip and retry instead of failing the whole run.
+    warm_left = args.warmup
+    while warm_left > 0:
+        try:
+            loss = step()
+            warm_left -= 1
+        except torch.OutOfMemoryError:
+            # skip semantics: 0 = checkpoint all (min memory), 1 = store
+            # all (max memory), k>=2 = store every k-th layer
+            if args.ckpt_skip >= 0 or skip == 0:
+                raise  # user-pinned or already at minimum memory
+            skip = 2 if skip == 1 else (
+                skip + 1 if skip < model.config.num_hidden_layers else 0)
+            opt.zero_grad()
+            torch.cuda.empty_cache()
+            model.gradient_checkpointing_enable(skip_interval=skip)
+            if rank == 0:
+                print(f"[bench] OOM in warmup; ckpt_skip -> {skip}",
+                      flush=True)
 
     if world > 1:
         dist.barrier()
