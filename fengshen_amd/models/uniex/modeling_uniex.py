@@ -255,3 +255,148 @@ class UniEXExtractor:
                               "type_score": float(tscores[t])})
             results.append(spans)
         return results
+
+
+# ---------------------------------------------------------------------------
+# Task metrics (ref modeling_uniex.py:44-245) and data-side encoding
+# ---------------------------------------------------------------------------
+def _entity_key(e):
+    return (e.get("entity_type"), tuple(map(tuple, e.get("entity_index",
+                                                         []))) if
+            isinstance(e.get("entity_index"), list)
+            else e.get("entity_index"))
+
+
+def get_entity_f1(test_data, pred_data):
+    """(f1, recall, precision) over unique (type, index) entities
+    (ref :44-100); falls back to spo subjects/objects when entity_list
+    is empty."""
+    corr = y_true = y_pred = 0
+    for t, p in zip(test_data, pred_data):
+        def collect(item):
+            out = []
+            for e in item.get("entity_list", []):
+                k = _entity_key(e)
+                if k not in out:
+                    out.append(k)
+            if not out:
+                for spo in item.get("spo_list", []):
+                    for side in ("subject", "object"):
+                        k = _entity_key(spo[side])
+                        if k not in out:
+                            out.append(k)
+            return out
+        tl, pl = collect(t), collect(p)
+        y_true += len(tl)
+        y_pred += len(pl)
+        corr += sum(1 for e in pl if e in tl)
+    precise = corr / y_pred if y_pred > 0 else 0
+    recall = corr / y_true if y_true > 0 else 0
+    f1 = 2 * precise * recall / (precise + recall) \
+        if precise + recall > 0 else 0
+    return f1, recall, precise
+
+
+def get_rel_f1(test_data, pred_data):
+    """(f1, recall, precision) over (predicate, subject, object) triples
+    (ref :157-162 wrapper over the strict entity matcher)."""
+    corr = y_true = y_pred = 0
+    for t, p in zip(test_data, pred_data):
+        def collect(item):
+            out = []
+            for spo in item.get("spo_list", []):
+                k = (spo.get("predicate"),
+                     _entity_key(spo.get("subject", {})),
+                     _entity_key(spo.get("object", {})))
+                if k not in out:
+                    out.append(k)
+            return out
+        tl, pl = collect(t), collect(p)
+        y_true += len(tl)
+        y_pred += len(pl)
+        corr += sum(1 for e in pl if e in tl)
+    precise = corr / y_pred if y_pred > 0 else 0
+    recall = corr / y_true if y_true > 0 else 0
+    f1 = 2 * precise * recall / (precise + recall) \
+        if precise + recall > 0 else 0
+    return f1, recall, precise
+
+
+class UniEXDataEncoder:
+    """Entity-task sample encoding (ref UniEXDataEncode :246-798, entity
+    subset): sequence = [CLS][unused-index][type_1]..[type_T][SEP] text;
+    label_token_idx covers the index token + type tokens; span labels
+    [s, s, 1+T] carry an index-head hit and a type-head hit per entity
+    span; span_labels_mask opens the text block for the index head and
+    text×text for type heads."""
+
+    def __init__(self, tokenizer, max_length: int = 128):
+        self.tokenizer = tokenizer
+        self.max_length = max_length
+
+    def encode(self, item: dict, entity_type_list: List[str]) -> dict:
+        tk = self.tokenizer
+        T = len(entity_type_list)
+        ids = [tk.cls_token_id]
+        label_idx = [len(ids) - 1]  # the [CLS] doubles as the index token
+        for et in entity_type_list:
+            label_idx.append(len(ids))
+            ids += tk.encode(et, add_special_tokens=False)
+        ids.append(tk.sep_token_id)
+        text_start = len(ids)
+        text_ids = tk.encode(item["text"], add_special_tokens=False)
+        ids += text_ids
+        ids = ids[:self.max_length]
+        text_token_idx = list(range(text_start,
+                                    min(len(ids), self.max_length)))
+        tlen = len(text_token_idx)
+
+        s = len(ids)
+        span_labels = torch.zeros(s, s, 1 + T)
+        span_mask = torch.full((s, s, 1 + T), -10000.0)
+        # index head: text block only
+        span_mask[text_start:, text_start:, 0] = 0.0
+        span_mask[text_start:, text_start:, 1:] = 0.0
+        for e in item.get("entity_list", []):
+            et = e["entity_type"]
+            if et not in entity_type_list:
+                continue
+            ti = entity_type_list.index(et)
+            for st_c, en_c in e.get("entity_index", []):
+                st = text_start + len(tk.encode(item["text"][:st_c],
+                                                add_special_tokens=False))
+                en = text_start + len(tk.encode(item["text"][:en_c + 1],
+                                                add_special_tokens=False)) - 1
+                if st < s and en < s:
+                    span_labels[st, en, 0] = 1        # index head
+                    span_labels[st, en, 1 + ti] = 1   # type head
+        return {
+            "input_ids": torch.tensor(ids, dtype=torch.long),
+            "label_token_idx": torch.tensor(label_idx, dtype=torch.long),
+            "text_token_idx": torch.tensor(text_token_idx,
+                                           dtype=torch.long),
+            "span_labels": span_labels,
+            "span_labels_mask": span_mask,
+            "text_start": text_start,
+            "tlen": tlen,
+        }
+
+    def collate(self, samples: List[dict]) -> dict:
+        smax = max(x["input_ids"].shape[0] for x in samples)
+        tmax = max(x["tlen"] for x in samples)
+        nlab = samples[0]["span_labels"].shape[-1]
+        b = len(samples)
+        ids = torch.zeros(b, smax, dtype=torch.long)
+        lab_idx = torch.stack([x["label_token_idx"] for x in samples])
+        txt_idx = torch.zeros(b, tmax, dtype=torch.long)
+        sl = torch.zeros(b, smax, smax, nlab)
+        sm = torch.full((b, smax, smax, nlab), -10000.0)
+        for i, x in enumerate(samples):
+            n = x["input_ids"].shape[0]
+            ids[i, :n] = x["input_ids"]
+            txt_idx[i, :x["tlen"]] = x["text_token_idx"]
+            sl[i, :n, :n] = x["span_labels"]
+            sm[i, :n, :n] = x["span_labels_mask"]
+        return {"input_ids": ids, "label_token_idx": lab_idx,
+                "text_token_idx": txt_idx, "span_labels": sl,
+                "span_labels_mask": sm}

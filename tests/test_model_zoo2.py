@@ -255,3 +255,57 @@ def test_int8_quantization_close():
         q = m(ids).logits
     rel = (q - ref).abs().max() / ref.abs().max()
     assert rel < 0.1, rel.item()
+
+
+def test_uniex_metrics():
+    """get_entity_f1 / get_rel_f1 (ref modeling_uniex.py:44-162)."""
+    from fengshen_amd.models.uniex.modeling_uniex import (
+        get_entity_f1, get_rel_f1)
+    test = [{"entity_list": [
+        {"entity_type": "人名", "entity_index": [[0, 1]]},
+        {"entity_type": "地名", "entity_index": [[4, 5]]}]}]
+    pred = [{"entity_list": [
+        {"entity_type": "人名", "entity_index": [[0, 1]]},
+        {"entity_type": "地名", "entity_index": [[3, 5]]}]}]
+    f1, r, p = get_entity_f1(test, pred)
+    assert abs(p - 0.5) < 1e-9 and abs(r - 0.5) < 1e-9
+    assert abs(f1 - 0.5) < 1e-9
+    # spo fallback when entity_list is empty
+    t2 = [{"spo_list": [{"predicate": "位于",
+                         "subject": {"entity_type": "机构",
+                                     "entity_index": [[0, 1]]},
+                         "object": {"entity_type": "地名",
+                                    "entity_index": [[4, 5]]}}]}]
+    f1e, _, _ = get_entity_f1(t2, t2)
+    assert f1e == 1.0
+    f1r, _, _ = get_rel_f1(t2, t2)
+    assert f1r == 1.0
+
+
+def test_uniex_data_encoder_trains():
+    """UniEXDataEncoder feeds the triaffine training path end-to-end and
+    the index-head grid marks the char-aligned entity span."""
+    from fengshen_amd.models.megatron_bert.configuration_megatron_bert import (
+        bert_tiny_config)
+    from fengshen_amd.models.uniex.modeling_uniex import (
+        UniEXDataEncoder, UniEXModel)
+    from fengshen_amd.tokenizer import SimpleCharTokenizer
+    torch.manual_seed(0)
+    tk = SimpleCharTokenizer()
+    enc = UniEXDataEncoder(tk, max_length=48)
+    types = ["人名", "地名"]
+    items = [{"text": "ab cd ef",
+              "entity_list": [{"entity_type": "人名",
+                               "entity_index": [[0, 1]]}]},
+             {"text": "xy zw", "entity_list": []}]
+    batch = enc.collate([enc.encode(it, types) for it in items])
+    assert batch["span_labels"].shape[-1] == 1 + len(types)
+    # index + type heads hit at the encoded span
+    s0 = enc.encode(items[0], types)
+    ts = s0["text_start"]
+    assert s0["span_labels"][ts, ts + 1, 0] == 1
+    assert s0["span_labels"][ts, ts + 1, 1] == 1
+    m = UniEXModel(bert_tiny_config(), triaffine_hidden_size=32)
+    out = m(**batch)
+    assert out.loss.isfinite()
+    out.loss.backward()
