@@ -29,6 +29,12 @@ def main():
     p.add_argument("--quant", choices=["bf16", "int8"], default="bf16")
     p.add_argument("--graph", action="store_true",
                    help="hipGraph-captured decode loop (serving fast path)")
+    p.add_argument("--sample", action="store_true",
+                   help="top-k/top-p sampling inside the graph "
+                        "(gumbel-max; greedy otherwise)")
+    p.add_argument("--top_k", type=int, default=50)
+    p.add_argument("--top_p", type=float, default=0.95)
+    p.add_argument("--temperature", type=float, default=0.8)
     args = p.parse_args()
     assert torch.cuda.is_available()
 
@@ -58,7 +64,9 @@ def main():
         from fengshen_amd.serving.graphed_decode import GraphedDecoder
         dec = GraphedDecoder(model, batch=args.batch,
                              max_len=args.prompt_len + args.new_tokens + 8,
-                             max_new_tokens=args.new_tokens)
+                             max_new_tokens=args.new_tokens,
+                             do_sample=args.sample, top_k=args.top_k,
+                             top_p=args.top_p, temperature=args.temperature)
         dec.generate(ids, max_new_tokens=4)  # warmup + capture
         torch.cuda.synchronize()
         t0 = time.perf_counter()

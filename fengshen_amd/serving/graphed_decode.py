@@ -27,9 +27,35 @@ import torch
 from fengshen_amd.ops import functional as F_ops
 
 
+def filter_logits(logits: torch.Tensor, temperature: torch.Tensor,
+                  top_k: int, top_p: torch.Tensor) -> torch.Tensor:
+    """HF-convention sampling filter, graph-safe (static shapes only).
+
+    temperature / top_p are device scalars so they can be retuned between
+    generates without recapturing; top_k is shape-affecting and fixed at
+    capture.  Order matches the reference's sample_sequence helpers
+    (fengshen/utils/transfo_xl_utils.py top_k_logits): temperature, then
+    top-k, then nucleus.
+    """
+    lg = logits.float() / temperature
+    ninf = torch.tensor(float("-inf"), device=lg.device)
+    if top_k > 0:
+        kth = lg.topk(top_k, dim=-1).values[..., -1:]
+        lg = torch.where(lg < kth, ninf, lg)
+    # nucleus: drop tokens once the cumulative prob BEFORE them >= top_p
+    # (the top-1 token always survives).  top_p >= 1 keeps everything.
+    srt, idx = lg.sort(dim=-1, descending=True)
+    probs = torch.softmax(srt, dim=-1)
+    shifted_cum = probs.cumsum(dim=-1) - probs
+    srt = torch.where(shifted_cum >= top_p, ninf, srt)
+    return torch.full_like(lg, float("-inf")).scatter(-1, idx, srt)
+
+
 class GraphedDecoder:
     def __init__(self, model, batch: int = 1, max_len: int = 512,
-                 max_new_tokens: int = 256):
+                 max_new_tokens: int = 256, do_sample: bool = False,
+                 top_k: int = 0, top_p: float = 1.0,
+                 temperature: float = 1.0):
         self.model = model.eval()
         cfg = model.config
         dev = next(model.parameters()).device
@@ -59,6 +85,16 @@ class GraphedDecoder:
             max_len, hd, base=getattr(cfg, "rope_base", 10000.0))
         self.cos = cos.to(dev)
         self.sin = sin.to(dev)
+        # sampling state: gumbel-max with a pre-generated noise buffer
+        # indexed by step_i — no RNG state inside the graph, and a fixed
+        # torch seed gives reproducible samples.
+        self.do_sample = do_sample
+        self.top_k = top_k
+        self.top_p = torch.tensor(float(top_p), device=dev)
+        self.temperature = torch.tensor(float(temperature), device=dev)
+        if do_sample:
+            self.gumbel = torch.zeros(max_new_tokens, batch,
+                                      cfg.vocab_size, device=dev)
         self._graph: Optional[torch.cuda.CUDAGraph] = None
 
     # ------------------------------------------------------------------
@@ -113,11 +149,24 @@ class GraphedDecoder:
         logits = self.model.lm_head(h)
         if isinstance(logits, tuple):
             logits = logits[0]
-        nxt = logits[:, -1, :].argmax(dim=-1, keepdim=True)  # [b, 1]
+        nxt = self._pick(logits[:, -1, :], self.step_i)  # [b, 1]
         self.out_tokens.index_copy_(1, self.step_i, nxt)
         self.tok.copy_(nxt)
         self.pos.add_(1)
         self.step_i.add_(1)
+
+    def _pick(self, last_logits: torch.Tensor,
+              step: torch.Tensor) -> torch.Tensor:
+        """Greedy argmax, or gumbel-max sample over the filtered logits.
+        `step` indexes the pre-generated noise buffer; graph-safe."""
+        if not self.do_sample:
+            return last_logits.argmax(dim=-1, keepdim=True)
+        lg = filter_logits(last_logits, self.temperature,
+                           self.top_k, self.top_p)
+        # clamp: warmup/capture advance step_i past the real range
+        g = self.gumbel.index_select(
+            0, step.clamp(max=self.max_new - 1)).squeeze(0)  # [b, vocab]
+        return (lg + g).argmax(dim=-1, keepdim=True)
 
     # ------------------------------------------------------------------
     @torch.no_grad()
@@ -142,7 +191,7 @@ class GraphedDecoder:
             self.v_cache[i][:, :, plen:].zero_()
         self.pos.fill_(plen)
         self.step_i.zero_()
-        self.tok.copy_(out.logits[:, -1, :].argmax(-1, keepdim=True))
+        self.tok.copy_(self._pick(out.logits[:, -1, :], self.step_i))
 
     def _capture(self):
         # warm up twice on a side stream (allocator + kernels), then capture
@@ -164,6 +213,10 @@ class GraphedDecoder:
                  eos_token_id: Optional[int] = None) -> torch.Tensor:
         """Greedy decode via graph replays.  Returns [b, plen + n]."""
         n = min(max_new_tokens or self.max_new, self.max_new)
+        if self.do_sample:
+            # fresh Gumbel(0,1) noise per call: -log(Exp(1)); seeded via
+            # torch.manual_seed for reproducible sampling.
+            self.gumbel.exponential_().log_().neg_()
         if self._graph is None:
             # capture against scratch state, then restore via real prefill
             self.pos.fill_(prompt_ids.shape[1])

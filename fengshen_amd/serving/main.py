@@ -26,6 +26,10 @@ class RequestData(BaseModel):
 class GenerateRequest(BaseModel):
     input_text: str
     max_new_tokens: int = 64
+    do_sample: bool = False
+    top_k: int = 0
+    top_p: float = 1.0
+    temperature: float = 1.0
 
 
 class APIConfig(BaseModel):
@@ -85,8 +89,12 @@ def build_app(config: APIConfig, pipeline=None):
         @app.post("/generate")
         def generate(req: GenerateRequest) -> Any:
             logger.info("generate: %.80s", req.input_text)
+            kw = {}
+            if req.do_sample:
+                kw = dict(do_sample=True, top_k=req.top_k,
+                          top_p=req.top_p, temperature=req.temperature)
             out = pipeline.generate(req.input_text,
-                                    max_new_tokens=req.max_new_tokens)
+                                    max_new_tokens=req.max_new_tokens, **kw)
             return {"result": out}
 
     return app

@@ -112,3 +112,30 @@ def test_graphed_decode_matches_eager_generate():
     # matching prefix (first 8 of 12 greedy tokens identical)
     assert torch.equal(out[:, 16:24], ref[:, 16:24]), (
         out[:, 16:].tolist(), ref[:, 16:].tolist())
+
+
+def test_graphed_decode_sampling():
+    """Sampled graphed decode: reproducible under a fixed torch seed
+    (noise buffer is drawn with torch RNG outside the graph), valid ids,
+    and diverse across different seeds."""
+    import torch
+    from fengshen_amd.models.llama.configuration_llama import (
+        llama_tiny_config)
+    from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+    from fengshen_amd.serving.graphed_decode import GraphedDecoder
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(llama_tiny_config()).to(
+        torch.bfloat16).to("cuda").eval()
+    ids = torch.randint(3, model.config.vocab_size, (2, 16), device="cuda")
+    dec = GraphedDecoder(model, batch=2, max_len=64, max_new_tokens=12,
+                         do_sample=True, top_k=20, top_p=0.95,
+                         temperature=0.8)
+    torch.manual_seed(7)
+    a = dec.generate(ids, max_new_tokens=12)
+    torch.manual_seed(7)
+    b = dec.generate(ids, max_new_tokens=12)
+    assert torch.equal(a, b)
+    torch.manual_seed(8)
+    c = dec.generate(ids, max_new_tokens=12)
+    assert not torch.equal(a, c)
+    assert int(a.max()) < model.config.vocab_size and int(a.min()) >= 0
