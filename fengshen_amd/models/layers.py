@@ -45,6 +45,18 @@ def scaled_init_normal(std: float, num_layers: int) -> Callable:
     return init_normal(std / math.sqrt(2.0 * num_layers))
 
 
+def small_init(dim: int) -> Callable:
+    """Nguyen & Salazar transformers-without-tears init: N(0, sqrt(2/5d))
+    (ref layers/init_functions.py small_init_init_method)."""
+    return init_normal(math.sqrt(2.0 / (5.0 * dim)))
+
+
+def wang_init(dim: int, num_layers: int) -> Callable:
+    """Ben Wang's GPT-J output projection init: std = 2/(L*sqrt(d))
+    (ref layers/init_functions.py wang_init_method)."""
+    return init_normal(2.0 / (num_layers * math.sqrt(dim)))
+
+
 class MergedColumnParallelLinear(ColumnParallelLinear):
     """Column-parallel GEMM whose output is a concat of logical segments
     (QKV or gate|up), arranged so each TP rank holds
@@ -93,11 +105,28 @@ class LayerNorm(nn.Module):
         return F_ops.layer_norm(x, self.weight, self.bias, self.eps)
 
 
+class ScaleNorm(nn.Module):
+    """Single learned scale over the l2-normalised vector
+    (ref norms.py:55 ScaleNorm): y = g * x / max(||x||/sqrt(d), eps)."""
+
+    def __init__(self, dim: int, eps: float = 1e-8, dtype=None):
+        super().__init__()
+        self.g = nn.Parameter(torch.ones(1, dtype=dtype))
+        self.dim = dim
+        self.eps = eps
+
+    def forward(self, x):
+        n = x.float().norm(dim=-1, keepdim=True) * self.dim ** -0.5
+        return (x.float() / n.clamp(min=self.eps)).to(x.dtype) * self.g
+
+
 def get_norm(kind: str, dim: int, eps: float, dtype=None) -> nn.Module:
     if kind == "rmsnorm":
         return RMSNorm(dim, eps, dtype)
     if kind == "layernorm":
         return LayerNorm(dim, eps, dtype)
+    if kind == "scalenorm":
+        return ScaleNorm(dim, dtype=dtype)
     raise ValueError(kind)
 
 

@@ -211,3 +211,27 @@ def test_trainer_ckpt_to_hf(tmp_path):
     for (k1, p1), (k2, p2) in zip(mod.model.named_parameters(),
                                   m2.named_parameters()):
         assert k1 == k2 and torch.allclose(p1, p2, atol=1e-6), k1
+
+
+def test_scalenorm_and_init_zoo():
+    """ScaleNorm normalises to unit RMS * g; small_init/wang_init std."""
+    import math
+    import torch
+    from fengshen_amd.models.layers import (ScaleNorm, get_norm,
+                                            small_init, wang_init)
+    torch.manual_seed(0)
+    x = torch.randn(4, 8, 64) * 3.0
+    sn = get_norm("scalenorm", 64, 1e-8)
+    assert isinstance(sn, ScaleNorm)
+    y = sn(x)
+    rms = y.pow(2).mean(-1).sqrt()
+    assert torch.allclose(rms, torch.ones_like(rms), atol=1e-4)
+    with torch.no_grad():
+        sn.g.fill_(2.0)
+    assert torch.allclose(sn(x).pow(2).mean(-1).sqrt(),
+                          2 * torch.ones_like(rms), atol=1e-3)
+    w = torch.empty(4096, 4096)
+    small_init(4096)(w)
+    assert abs(w.std().item() - math.sqrt(2 / (5 * 4096))) < 1e-3
+    wang_init(4096, 24)(w)
+    assert abs(w.std().item() - 2 / (24 * math.sqrt(4096))) < 1e-3
