@@ -54,6 +54,8 @@ void fs_flash_attn_bwd(const void*, const void*, const void*, const void*,
                        const void*, const float*, void*, void*, void*, float*,
                        const int*, int, int, int, int, int, int, float,
                        float, unsigned long long, hipStream_t);
+void fs_bf16_gemv(const void*, const void*, void*, int, int, int,
+                  hipStream_t);
 void fs_w8_gemv(const void*, const float*, const void*, void*, int, int, int,
                 hipStream_t);
 void fs_vocab_ce_fwd(const void*, const long*, float*, float*, float*,
@@ -355,8 +357,24 @@ static at::Tensor w8_gemv(at::Tensor q8, at::Tensor scale, at::Tensor x) {
   return y;
 }
 
+// x [b, in] bf16 (b <= 8), w [out, in] bf16 -> y [b, out] bf16
+static at::Tensor bf16_gemv(at::Tensor w, at::Tensor x) {
+  TORCH_CHECK(w.scalar_type() == at::kBFloat16 &&
+              x.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(w.is_contiguous() && x.is_contiguous());
+  const int out = w.size(0), in = w.size(1);
+  TORCH_CHECK(in % 16 == 0, "in_features must be divisible by 16");
+  const int b = x.numel() / in;
+  TORCH_CHECK(b >= 1 && b <= 8, "bf16_gemv: batch must be 1..8");
+  auto y = at::empty({b, out}, x.options());
+  fs_bf16_gemv(w.data_ptr(), x.data_ptr(), y.data_ptr(), b, in, out,
+               cur_stream());
+  return y;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("w8_gemv", &w8_gemv);
+  mod.def("bf16_gemv", &bf16_gemv);
   mod.def("flash_attn_fwd", &flash_attn_fwd);
   mod.def("vocab_ce_fwd", [](at::Tensor logits2d, at::Tensor targets,
                              int64_t vstart, int64_t vend) {

@@ -898,6 +898,71 @@ extern "C" void fs_w8_gemv(const void* q8, const float* scale, const void* x,
 
 
 // ===========================================================================
+// bf16 GEMV: y[b, o] = sum_i w[o, i] * x[b, i]   (bf16 decode path)
+// hipBLASLt's M=1 GEMM leaves most of HBM bandwidth on the table on
+// gfx950 (tile quantization: tiny M forces skinny tiles).  Decode is a
+// pure weight stream — one wave per output row, lanes stride K with
+// 32-byte loads, and a compile-time batch B<=8 reuses each weight dword
+// for every sequence in the batch.
+// ===========================================================================
+
+template <int B>
+__global__ __launch_bounds__(256)
+void bf16_gemv_kernel(const bf16_t* __restrict__ w,
+                      const bf16_t* __restrict__ x,
+                      bf16_t* __restrict__ y,
+                      int in_features, int out_features) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  for (int row = blockIdx.x * 4 + wid; row < out_features;
+       row += gridDim.x * 4) {
+    const bf16_t* wr = w + (long)row * in_features;
+    float acc[B];
+#pragma unroll
+    for (int b = 0; b < B; ++b) acc[b] = 0.f;
+    for (int i = lane * 16; i + 15 < in_features; i += 64 * 16) {
+      float wv[8], wv2[8];
+      load8<bf16_t>(wr + i, wv);
+      load8<bf16_t>(wr + i + 8, wv2);
+#pragma unroll
+      for (int b = 0; b < B; ++b) {
+        const bf16_t* xb = x + (long)b * in_features;
+        float xv[8], xv2[8];
+        load8<bf16_t>(xb + i, xv);
+        load8<bf16_t>(xb + i + 8, xv2);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc[b] += wv[j] * xv[j] + wv2[j] * xv2[j];
+      }
+    }
+#pragma unroll
+    for (int b = 0; b < B; ++b) {
+      float r = wave_reduce_sum(acc[b]);
+      if (lane == 0)
+        y[(long)b * out_features + row] = from_f32<bf16_t>(r);
+    }
+  }
+}
+
+extern "C" void fs_bf16_gemv(const void* w, const void* x, void* y, int batch,
+                             int in_features, int out_features,
+                             hipStream_t s) {
+  int grid = (out_features + 3) / 4;
+  if (grid > FS_MAX_BLOCKS) grid = FS_MAX_BLOCKS;
+#define FS_GEMV_CASE(B)                                                      \
+  case B:                                                                    \
+    hipLaunchKernelGGL((bf16_gemv_kernel<B>), dim3(grid), dim3(256), 0, s,   \
+                       (const bf16_t*)w, (const bf16_t*)x, (bf16_t*)y,       \
+                       in_features, out_features);                           \
+    break;
+  switch (batch) {
+    FS_GEMV_CASE(1) FS_GEMV_CASE(2) FS_GEMV_CASE(3) FS_GEMV_CASE(4)
+    FS_GEMV_CASE(5) FS_GEMV_CASE(6) FS_GEMV_CASE(7) FS_GEMV_CASE(8)
+  }
+#undef FS_GEMV_CASE
+}
+
+// ===========================================================================
 // Fused vocab cross-entropy (TP-shard-aware)
 // ===========================================================================
 // Replaces the composite vocab_parallel_cross_entropy hot path, which

@@ -308,6 +308,25 @@ def bias_dropout_add(x, bias, residual, p: float, training: bool):
     return _BiasDropoutAdd.apply(x, bias, residual, p, training)
 
 
+def linear(x: torch.Tensor, weight: torch.Tensor,
+           bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """F.linear with a decode fast path: for inference-shaped inputs
+    (<=8 rows, bf16, no bias) the hand-written bf16 GEMV kernel streams
+    the weight matrix at HBM rate instead of hipBLASLt's skinny-M GEMM
+    tiles.  Falls back to F.linear everywhere else (training, CPU,
+    other dtypes) so autograd semantics are unchanged."""
+    if (bias is None and x.is_cuda and x.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16
+            and not torch.is_grad_enabled()
+            and weight.shape[1] % 16 == 0
+            and x.shape[-1] == weight.shape[1]):
+        rows = x.numel() // x.shape[-1]
+        if 1 <= rows <= 8 and use_hip(x):
+            y = get_ext().bf16_gemv(weight.contiguous(), x.contiguous())
+            return y.view(*x.shape[:-1], weight.shape[0])
+    return torch.nn.functional.linear(x, weight, bias)
+
+
 # ---------------------------------------------------------------------------
 # rotary embedding
 # ---------------------------------------------------------------------------

@@ -14,6 +14,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 import torch.nn.init as init
 
+from fengshen_amd.ops import functional as F_ops
 from fengshen_amd.parallel import groups
 from fengshen_amd.parallel.mappings import (
     copy_to_tensor_model_parallel_region,
@@ -140,7 +141,7 @@ class ColumnParallelLinear(nn.Module):
     def forward(self, input_: torch.Tensor):
         input_parallel = copy_to_tensor_model_parallel_region(input_)
         bias = self.bias if not self.skip_bias_add else None
-        output_parallel = F.linear(input_parallel, self.weight, bias)
+        output_parallel = F_ops.linear(input_parallel, self.weight, bias)
         output = (gather_from_tensor_model_parallel_region(output_parallel)
                   if self.gather_output else output_parallel)
         if self.skip_bias_add:
@@ -180,7 +181,7 @@ class RowParallelLinear(nn.Module):
     def forward(self, input_: torch.Tensor):
         input_parallel = (input_ if self.input_is_parallel
                           else scatter_to_tensor_model_parallel_region(input_))
-        output_parallel = F.linear(input_parallel, self.weight)
+        output_parallel = F_ops.linear(input_parallel, self.weight)
         output_ = reduce_from_tensor_model_parallel_region(output_parallel)
         if self.skip_bias_add:
             return output_, self.bias

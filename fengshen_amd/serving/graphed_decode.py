@@ -37,17 +37,19 @@ def filter_logits(logits: torch.Tensor, temperature: torch.Tensor,
     (fengshen/utils/transfo_xl_utils.py top_k_logits): temperature, then
     top-k, then nucleus.
     """
+    # NOTE: no torch.tensor(..., device=...) here — an H2D copy is
+    # hipErrorStreamCaptureUnsupported inside graph capture; masked_fill
+    # with a python scalar lowers to a capture-safe fill kernel.
     lg = logits.float() / temperature
-    ninf = torch.tensor(float("-inf"), device=lg.device)
     if top_k > 0:
         kth = lg.topk(top_k, dim=-1).values[..., -1:]
-        lg = torch.where(lg < kth, ninf, lg)
+        lg = lg.masked_fill(lg < kth, float("-inf"))
     # nucleus: drop tokens once the cumulative prob BEFORE them >= top_p
     # (the top-1 token always survives).  top_p >= 1 keeps everything.
     srt, idx = lg.sort(dim=-1, descending=True)
     probs = torch.softmax(srt, dim=-1)
     shifted_cum = probs.cumsum(dim=-1) - probs
-    srt = torch.where(shifted_cum >= top_p, ninf, srt)
+    srt = srt.masked_fill(shifted_cum >= top_p, float("-inf"))
     return torch.full_like(lg, float("-inf")).scatter(-1, idx, srt)
 
 

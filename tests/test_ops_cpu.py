@@ -98,3 +98,19 @@ def test_bias_dropout_add_eval_path():
     torch.manual_seed(0)
     out = F.bias_dropout_add(x, bias, res, p=0.5, training=True)
     assert out.shape == x.shape
+
+
+def test_fast_linear_cpu_fallback():
+    """F_ops.linear == F.linear on CPU (fast path requires CUDA+bf16) and
+    under autograd."""
+    import torch
+    from fengshen_amd.ops import functional as F_ops
+    torch.manual_seed(0)
+    x = torch.randn(2, 1, 64, requires_grad=True)
+    w = torch.randn(48, 64, requires_grad=True)
+    b = torch.randn(48)
+    y = F_ops.linear(x, w, b)
+    ref = torch.nn.functional.linear(x, w, b)
+    assert torch.equal(y, ref)
+    y.sum().backward()
+    assert x.grad is not None and w.grad is not None
