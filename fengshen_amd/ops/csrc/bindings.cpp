@@ -56,6 +56,9 @@ void fs_flash_attn_bwd(const void*, const void*, const void*, const void*,
                        float, unsigned long long, hipStream_t);
 void fs_bf16_gemv(const void*, const void*, void*, int, int, int,
                   hipStream_t);
+void fs_decode_attn(const void*, void*, void*, const float*, const float*,
+                    const void*, void*, int, int, int, int, float, int,
+                    hipStream_t);
 void fs_w8_gemv(const void*, const float*, const void*, void*, int, int, int,
                 hipStream_t);
 void fs_vocab_ce_fwd(const void*, const long*, float*, float*, float*,
@@ -372,9 +375,32 @@ static at::Tensor bf16_gemv(at::Tensor w, at::Tensor x) {
   return y;
 }
 
+// Fused decode step: qkv [b, 3*H] bf16, kc/vc [b, nh, L, d] bf16 caches
+// (written in-place at pos), cos/sin [max_len, d] fp32, pos [1] long.
+// Returns ctx [b, H] bf16.
+static at::Tensor decode_attn(at::Tensor qkv, at::Tensor kc, at::Tensor vc,
+                              at::Tensor cos_t, at::Tensor sin_t,
+                              at::Tensor pos, double scale, bool rope) {
+  TORCH_CHECK(qkv.scalar_type() == at::kBFloat16 && qkv.is_contiguous());
+  TORCH_CHECK(kc.is_contiguous() && vc.is_contiguous());
+  TORCH_CHECK(cos_t.scalar_type() == at::kFloat &&
+              sin_t.scalar_type() == at::kFloat);
+  TORCH_CHECK(pos.scalar_type() == at::kLong);
+  const int b = kc.size(0), nh = kc.size(1), L = kc.size(2), d = kc.size(3);
+  TORCH_CHECK(d == 64 || d == 128, "decode_attn: head dim must be 64/128");
+  TORCH_CHECK(qkv.numel() == (long)b * 3 * nh * d, "qkv shape mismatch");
+  auto ctx = at::empty({b, nh * d}, qkv.options());
+  fs_decode_attn(qkv.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                 cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
+                 pos.data_ptr(), ctx.data_ptr(), b, nh, d, L, (float)scale,
+                 rope ? 1 : 0, cur_stream());
+  return ctx;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("w8_gemv", &w8_gemv);
   mod.def("bf16_gemv", &bf16_gemv);
+  mod.def("decode_attn", &decode_attn);
   mod.def("flash_attn_fwd", &flash_attn_fwd);
   mod.def("vocab_ce_fwd", [](at::Tensor logits2d, at::Tensor targets,
                              int64_t vstart, int64_t vend) {

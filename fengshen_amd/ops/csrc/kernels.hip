@@ -963,6 +963,151 @@ extern "C" void fs_bf16_gemv(const void* w, const void* x, void* y, int batch,
 }
 
 // ===========================================================================
+// Fused single-query decode attention (serving path)
+// ===========================================================================
+// One launch per layer replaces ~10 eager ops inside the captured decode
+// graph: rope(q,k) at pos -> KV-cache write -> online-softmax attention
+// over the cache prefix [0..pos] -> context vector.  The eager tail
+// (rotate-half cats, fp32 casts of the cache, index_copy, softmax) was
+// ~40% of the 13B per-token time (profiles/decprof).
+//   qkv    [b, 3*H]     H = nh*D, straight out of the fused qkv GEMV
+//   kc/vc  [b, nh, L, D] static caches, row `pos` written here
+//   cos/sin [max_len, D] fp32 rope tables; pos = device scalar (long)
+//   ctx    [b, H]
+// Grid (b*nh) x 128 threads.  K tiles staged in LDS as fp32 rows padded
+// to D+1 (bank-conflict-free per-thread row reads); V read coalesced by
+// the D lanes in the accumulate phase.
+
+#define FS_DEC_BLK 128
+
+template <int D, bool ROPE>
+__global__ __launch_bounds__(FS_DEC_BLK)
+void decode_attn_kernel(const bf16_t* __restrict__ qkv,
+                        bf16_t* __restrict__ kc, bf16_t* __restrict__ vc,
+                        const float* __restrict__ cos_t,
+                        const float* __restrict__ sin_t,
+                        const long* __restrict__ pos_p,
+                        bf16_t* __restrict__ ctx,
+                        int nh, int cache_len, float scale) {
+  const int tid = threadIdx.x;
+  const int bh = blockIdx.x;  // b*nh + h
+  const int b = bh / nh, h = bh % nh;
+  const int pos = (int)pos_p[0];
+  const int H = nh * D;
+
+  __shared__ float qraw[D], kraw[D];
+  __shared__ float qs[D], ks_new[D];
+  __shared__ float red[16];
+  __shared__ float pbuf[FS_DEC_BLK];
+  __shared__ float ktile[FS_DEC_BLK][D + 1];
+
+  // ---- phase 0: rope q/k at pos, write the cache row -------------------
+  float v_new = 0.f;
+  if (tid < D) {
+    qraw[tid] = to_f32<bf16_t>(qkv[(long)b * 3 * H + h * D + tid]);
+    kraw[tid] = to_f32<bf16_t>(qkv[(long)b * 3 * H + H + h * D + tid]);
+    v_new = to_f32<bf16_t>(qkv[(long)b * 3 * H + 2 * H + h * D + tid]);
+  }
+  __syncthreads();
+  if (tid < D) {
+    float q = qraw[tid], k = kraw[tid];
+    if (ROPE) {
+      float c = cos_t[(long)pos * D + tid];
+      float s = sin_t[(long)pos * D + tid];
+      float qr = (tid < D / 2) ? -qraw[tid + D / 2] : qraw[tid - D / 2];
+      float kr = (tid < D / 2) ? -kraw[tid + D / 2] : kraw[tid - D / 2];
+      q = q * c + qr * s;
+      k = k * c + kr * s;
+    }
+    qs[tid] = q * scale;
+    ks_new[tid] = k;
+    const long coff = (((long)b * nh + h) * cache_len + pos) * D + tid;
+    kc[coff] = from_f32<bf16_t>(k);
+    vc[coff] = from_f32<bf16_t>(v_new);
+  }
+  // make the cache row visible to this block's phase-2 V reads
+  __threadfence();
+  __syncthreads();
+
+  // ---- phases 1+2: online softmax over chunks of 128 positions ---------
+  const bf16_t* kbase = kc + ((long)b * nh + h) * (long)cache_len * D;
+  const bf16_t* vbase = vc + ((long)b * nh + h) * (long)cache_len * D;
+  float m_run = -INFINITY, l_run = 0.f, oacc = 0.f;
+  for (int p0 = 0; p0 <= pos; p0 += FS_DEC_BLK) {
+    const int nrow = min(FS_DEC_BLK, pos + 1 - p0);
+    // stage K rows [p0, p0+nrow) -> fp32 LDS (coalesced 16B global loads)
+    for (int idx = tid * 8; idx < nrow * D; idx += FS_DEC_BLK * 8) {
+      const int r = idx / D, c = idx % D;  // D % 8 == 0: no row straddle
+      float kv[8];
+      load8<bf16_t>(kbase + (long)(p0 + r) * D + c, kv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ktile[r][c + j] = kv[j];
+    }
+    __syncthreads();
+    // the row written this step comes from LDS (no coherence question)
+    if (pos >= p0 && pos < p0 + FS_DEC_BLK && tid < D)
+      ktile[pos - p0][tid] = ks_new[tid];
+    __syncthreads();
+    // scores: thread t owns position p0+t
+    float s_t = -INFINITY;
+    if (tid < nrow) {
+      float acc = 0.f;
+#pragma unroll
+      for (int j = 0; j < D; j += 8) {
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          acc += qs[j + jj] * ktile[tid][j + jj];
+      }
+      s_t = acc;  // scale folded into qs
+    }
+    const float m_chunk = block_reduce_max(s_t, red);
+    const float m_new = fmaxf(m_run, m_chunk);
+    const float alpha = __expf(m_run - m_new);
+    const float pr = (tid < nrow) ? __expf(s_t - m_new) : 0.f;
+    pbuf[tid] = pr;
+    const float psum = block_reduce_sum(pr, red);  // also syncs pbuf
+    l_run = l_run * alpha + psum;
+    m_run = m_new;
+    // context accumulate: thread tid < D owns ctx element tid
+    if (tid < D) {
+      float acc = oacc * alpha;
+      for (int t = 0; t < nrow; ++t)
+        acc += pbuf[t] * to_f32<bf16_t>(vbase[(long)(p0 + t) * D + tid]);
+      oacc = acc;
+    }
+    __syncthreads();  // pbuf/ktile reuse in next chunk
+  }
+  if (tid < D)
+    ctx[(long)b * H + h * D + tid] = from_f32<bf16_t>(oacc / l_run);
+}
+
+extern "C" void fs_decode_attn(const void* qkv, void* kc, void* vc,
+                               const float* cos_t, const float* sin_t,
+                               const void* pos, void* ctx, int b, int nh,
+                               int d, int cache_len, float scale, int rope,
+                               hipStream_t s) {
+#define FS_DEC_CASE(D)                                                       \
+  if (d == D) {                                                              \
+    if (rope)                                                                \
+      hipLaunchKernelGGL((decode_attn_kernel<D, true>), dim3(b * nh),        \
+                         dim3(FS_DEC_BLK), 0, s, (const bf16_t*)qkv,         \
+                         (bf16_t*)kc, (bf16_t*)vc, cos_t, sin_t,             \
+                         (const long*)pos, (bf16_t*)ctx, nh, cache_len,      \
+                         scale);                                             \
+    else                                                                     \
+      hipLaunchKernelGGL((decode_attn_kernel<D, false>), dim3(b * nh),       \
+                         dim3(FS_DEC_BLK), 0, s, (const bf16_t*)qkv,         \
+                         (bf16_t*)kc, (bf16_t*)vc, cos_t, sin_t,             \
+                         (const long*)pos, (bf16_t*)ctx, nh, cache_len,      \
+                         scale);                                             \
+    return;                                                                  \
+  }
+  FS_DEC_CASE(64)
+  FS_DEC_CASE(128)
+#undef FS_DEC_CASE
+}
+
+// ===========================================================================
 // Fused vocab cross-entropy (TP-shard-aware)
 // ===========================================================================
 // Replaces the composite vocab_parallel_cross_entropy hot path, which

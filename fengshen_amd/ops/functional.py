@@ -321,7 +321,10 @@ def linear(x: torch.Tensor, weight: torch.Tensor,
             and weight.shape[1] % 16 == 0
             and x.shape[-1] == weight.shape[1]):
         rows = x.numel() // x.shape[-1]
-        if 1 <= rows <= 8 and use_hip(x):
+        # rows == 1 only: the B>1 GEMV re-reads x per weight vector and
+        # measured SLOWER than hipBLASLt's M=8 GEMM (17.2 vs 11.0 ms/tok
+        # at batch 8) — multi-row decode stays on the library path.
+        if rows == 1 and use_hip(x):
             y = get_ext().bf16_gemv(weight.contiguous(), x.contiguous())
             return y.view(*x.shape[:-1], weight.shape[0])
     return torch.nn.functional.linear(x, weight, bias)

@@ -97,6 +97,16 @@ class GraphedDecoder:
         if do_sample:
             self.gumbel = torch.zeros(max_new_tokens, batch,
                                       cfg.vocab_size, device=dev)
+        # fused rope+cache+attention kernel (one launch per layer instead
+        # of ~10 eager ops); requires bf16 and head dim 64/128
+        from fengshen_amd.ops import get_ext
+        import os
+        self._ext = get_ext() if dev.type == "cuda" else None
+        self._fused_attn = (
+            self._ext is not None and hd in (64, 128)
+            and dt == torch.bfloat16
+            and os.environ.get("FENGSHEN_AMD_FORCE_EAGER") != "1"
+            and os.environ.get("FENGSHEN_DECODE_FUSED", "1") == "1")
         self._graph: Optional[torch.cuda.CUDAGraph] = None
 
     # ------------------------------------------------------------------
@@ -111,18 +121,37 @@ class GraphedDecoder:
         b, nh, hd = self.batch, self.nh, self.hd
         pos = self.pos
         h = m.embed_tokens(self.tok)  # [b, 1, H]
-        c = self.cos.index_select(0, pos).to(self.dt)  # [1, hd]
-        s = self.sin.index_select(0, pos).to(self.dt)
-        # additive mask over the static window: visible iff idx <= pos
-        amask = torch.where(self.ar <= pos,
-                            torch.zeros((), device=self.dev),
-                            torch.full((), float("-inf"), device=self.dev))
+        if not self._fused_attn:
+            c = self.cos.index_select(0, pos).to(self.dt)  # [1, hd]
+            s = self.sin.index_select(0, pos).to(self.dt)
+            # additive mask over the static window: visible iff idx <= pos
+            amask = torch.where(self.ar <= pos,
+                                torch.zeros((), device=self.dev),
+                                torch.full((), float("-inf"),
+                                           device=self.dev))
         for i, layer in enumerate(m.layers):
             res = h
             x = layer.input_norm(h)
             qkv = layer.attention.qkv_proj(x)
             if isinstance(qkv, tuple):
                 qkv = qkv[0]
+            if self._fused_attn:
+                ctx = self._ext.decode_attn(
+                    qkv.view(b, 3 * nh * hd), self.k_cache[i],
+                    self.v_cache[i], self.cos, self.sin, pos,
+                    self.scale, bool(layer.attention.rotary))
+                ctx = ctx.view(b, 1, nh * hd)
+                out = layer.attention.out_proj(ctx)
+                if isinstance(out, tuple):
+                    out = out[0]
+                h = res + out
+                res = h
+                x = layer.post_attention_norm(h)
+                mlp = layer.mlp(x)
+                if isinstance(mlp, tuple):
+                    mlp = mlp[0]
+                h = res + mlp
+                continue
             q, k, v = qkv.chunk(3, dim=-1)
             q = q.view(b, 1, nh, hd).transpose(1, 2)  # [b, nh, 1, hd]
             k = k.view(b, 1, nh, hd).transpose(1, 2)
