@@ -20,7 +20,7 @@ table) and PERF_ROADMAP item 5 (hipGraph capture of the decode loop).
 from __future__ import annotations
 
 import math
-from typing import List, Optional
+from typing import Optional
 
 import torch
 
