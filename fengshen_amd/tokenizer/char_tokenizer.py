@@ -80,6 +80,30 @@ class SimpleCharTokenizer:
         return {"input_ids": ids, "attention_mask": attn,
                 "token_type_ids": tok_type}
 
+    def __call__(self, text, padding=True, truncation=True,
+                 max_length: int = 512, return_tensors=None, **_kw):
+        """HF-style batch encode: str or list of str; padding to the
+        longest sequence (or max_length when padding='max_length')."""
+        texts = [text] if isinstance(text, str) else list(text)
+        encs = []
+        for t in texts:
+            ids = self.encode(t)
+            if truncation and len(ids) > max_length:
+                ids = ids[:max_length - 1] + [self.sep_token_id]
+            encs.append(ids)
+        tgt = (max_length if padding == "max_length"
+               else max(len(e) for e in encs))
+        input_ids, attn = [], []
+        for e in encs:
+            pad = tgt - len(e) if padding else 0
+            input_ids.append(e + [self.pad_token_id] * pad)
+            attn.append([1] * len(e) + [0] * pad)
+        out = {"input_ids": input_ids, "attention_mask": attn}
+        if return_tensors == "pt":
+            import torch
+            out = {k: torch.tensor(v) for k, v in out.items()}
+        return out
+
     def decode(self, ids, skip_special_tokens: bool = True) -> str:
         out = []
         for i in ids:

@@ -179,3 +179,40 @@ def test_graphed_decoder_sampling_cpu_pick():
     dec.max_new = 4
     dec.gumbel = torch.zeros(4, 3, 30).exponential_().log_().neg_()
     assert torch.equal(dec._pick(logits, step), greedy)
+
+
+def test_text_generation_pipeline_cpu():
+    """text_generation pipeline: greedy + sampled generation on a tiny
+    llama (CPU falls back to HF generate), serving /generate wiring, and
+    the causal-LM collator."""
+    from fengshen_amd.models.llama.configuration_llama import (
+        llama_tiny_config)
+    from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+    from fengshen_amd.pipelines import text_generation
+    from tests.test_data import FakeTokenizer
+
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(llama_tiny_config())
+    tok = FakeTokenizer()
+    pipe = text_generation.Pipeline(model=model, tokenizer=tok,
+                                    max_new_tokens=8)
+    out = pipe.generate("你好 世界", max_new_tokens=4)
+    assert isinstance(out, str)
+    sampled = pipe("你好", do_sample=True, top_k=5, temperature=0.7,
+                   max_new_tokens=4)
+    assert isinstance(sampled, str)
+    # collator
+    coll = text_generation._LMCollator(tok, max_length=16)
+    batch = coll([{"text": "你好 世界"}, {"text": "好"}])
+    assert batch["labels"].shape == batch["input_ids"].shape
+    assert (batch["labels"][batch["attention_mask"] == 0] == -100).all()
+    # serving endpoint
+    from fengshen_amd.serving.main import APIConfig, build_app
+    from fastapi.testclient import TestClient
+    app = build_app(APIConfig(pipeline_type="text_generation"),
+                    pipeline=pipe)
+    c = TestClient(app)
+    r = c.post("/generate", json={"input_text": "你好",
+                                  "max_new_tokens": 4,
+                                  "do_sample": True, "top_k": 5})
+    assert r.status_code == 200 and isinstance(r.json()["result"], str)
