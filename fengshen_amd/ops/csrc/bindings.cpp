@@ -56,6 +56,8 @@ void fs_flash_attn_bwd(const void*, const void*, const void*, const void*,
                        float, unsigned long long, hipStream_t);
 void fs_bf16_gemv(const void*, const void*, void*, int, int, int,
                   hipStream_t);
+void fs_add_rms_norm(const void*, const void*, const void*, void*, void*,
+                     int, int, float, hipStream_t);
 void fs_decode_attn(const void*, void*, void*, const float*, const float*,
                     const void*, void*, int, int, int, int, float, int,
                     hipStream_t);
@@ -397,10 +399,27 @@ static at::Tensor decode_attn(at::Tensor qkv, at::Tensor kc, at::Tensor vc,
   return ctx;
 }
 
+// a,b [rows, H] bf16 -> (sum=a+b, y=rmsnorm(sum)*w), inference-only
+static std::vector<at::Tensor> add_rms_norm(at::Tensor a, at::Tensor b,
+                                            at::Tensor w, double eps) {
+  TORCH_CHECK(a.scalar_type() == at::kBFloat16 && a.is_contiguous() &&
+              b.is_contiguous() && w.is_contiguous());
+  const int H = a.size(-1);
+  TORCH_CHECK(H % 8 == 0 && w.numel() == H && b.sizes() == a.sizes());
+  const int rows = a.numel() / H;
+  auto sum_out = at::empty_like(a);
+  auto y = at::empty_like(a);
+  fs_add_rms_norm(a.data_ptr(), b.data_ptr(), w.data_ptr(),
+                  sum_out.data_ptr(), y.data_ptr(), rows, H, (float)eps,
+                  cur_stream());
+  return {sum_out, y};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("w8_gemv", &w8_gemv);
   mod.def("bf16_gemv", &bf16_gemv);
   mod.def("decode_attn", &decode_attn);
+  mod.def("add_rms_norm", &add_rms_norm);
   mod.def("flash_attn_fwd", &flash_attn_fwd);
   mod.def("vocab_ce_fwd", [](at::Tensor logits2d, at::Tensor targets,
                              int64_t vstart, int64_t vend) {

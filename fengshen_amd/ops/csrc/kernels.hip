@@ -898,6 +898,59 @@ extern "C" void fs_w8_gemv(const void* q8, const float* scale, const void* x,
 
 
 // ===========================================================================
+// Fused residual-add + RMSNorm (decode/inference): sum = a + b written
+// out for the next residual, y = rmsnorm(sum) * w.  One launch instead
+// of add + norm (the captured 13B decode graph carries 80 such pairs
+// per token).  Inference-only: no invrms saved.
+// ===========================================================================
+__global__ void add_rms_norm_kernel(const bf16_t* __restrict__ a,
+                                    const bf16_t* __restrict__ b,
+                                    const bf16_t* __restrict__ w,
+                                    bf16_t* __restrict__ sum_out,
+                                    bf16_t* __restrict__ y,
+                                    int rows, int H, float eps) {
+  extern __shared__ float lds[];  // [H]
+  __shared__ float red[32];
+  for (int row = blockIdx.x; row < rows; row += gridDim.x) {
+    const bf16_t* ar = a + (long)row * H;
+    const bf16_t* br = b + (long)row * H;
+    float ss = 0.f;
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+      float va[8], vb[8], o[8];
+      load8<bf16_t>(ar + c, va);
+      load8<bf16_t>(br + c, vb);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        o[i] = va[i] + vb[i];
+        lds[c + i] = o[i];
+        ss += o[i] * o[i];
+      }
+      store8<bf16_t>(sum_out + (long)row * H + c, o);
+    }
+    ss = block_reduce_sum(ss, red);
+    const float ir = rsqrtf(ss / H + eps);
+    for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+      float wv[8], o[8];
+      load8<bf16_t>(w + c, wv);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) o[i] = lds[c + i] * ir * wv[i];
+      store8<bf16_t>(y + (long)row * H + c, o);
+    }
+    __syncthreads();
+  }
+}
+
+extern "C" void fs_add_rms_norm(const void* a, const void* b, const void* w,
+                                void* sum_out, void* y, int rows, int H,
+                                float eps, hipStream_t s) {
+  int grid = rows < FS_MAX_BLOCKS ? rows : FS_MAX_BLOCKS;
+  size_t lds = (size_t)H * 4;
+  hipLaunchKernelGGL(add_rms_norm_kernel, dim3(grid), dim3(256), lds, s,
+                     (const bf16_t*)a, (const bf16_t*)b, (const bf16_t*)w,
+                     (bf16_t*)sum_out, (bf16_t*)y, rows, H, eps);
+}
+
+// ===========================================================================
 // bf16 GEMV: y[b, o] = sum_i w[o, i] * x[b, i]   (bf16 decode path)
 // hipBLASLt's M=1 GEMM leaves most of HBM bandwidth on the table on
 // gfx950 (tile quantization: tiny M forces skinny tiles).  Decode is a

@@ -115,6 +115,17 @@ class GraphedDecoder:
         x1, x2 = x.chunk(2, dim=-1)
         return torch.cat((-x2, x1), dim=-1)
 
+    def _add_norm(self, a, b_, norm):
+        """(a + b_, norm(a + b_)) — fused into one kernel when norm is an
+        RMSNorm on the HIP path (the 13B decode graph has 81 such pairs
+        per token)."""
+        if self._fused_attn and type(norm).__name__ == "RMSNorm":
+            s, y = self._ext.add_rms_norm(a.contiguous(), b_.contiguous(),
+                                          norm.weight, norm.eps)
+            return s.view_as(a), y.view_as(a)
+        s = a + b_
+        return s, norm(s)
+
     def _step(self):
         """Graph-safe single-token decode; advances tok/pos/step_i."""
         m = self.model.model
@@ -129,29 +140,47 @@ class GraphedDecoder:
                                 torch.zeros((), device=self.dev),
                                 torch.full((), float("-inf"),
                                            device=self.dev))
+        if self._fused_attn:
+            # residual adds ride the next norm's kernel (add_rms_norm),
+            # including across layer boundaries and into the final norm
+            pend = None  # (residual, branch) awaiting add+norm
+            for i, layer in enumerate(m.layers):
+                if pend is None:
+                    x = layer.input_norm(h)
+                else:
+                    h, x = self._add_norm(pend[0], pend[1],
+                                          layer.input_norm)
+                qkv = layer.attention.qkv_proj(x)
+                if isinstance(qkv, tuple):
+                    qkv = qkv[0]
+                ctx = self._ext.decode_attn(
+                    qkv.view(b, 3 * nh * hd), self.k_cache[i],
+                    self.v_cache[i], self.cos, self.sin, pos,
+                    self.scale, bool(layer.attention.rotary))
+                out = layer.attention.out_proj(ctx.view(b, 1, nh * hd))
+                if isinstance(out, tuple):
+                    out = out[0]
+                h, x = self._add_norm(h, out, layer.post_attention_norm)
+                mlp = layer.mlp(x)
+                if isinstance(mlp, tuple):
+                    mlp = mlp[0]
+                pend = (h, mlp)
+            _, h = self._add_norm(pend[0], pend[1], m.norm)
+            logits = self.model.lm_head(h)
+            if isinstance(logits, tuple):
+                logits = logits[0]
+            nxt = self._pick(logits[:, -1, :], self.step_i)
+            self.out_tokens.index_copy_(1, self.step_i, nxt)
+            self.tok.copy_(nxt)
+            self.pos.add_(1)
+            self.step_i.add_(1)
+            return
         for i, layer in enumerate(m.layers):
             res = h
             x = layer.input_norm(h)
             qkv = layer.attention.qkv_proj(x)
             if isinstance(qkv, tuple):
                 qkv = qkv[0]
-            if self._fused_attn:
-                ctx = self._ext.decode_attn(
-                    qkv.view(b, 3 * nh * hd), self.k_cache[i],
-                    self.v_cache[i], self.cos, self.sin, pos,
-                    self.scale, bool(layer.attention.rotary))
-                ctx = ctx.view(b, 1, nh * hd)
-                out = layer.attention.out_proj(ctx)
-                if isinstance(out, tuple):
-                    out = out[0]
-                h = res + out
-                res = h
-                x = layer.post_attention_norm(h)
-                mlp = layer.mlp(x)
-                if isinstance(mlp, tuple):
-                    mlp = mlp[0]
-                h = res + mlp
-                continue
             q, k, v = qkv.chunk(3, dim=-1)
             q = q.view(b, 1, nh, hd).transpose(1, 2)  # [b, nh, 1, hd]
             k = k.view(b, 1, nh, hd).transpose(1, 2)

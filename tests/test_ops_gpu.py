@@ -596,3 +596,22 @@ def test_decode_attn_matches_eager():
         ref = torch.einsum("bhl,bhld->bhd", att.softmax(-1), vr)
         err = (ctx.view(b, nh, d).float() - ref).abs().max()
         assert err < 0.03, (nh, d, L, pos_i, float(err))
+
+
+def test_add_rms_norm_matches_eager():
+    """Fused residual-add + RMSNorm (decode) vs fp32 eager."""
+    import torch
+    from fengshen_amd.ops import get_ext
+    ext = get_ext()
+    assert ext is not None
+    torch.manual_seed(0)
+    for (rows, H) in [(1, 5120), (8, 5120), (3, 256)]:
+        a = (torch.randn(rows, H, device="cuda") * 0.7).bfloat16()
+        b = (torch.randn(rows, H, device="cuda") * 0.7).bfloat16()
+        w = (torch.randn(H, device="cuda") * 0.1 + 1).bfloat16()
+        s, y = ext.add_rms_norm(a, b, w, 1e-6)
+        sf = a.float() + b.float()
+        ref = sf * torch.rsqrt(sf.pow(2).mean(-1, keepdim=True) + 1e-6) \
+            * w.float()
+        assert (s.float() - sf).abs().max() < 0.02
+        assert (y.float() - ref).abs().max() < 0.05, (rows, H)
