@@ -216,3 +216,15 @@ def test_text_generation_pipeline_cpu():
                                   "max_new_tokens": 4,
                                   "do_sample": True, "top_k": 5})
     assert r.status_code == 200 and isinstance(r.json()["result"], str)
+
+
+def test_graphed_decoder_add_norm_eager_fallback():
+    """_add_norm without the HIP ext: plain add + norm, any norm type."""
+    from fengshen_amd.serving.graphed_decode import GraphedDecoder
+    dec = GraphedDecoder.__new__(GraphedDecoder)
+    dec._fused_attn = False
+    norm = torch.nn.LayerNorm(16)
+    a, b = torch.randn(2, 1, 16), torch.randn(2, 1, 16)
+    s, y = dec._add_norm(a, b, norm)
+    assert torch.allclose(s, a + b)
+    assert torch.allclose(y, norm(a + b))
