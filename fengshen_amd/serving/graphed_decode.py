@@ -1,4 +1,4 @@
-"""hipGraph-captured greedy decode for LLaMA serving.
+"""hipGraph-captured decode for LLaMA serving (greedy or sampled).
 
 Eager decode of a 13B model launches ~400 kernels per token (40 layers x
 ~10 ops); at ~8 us a launch that is several ms of pure launch overhead on
@@ -6,13 +6,20 @@ a path that is otherwise weight-bandwidth-bound.  This engine captures ONE
 self-advancing decode step into a hipGraph (torch.cuda.CUDAGraph is
 hipGraph on ROCm) and replays it per token:
 
-- static KV cache [b, heads, max_len, hd] per layer, in-place index_copy_
-  at a device position tensor;
-- rope cos/sin gathered by the position tensor (no host sync);
-- attention over the full static window with an additive (pos-driven) mask;
-- greedy argmax feeds the token buffer consumed by the NEXT replay, so a
-  whole max_new_tokens decode is just N graph replays with zero host
-  round-trips.
+- static KV cache [b, heads, max_len, hd] per layer, position driven by a
+  device tensor (no host sync anywhere in the loop);
+- inside the graph the hot path is hand-written HIP (kernels.hip):
+  bf16_gemv (weights streamed at 5.7-6.8 TB/s vs hipBLASLt's ~2.5-3 at
+  M=1), decode_attn (rope + cache write + online-softmax attention in one
+  launch per layer) and add_rms_norm (residual add fused into the norm);
+  eager torch fallback when the extension/dtype/head-dim doesn't fit;
+- greedy argmax — or gumbel-max top-k/top-p sampling over a pre-generated
+  seeded noise buffer indexed by the step tensor — feeds the token buffer
+  consumed by the NEXT replay, so a whole max_new_tokens decode is just N
+  graph replays with zero host round-trips.
+
+Measured (13B, 1x MI355X): 5.92 ms/token bf16, 5.06 int8, 0.99
+ms/seq/token at batch 8 (profiles/decode_kernels_r2.md).
 
 Ref context: examples/ziya_inference (the reference's only throughput
 table) and PERF_ROADMAP item 5 (hipGraph capture of the decode loop).
