@@ -235,3 +235,42 @@ def test_scalenorm_and_init_zoo():
     assert abs(w.std().item() - math.sqrt(2 / (5 * 4096))) < 1e-3
     wang_init(4096, 24)(w)
     assert abs(w.std().item() - 2 / (24 * math.sqrt(4096))) < 1e-3
+
+
+def test_alibi_softembedding_gmlp():
+    """Layer-library extras: ALiBi slopes/bias shape, SoftEmbedding
+    prompt prepend, gMLP block identity-at-init gating."""
+    import math
+    import torch
+    from fengshen_amd.models.layers import AliBi, GMLPBlock, SoftEmbedding
+
+    torch.manual_seed(0)
+    # ALiBi: 8 heads -> geometric slopes starting at 2^-1 ... ref formula
+    ab = AliBi(num_heads=8)
+    assert ab.slopes.shape == (8,)
+    assert abs(ab.slopes[0].item() - 2 ** -1) < 1e-6
+    assert abs(ab.slopes[-1].item() - 2 ** -8) < 1e-6
+    bias = ab(4, 6, torch.device("cpu"), torch.float32)
+    assert bias.shape == (8, 4, 6)
+    # bias is slope * (j - i): zero on the diagonal band start
+    assert torch.allclose(bias[:, 0, 0], torch.zeros(8))
+    # non-power-of-two head count path
+    assert len(AliBi._slopes(12)) == 12
+    # TP-aware slicing
+    ab1 = AliBi(num_heads=8, mp_size=2, mp_rank=1)
+    assert torch.allclose(ab1.slopes, ab.slopes[4:])
+
+    # SoftEmbedding
+    wte = torch.nn.Embedding(50, 16)
+    se = SoftEmbedding(wte, n_tokens=3)
+    out = se(torch.randint(0, 50, (2, 5)))
+    assert out.shape == (2, 8, 16)
+    assert torch.allclose(out[0, :3], se.soft_prompt)
+
+    # gMLP: zero-weight/one-bias SGU projection gates with norm(gate)*1
+    blk = GMLPBlock(dim=16, dim_ff=32, seq_len=6)
+    x = torch.randn(2, 6, 16)
+    y = blk(x)
+    assert y.shape == x.shape
+    y.sum().backward()
+    assert blk.sgu.proj.weight.grad is not None
