@@ -135,3 +135,20 @@ def test_tf_bert_name_mapping(tmp_path):
     assert "bert.embeddings.word_embeddings.weight" in sd
     assert "cls.predictions.bias" in sd
     assert not any("adam_m" in k for k in sd)
+
+
+def test_ner_bio_bios_decode():
+    """BIO/BIOS entity decode (ref metric/utils_ner.py behavior)."""
+    from fengshen_amd.metric.utils_ner import get_entities
+
+    bios = ["O", "B-PER", "I-PER", "O", "S-LOC", "B-ORG", "I-ORG"]
+    assert get_entities(bios, "bios") == [("PER", 1, 2), ("LOC", 4, 4),
+                                          ("ORG", 5, 6)]
+    bio = ["B-PER", "I-PER", "O", "B-LOC", "B-ORG", "I-ORG"]
+    assert get_entities(bio, "bio") == [("PER", 0, 1), ("LOC", 3, 3),
+                                        ("ORG", 4, 5)]
+    # I- with mismatched type does not extend the chunk
+    assert get_entities(["B-PER", "I-LOC", "O"], "bio") == [("PER", 0, 0)]
+    # trailing entity at sequence end is flushed
+    assert get_entities(["O", "B-LOC"], "bio") == [("LOC", 1, 1)]
+    assert get_entities([], "bios") == []
