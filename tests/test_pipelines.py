@@ -228,3 +228,16 @@ def test_graphed_decoder_add_norm_eager_fallback():
     s, y = dec._add_norm(a, b, norm)
     assert torch.allclose(s, a + b)
     assert torch.allclose(y, norm(a + b))
+
+
+def test_all_pipeline_modules_importable():
+    """Every pipeline module the CLI can resolve exposes Pipeline with
+    the argparse hook."""
+    from importlib import import_module
+    for task in ["text_classification", "sequence_tagging",
+                 "information_extraction", "multiplechoice", "tcbert",
+                 "text_generation"]:
+        mod = import_module(f"fengshen_amd.pipelines.{task}")
+        assert hasattr(mod, "Pipeline"), task
+        assert callable(getattr(mod.Pipeline, "add_pipeline_specific_args",
+                                None)), task
