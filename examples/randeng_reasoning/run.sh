@@ -1,0 +1,5 @@
+#!/bin/bash
+# Randeng reasoning generation demo (tiny random-init unless --model_path).
+set -e
+cd "$(dirname "$0")/../.."
+python examples/randeng_reasoning/reasoning_generate.py "$@"

@@ -26,6 +26,9 @@ def _run(rel, *extra):
     ("examples/transfo_xl_denoise/generate.py", ("--seq_len", "28")),
     ("examples/fastdemo/qa_demo.py", ("--smoke",)),
     ("examples/disco_project/clip_guided_generate.py", ("--steps", "4")),
+    ("examples/randeng_reasoning/reasoning_generate.py",
+     ("--max_out_seq", "10")),
+    ("examples/longformer/longformer_mlm.py", ("--seq_len", "128")),
 ])
 def test_example_smokes(rel, extra):
     _run(rel, *extra)
