@@ -152,3 +152,40 @@ def test_ner_bio_bios_decode():
     # trailing entity at sequence end is flushed
     assert get_entities(["O", "B-LOC"], "bio") == [("LOC", 1, 1)]
     assert get_entities([], "bios") == []
+
+
+def test_scheduler_registry_formulas():
+    """Scheduler registry (ref model_utils.py:101-254): warmup ramp,
+    polynomial floor at lr_end, inverse-sqrt decay, direct constant."""
+    import math
+    import torch
+    from fengshen_amd.models.model_utils import get_scheduler
+
+    def lrs(name, n, **kw):
+        p = torch.nn.Parameter(torch.zeros(1))
+        opt = torch.optim.SGD([p], lr=1e-4)
+        sch = get_scheduler(name, opt, **kw)
+        out = []
+        for _ in range(n):
+            out.append(opt.param_groups[0]["lr"])
+            opt.step()
+            sch.step()
+        return out
+
+    poly = lrs("polynomial", 30, num_warmup_steps=5, num_training_steps=20,
+               lr_end=1e-6, lr_init=1e-4)
+    assert poly[0] == 0.0 and abs(poly[5] - 1e-4) < 1e-9   # ramp to peak
+    assert abs(poly[-1] - 1e-6) < 1e-9                      # floor at lr_end
+    assert all(a >= b - 1e-12 for a, b in zip(poly[5:], poly[6:]))
+
+    inv = lrs("inverse_sqrt", 25, num_warmup_steps=4)
+    assert abs(inv[4] - 1e-4) < 1e-9
+    assert abs(inv[16] - 1e-4 * math.sqrt(4 / 16)) < 1e-9   # ~ 1/sqrt(t)
+
+    direct = lrs("direct", 10, num_warmup_steps=4)
+    assert abs(direct[2] - 1e-4 * 2 / 4) < 1e-9
+    assert all(abs(v - 1e-4) < 1e-9 for v in direct[4:])
+
+    cos = lrs("cosine", 21, num_warmup_steps=0, num_training_steps=20)
+    assert abs(cos[10] - 5e-5) < 1e-7                       # half way
+    assert cos[-1] < 1e-5
