@@ -173,14 +173,7 @@ class GraphedDecoder:
                     mlp = mlp[0]
                 pend = (h, mlp)
             _, h = self._add_norm(pend[0], pend[1], m.norm)
-            logits = self.model.lm_head(h)
-            if isinstance(logits, tuple):
-                logits = logits[0]
-            nxt = self._pick(logits[:, -1, :], self.step_i)
-            self.out_tokens.index_copy_(1, self.step_i, nxt)
-            self.tok.copy_(nxt)
-            self.pos.add_(1)
-            self.step_i.add_(1)
+            self._emit(h)
             return
         for i, layer in enumerate(m.layers):
             res = h
@@ -213,6 +206,10 @@ class GraphedDecoder:
                 mlp = mlp[0]
             h = res + mlp
         h = m.norm(h)
+        self._emit(h)
+
+    def _emit(self, h):
+        """lm_head -> pick -> record token and advance pos/step_i."""
         logits = self.model.lm_head(h)
         if isinstance(logits, tuple):
             logits = logits[0]
