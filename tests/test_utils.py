@@ -189,3 +189,24 @@ def test_scheduler_registry_formulas():
     cos = lrs("cosine", 21, num_warmup_steps=0, num_training_steps=20)
     assert abs(cos[10] - 5e-5) < 1e-7                       # half way
     assert cos[-1] < 1e-5
+
+
+def test_rouge_score_hand_values():
+    """RougeScore against hand-computed rouge-1/2/L on token strings."""
+    from fengshen_amd.metric.rouge import RougeScore
+
+    rs = RougeScore()
+    rs.update(["a b c d"], ["a b e d"])
+    out = rs.compute()
+    # unigrams: overlap {a,b,d} = 3 of 4 -> P=R=F=0.75
+    assert abs(out["rouge1_fmeasure"] - 0.75) < 1e-6
+    # bigrams: pred {ab,bc,cd}, ref {ab,be,ed} -> overlap {ab} = 1/3
+    assert abs(out["rouge2_precision"] - 1 / 3) < 1e-6
+    # LCS("abcd","abed") = "abd" (3) -> F = 0.75
+    assert abs(out["rougeL_fmeasure"] - 0.75) < 1e-6
+    # perfect match accumulates with the first sample
+    rs.update(["x y"], ["x y"])
+    out2 = rs.compute()
+    assert abs(out2["rouge1_fmeasure"] - (0.75 + 1.0) / 2) < 1e-6
+    rs.reset()
+    assert rs.compute()["rouge1_fmeasure"] == 0.0
