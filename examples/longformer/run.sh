@@ -2,4 +2,4 @@
 # Longformer MLM demo at 1k context (tiny random-init unless --model_path).
 set -e
 cd "$(dirname "$0")/../.."
-python examples/longformer/longformer_mlm.py "$@"
+exec python examples/longformer/longformer_mlm.py "$@"

@@ -2,4 +2,4 @@
 # Randeng reasoning generation demo (tiny random-init unless --model_path).
 set -e
 cd "$(dirname "$0")/../.."
-python examples/randeng_reasoning/reasoning_generate.py "$@"
+exec python examples/randeng_reasoning/reasoning_generate.py "$@"
