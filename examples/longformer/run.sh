@@ -1,5 +1,5 @@
 #!/bin/bash
 # Longformer MLM demo at 1k context (tiny random-init unless --model_path).
-set -e
-cd "$(dirname "$0")/../.."
-exec python examples/longformer/longformer_mlm.py "$@"
+set -euo pipefail
+cd "$(dirname "$0")"
+exec python longformer_mlm.py "$@"

@@ -1,5 +1,5 @@
 #!/bin/bash
 # Randeng reasoning generation demo (tiny random-init unless --model_path).
-set -e
-cd "$(dirname "$0")/../.."
-exec python examples/randeng_reasoning/reasoning_generate.py "$@"
+set -euo pipefail
+cd "$(dirname "$0")"
+exec python reasoning_generate.py "$@"
