@@ -358,3 +358,25 @@ def test_ubert_offset_mapping():
     mapping = OffsetMapping().rematch("Hello ab", list("hello ab"))
     assert mapping[0] == [0]
     assert mapping[6] == [6]
+
+
+def test_megatron_t5_idea_diffs():
+    """The @IDEA-modified T5 mechanics (ref modeling_megatron_t5.py):
+    biased dense layers, standard LayerNorm (with bias), absolute
+    position embeddings instead of relative bias."""
+    from fengshen_amd.models.t5.modeling_t5 import (
+        T5ForConditionalGeneration, t5_tiny_config)
+    m = T5ForConditionalGeneration(t5_tiny_config())
+    core = m.t5 if hasattr(m, "t5") else m
+    assert hasattr(core, "enc_pos") and hasattr(core, "dec_pos")
+    # biased dense layers (column/row-parallel linears carry .bias here)
+    biased = [mod for mod in m.modules()
+              if hasattr(mod, "weight") and getattr(mod, "bias", None)
+              is not None and getattr(mod, "weight", None) is not None
+              and mod.weight.dim() == 2]
+    assert biased, "IDEA T5 uses bias=True dense layers"
+    assert not any("relative_attention_bias" in n
+                   for n, _ in m.named_parameters())
+    # standard LayerNorm (has a bias term; T5LayerNorm/RMS does not)
+    assert any("ln" in n or "norm" in n for n, _ in m.named_parameters()
+               if n.endswith(".bias"))
