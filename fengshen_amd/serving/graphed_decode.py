@@ -282,8 +282,11 @@ class GraphedDecoder:
             # torch.manual_seed for reproducible sampling.
             self.gumbel.exponential_().log_().neg_()
         if self._graph is None:
-            # capture against scratch state, then restore via real prefill
-            self.pos.fill_(prompt_ids.shape[1])
+            # capture against scratch state, then restore via real prefill.
+            # The 2 warmup steps + capture step advance pos by 3 total, so
+            # start low enough that every scratch write stays inside the
+            # static window even for tiny max_new_tokens.
+            self.pos.fill_(min(prompt_ids.shape[1], self.max_len - 3))
             self._capture()
         self._prefill(prompt_ids)
         # first generated token came from prefill logits
