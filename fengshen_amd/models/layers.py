@@ -439,3 +439,62 @@ class GMLPBlock(nn.Module):
         h = F_ops.eager_gelu(self.fc_in(h).float()).to(h.dtype)
         h = self.sgu(h)
         return x + self.fc_out(h)
+
+
+def expand_attention_types(attention_config, num_layers: int):
+    """Expand a [[types, repeat], ...] spec into a per-layer type list
+    (ref models/llama/modeling_llama.py:37-64 semantics): each entry
+    contributes `repeat` layers cycling through its `types`; "all" in a
+    spec means the same types tile the whole depth.
+
+    >>> expand_attention_types([[["global"], 2], [["sparse_fixed"], 2]], 4)
+    ['global', 'global', 'sparse_fixed', 'sparse_fixed']
+    """
+    if attention_config is None:
+        return ["global"] * num_layers
+    out = []
+    for types, repeat in attention_config:
+        if repeat == "all":
+            for i in range(num_layers):
+                out.append(types[i % len(types)])
+            break
+        for i in range(int(repeat)):
+            out.append(types[i % len(types)])
+    assert len(out) >= num_layers, (
+        f"attention_config covers {len(out)} layers, model has {num_layers}")
+    return out[:num_layers]
+
+
+def get_ltor_masks_and_position_ids(data: torch.Tensor, eod_token: int,
+                                    reset_position_ids: bool = False,
+                                    reset_attention_mask: bool = False,
+                                    eod_mask_loss: bool = False):
+    """Left-to-right (causal) masks + position ids for packed GPT batches
+    (ref layers/utils.py:38 get_ltor_masks_and_position_ids).
+
+    Returns (attention_mask [b,1,s,s] bool — True where MASKED, the
+    megatron convention — loss_mask [b,s] float, position_ids [b,s]).
+    reset_attention_mask blocks attention across EOD boundaries;
+    reset_position_ids restarts positions after each EOD; eod_mask_loss
+    zeroes the loss at EOD tokens.
+    """
+    b, s = data.shape
+    att = torch.tril(torch.ones(s, s, device=data.device)) \
+        .unsqueeze(0).repeat(b, 1, 1)
+    loss_mask = torch.ones(b, s, dtype=torch.float, device=data.device)
+    if eod_mask_loss:
+        loss_mask[data == eod_token] = 0.0
+    position_ids = torch.arange(s, device=data.device) \
+        .unsqueeze(0).repeat(b, 1)
+    if reset_position_ids or reset_attention_mask:
+        position_ids = position_ids.clone()
+        for bi in range(b):
+            eod_idx = position_ids[bi, data[bi] == eod_token]
+            prev = 0
+            for j in eod_idx.tolist():
+                if reset_attention_mask:
+                    att[bi, j + 1:, : j + 1] = 0
+                if reset_position_ids:
+                    position_ids[bi, j + 1:] -= j + 1 - prev
+                    prev = j + 1
+    return (att < 0.5).unsqueeze(1), loss_mask, position_ids

@@ -312,3 +312,37 @@ def test_encoder_decoder_layers_direct():
     assert y2.shape == (2, 5, H)
     y2.sum().backward()
     assert dl.cross_attn.q_proj.weight.grad is not None
+
+
+def test_expand_attention_types_and_ltor_masks():
+    """Per-layer attention-config expansion + packed-batch EOD masks
+    (ref modeling_llama.py:37-64, layers/utils.py:38)."""
+    import torch
+    from fengshen_amd.models.layers import (
+        expand_attention_types, get_ltor_masks_and_position_ids)
+
+    assert expand_attention_types(None, 3) == ["global"] * 3
+    assert expand_attention_types(
+        [[["global"], 2], [["sparse_fixed"], 2]], 4) == \
+        ["global", "global", "sparse_fixed", "sparse_fixed"]
+    assert expand_attention_types(
+        [[["a", "b"], "all"]], 5) == ["a", "b", "a", "b", "a"]
+
+    data = torch.tensor([[5, 6, 0, 7, 8]])  # EOD token = 0 at index 2
+    att, loss, pos = get_ltor_masks_and_position_ids(
+        data, eod_token=0, reset_position_ids=True,
+        reset_attention_mask=True, eod_mask_loss=True)
+    assert att.shape == (1, 1, 5, 5) and att.dtype == torch.bool
+    # causal: future masked
+    assert bool(att[0, 0, 0, 1])
+    # cross-EOD attention blocked: token 3 cannot see tokens 0..2
+    assert bool(att[0, 0, 3, 0]) and bool(att[0, 0, 3, 2])
+    assert not bool(att[0, 0, 3, 3])
+    # positions restart after EOD
+    assert pos[0].tolist() == [0, 1, 2, 0, 1]
+    # loss masked at EOD only
+    assert loss[0].tolist() == [1.0, 1.0, 0.0, 1.0, 1.0]
+    # plain mode: pure causal
+    att2, loss2, pos2 = get_ltor_masks_and_position_ids(data, 0)
+    assert pos2[0].tolist() == [0, 1, 2, 3, 4]
+    assert loss2.sum() == 5 and not bool(att2[0, 0, 4, 0])
