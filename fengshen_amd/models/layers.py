@@ -140,9 +140,10 @@ class ParallelAttention(nn.Module):
                  attention_dropout: float = 0.0, hidden_dropout: float = 0.0,
                  bias: bool = True, init_method=None, output_init_method=None,
                  dtype=None, num_kv_heads: Optional[int] = None,
-                 layer_idx: int = 0):
+                 layer_idx: int = 0, reduce_output: bool = True):
         super().__init__()
         self.layer_idx = layer_idx
+        self.reduce_output = reduce_output
         tp = groups.get_tensor_model_parallel_world_size()
         self.hidden_size = hidden_size
         self.num_heads = num_heads
@@ -163,7 +164,8 @@ class ParallelAttention(nn.Module):
             init_method=init_method, dtype=dtype)
         self.out_proj = RowParallelLinear(
             hidden_size, hidden_size, bias=bias, input_is_parallel=True,
-            init_method=output_init_method, dtype=dtype)
+            init_method=output_init_method, dtype=dtype,
+            reduce_output=reduce_output)
 
         if rotary:
             # lazy host-precomputed cos/sin tables (guide App. B: no on-device
@@ -219,9 +221,9 @@ class ParallelAttention(nn.Module):
                               training=self.training, scale=self.norm_factor)
         ctx = ctx.transpose(1, 2).reshape(b, sq, np_ * hn)
         out = self.out_proj(ctx)
-        if isinstance(out, tuple):
+        if isinstance(out, tuple) and self.reduce_output:
             out = out[0]
-        return out
+        return out  # (partial, bias) when reduce_output=False
 
 
 class ParallelMLP(nn.Module):
@@ -229,16 +231,18 @@ class ParallelMLP(nn.Module):
 
     def __init__(self, hidden_size: int, ffn_hidden_size: Optional[int] = None,
                  bias: bool = True, init_method=None, output_init_method=None,
-                 dtype=None):
+                 dtype=None, reduce_output: bool = True):
         super().__init__()
         ffn = ffn_hidden_size or 4 * hidden_size
         init_method = init_method or init.xavier_normal_
+        self.reduce_output = reduce_output
         self.fc_in = ColumnParallelLinear(
             hidden_size, ffn, bias=bias, gather_output=False,
             init_method=init_method, skip_bias_add=bias, dtype=dtype)
         self.fc_out = RowParallelLinear(
             ffn, hidden_size, bias=bias, input_is_parallel=True,
-            init_method=output_init_method or init_method, dtype=dtype)
+            init_method=output_init_method or init_method, dtype=dtype,
+            reduce_output=reduce_output)
 
     def forward(self, x):
         h = self.fc_in(x)
@@ -248,9 +252,9 @@ class ParallelMLP(nn.Module):
         else:
             h = F_ops.eager_gelu(h.float()).to(h.dtype)
         out = self.fc_out(h)
-        if isinstance(out, tuple):
+        if isinstance(out, tuple) and self.reduce_output:
             out = out[0]
-        return out
+        return out  # (partial, bias) when reduce_output=False
 
 
 class LLaMAParallelMLP(nn.Module):
@@ -259,8 +263,10 @@ class LLaMAParallelMLP(nn.Module):
 
     def __init__(self, hidden_size: int, intermediate_size: Optional[int] = None,
                  multiple_of: int = 256, init_method=None,
-                 output_init_method=None, dtype=None):
+                 output_init_method=None, dtype=None,
+                 reduce_output: bool = True):
         super().__init__()
+        self.reduce_output = reduce_output
         if intermediate_size is None:
             # ref :589-590 rounding
             intermediate_size = int(2 * (4 * hidden_size) / 3)
@@ -273,12 +279,13 @@ class LLaMAParallelMLP(nn.Module):
             init_method=init_method, dtype=dtype)
         self.down_proj = RowParallelLinear(
             intermediate_size, hidden_size, bias=False, input_is_parallel=True,
-            init_method=output_init_method or init_method, dtype=dtype)
+            init_method=output_init_method or init_method, dtype=dtype,
+            reduce_output=reduce_output)
 
     def forward(self, x):
         packed = self.gate_up_proj(x)
         h = F_ops.swiglu(packed)
-        return self.down_proj(h)
+        return self.down_proj(h)  # (partial, bias) when reduce_output=False
 
 
 class ParallelTransformerLayer(nn.Module):
@@ -292,8 +299,13 @@ class ParallelTransformerLayer(nn.Module):
                  max_positions: int = 4096,
                  attention_dropout: float = 0.0, hidden_dropout: float = 0.0,
                  bias: bool = True, init_method=None, output_init_method=None,
-                 dtype=None, layer_idx: int = 0):
+                 dtype=None, layer_idx: int = 0,
+                 parallel_residual: bool = False):
         super().__init__()
+        # parallel_residual: GPT-J composition x + attn(ln1 x) + mlp(ln2 x)
+        # with ONE deferred TP all-reduce over the summed partials
+        # (ref transformer.py:710-752) — halves the per-layer TP comms
+        self.parallel_residual = parallel_residual
         self.input_norm = get_norm(norm, hidden_size, norm_eps, dtype)
         self.attention = ParallelAttention(
             hidden_size, num_heads, causal=causal, rotary=rotary,
@@ -301,19 +313,42 @@ class ParallelTransformerLayer(nn.Module):
             attention_dropout=attention_dropout, hidden_dropout=hidden_dropout,
             bias=bias, init_method=init_method,
             output_init_method=output_init_method, dtype=dtype,
-            layer_idx=layer_idx)
+            layer_idx=layer_idx, reduce_output=not parallel_residual)
         self.post_attention_norm = get_norm(norm, hidden_size, norm_eps, dtype)
         if mlp_type == "swiglu":
             self.mlp = LLaMAParallelMLP(
                 hidden_size, ffn_hidden_size, init_method=init_method,
-                output_init_method=output_init_method, dtype=dtype)
+                output_init_method=output_init_method, dtype=dtype,
+                reduce_output=not parallel_residual)
         else:
             self.mlp = ParallelMLP(
                 hidden_size, ffn_hidden_size, bias=bias, init_method=init_method,
-                output_init_method=output_init_method, dtype=dtype)
+                output_init_method=output_init_method, dtype=dtype,
+                reduce_output=not parallel_residual)
         self.hidden_dropout = hidden_dropout
 
+    @staticmethod
+    def _split_partial(out):
+        if isinstance(out, tuple):
+            return out[0], out[1]
+        return out, None
+
     def forward(self, x, attention_mask=None, cache=None):
+        if self.parallel_residual:
+            from fengshen_amd.parallel.mappings import (
+                reduce_from_tensor_model_parallel_region)
+            ap, ab = self._split_partial(self.attention(
+                self.input_norm(x), attention_mask=attention_mask,
+                cache=cache))
+            mp, mb = self._split_partial(
+                self.mlp(self.post_attention_norm(x)))
+            total = reduce_from_tensor_model_parallel_region(ap + mp)
+            bias = None
+            if ab is not None or mb is not None:
+                bias = (ab if ab is not None else 0)                     + (mb if mb is not None else 0)
+            return F_ops.bias_dropout_add(total, bias, x,
+                                          self.hidden_dropout,
+                                          self.training)
         residual = x
         h = self.input_norm(x)
         attn_out = self.attention(h, attention_mask=attention_mask, cache=cache)

@@ -157,12 +157,18 @@ class RowParallelLinear(nn.Module):
                  input_is_parallel: bool = False,
                  init_method: Callable = init.xavier_normal_,
                  stride: int = 1, skip_bias_add: bool = False,
-                 dtype: Optional[torch.dtype] = None):
+                 dtype: Optional[torch.dtype] = None,
+                 reduce_output: bool = True):
         super().__init__()
         self.input_size = input_size
         self.output_size = output_size
         self.input_is_parallel = input_is_parallel
         self.skip_bias_add = skip_bias_add
+        # reduce_output=False returns the rank-local partial sum so a
+        # caller can DEFER the TP all-reduce and reduce several branches
+        # once (the reference's GPT-J parallel-residual single deferred
+        # all-reduce, transformer.py:710-752)
+        self.reduce_output = reduce_output
         tp = groups.get_tensor_model_parallel_world_size()
         self.input_size_per_partition = divide(input_size, tp)
         dtype = dtype or torch.get_default_dtype()
@@ -182,6 +188,12 @@ class RowParallelLinear(nn.Module):
         input_parallel = (input_ if self.input_is_parallel
                           else scatter_to_tensor_model_parallel_region(input_))
         output_parallel = F_ops.linear(input_parallel, self.weight)
+        if not self.reduce_output:
+            # partial sum: the caller owns the (deferred) all-reduce;
+            # bias must be added after that reduce, so hand it back
+            if self.skip_bias_add or self.bias is not None:
+                return output_parallel, self.bias
+            return output_parallel
         output_ = reduce_from_tensor_model_parallel_region(output_parallel)
         if self.skip_bias_add:
             return output_, self.bias

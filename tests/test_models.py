@@ -346,3 +346,30 @@ def test_expand_attention_types_and_ltor_masks():
     att2, loss2, pos2 = get_ltor_masks_and_position_ids(data, 0)
     assert pos2[0].tolist() == [0, 1, 2, 3, 4]
     assert loss2.sum() == 5 and not bool(att2[0, 0, 4, 0])
+
+
+def test_parallel_residual_layer_tp1():
+    """GPT-J parallel residual (ref transformer.py:710-752): output is
+    x + attn(ln1 x) + mlp(ln2 x), for both gelu and swiglu MLPs."""
+    import torch
+    from fengshen_amd.models.layers import ParallelTransformerLayer
+    torch.manual_seed(0)
+    for mlp_type, norm in [("gelu", "layernorm"), ("swiglu", "rmsnorm")]:
+        layer = ParallelTransformerLayer(
+            32, 4, causal=True, mlp_type=mlp_type, norm=norm,
+            parallel_residual=True).eval()
+        x = torch.randn(2, 6, 32)
+        y = layer(x)
+        with torch.no_grad():
+            a = layer.attention(layer.input_norm(x))
+            a, ab = a if isinstance(a, tuple) else (a, None)
+            m = layer.mlp(layer.post_attention_norm(x))
+            m, mb = m if isinstance(m, tuple) else (m, None)
+            ref = x + a + m
+            if ab is not None:
+                ref = ref + ab
+            if mb is not None:
+                ref = ref + mb
+        assert (y - ref).abs().max() < 1e-5, mlp_type
+        y.sum().backward()
+        assert layer.attention.qkv_proj.weight.grad is not None
