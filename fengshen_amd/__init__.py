@@ -24,3 +24,37 @@ from fengshen_amd.trainer.module import FengshenModule  # noqa: F401
 from fengshen_amd.trainer.trainer import Trainer  # noqa: F401
 from fengshen_amd.data.universal_datamodule import UniversalDataModule  # noqa: F401
 from fengshen_amd.utils.universal_checkpoint import UniversalCheckpoint  # noqa: F401
+
+# Reference-style top-level model exports (the reference README documents
+# `from fengshen import LongformerModel` etc.) — lazy so importing
+# fengshen_amd stays light.
+_LAZY_EXPORTS = {
+    "LongformerModel": "fengshen_amd.models.longformer.modeling_longformer",
+    "LongformerConfig": "fengshen_amd.models.longformer.modeling_longformer",
+    "LongformerForMaskedLM":
+        "fengshen_amd.models.longformer.modeling_longformer",
+    "RoFormerModel": "fengshen_amd.models.roformer.modeling_roformer",
+    "RoFormerConfig": "fengshen_amd.models.roformer.modeling_roformer",
+    "LlamaForCausalLM": "fengshen_amd.models.llama.modeling_llama",
+    "LlamaConfig": "fengshen_amd.models.llama.configuration_llama",
+    "MegatronBertModel":
+        "fengshen_amd.models.megatron_bert.modeling_megatron_bert",
+    "MegatronBertForPreTraining":
+        "fengshen_amd.models.megatron_bert.modeling_megatron_bert",
+    "GPT2LMHeadModel": "fengshen_amd.models.gpt2.modeling_gpt2",
+    "T5ForConditionalGeneration": "fengshen_amd.models.t5.modeling_t5",
+    "UbertModel": "fengshen_amd.models.ubert.modeling_ubert",
+    "UniMCModel": "fengshen_amd.models.unimc.modeling_unimc",
+}
+
+
+def __getattr__(name):
+    mod = _LAZY_EXPORTS.get(name)
+    if mod is not None:
+        import importlib
+        return getattr(importlib.import_module(mod), name)
+    raise AttributeError(f"module 'fengshen_amd' has no attribute {name!r}")
+
+
+def __dir__():
+    return sorted(list(globals()) + list(_LAZY_EXPORTS))

@@ -373,3 +373,15 @@ def test_parallel_residual_layer_tp1():
         assert (y - ref).abs().max() < 1e-5, mlp_type
         y.sum().backward()
         assert layer.attention.qkv_proj.weight.grad is not None
+
+
+def test_top_level_lazy_exports():
+    """Reference README style: `from fengshen import LongformerModel`."""
+    import fengshen_amd
+    from fengshen_amd import (LlamaForCausalLM, LongformerConfig,
+                              LongformerModel, RoFormerModel)
+    assert LongformerModel.__name__ == "LongformerModel"
+    assert "UniMCModel" in dir(fengshen_amd)
+    import pytest
+    with pytest.raises(AttributeError):
+        fengshen_amd.NoSuchModel
