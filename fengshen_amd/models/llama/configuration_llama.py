@@ -27,8 +27,11 @@ class LlamaConfig(PretrainedConfig):
                  bos_token_id: int = 1,
                  eos_token_id: int = 2,
                  tie_word_embeddings: bool = False,
+                 parallel_residual: bool = False,
                  torch_dtype="bfloat16",
                  **kwargs):
+        # GPT-J composition with one deferred TP all-reduce per layer
+        self.parallel_residual = parallel_residual
         self.vocab_size = vocab_size
         self.hidden_size = hidden_size
         self.num_hidden_layers = num_hidden_layers

@@ -66,7 +66,9 @@ class LlamaModel(LlamaPreTrainedModel):
                 attention_dropout=config.attention_dropout,
                 hidden_dropout=config.hidden_dropout,
                 bias=False, init_method=im, output_init_method=om,
-                layer_idx=i)
+                layer_idx=i,
+                parallel_residual=getattr(config, "parallel_residual",
+                                          False))
             for i in range(config.num_hidden_layers)])
         self.norm = RMSNorm(config.hidden_size, eps=config.rms_norm_epsilon)
         self.gradient_checkpointing = False

@@ -385,3 +385,22 @@ def test_top_level_lazy_exports():
     import pytest
     with pytest.raises(AttributeError):
         fengshen_amd.NoSuchModel
+
+
+def test_llama_parallel_residual_config():
+    """config.parallel_residual=True builds GPT-J-composition layers and
+    trains; default stays sequential pre-LN."""
+    import torch
+    from fengshen_amd.models.llama.configuration_llama import (
+        llama_tiny_config)
+    from fengshen_amd.models.llama.modeling_llama import LlamaForCausalLM
+    torch.manual_seed(0)
+    cfg = llama_tiny_config(parallel_residual=True)
+    m = LlamaForCausalLM(cfg)
+    assert m.model.layers[0].parallel_residual
+    ids = torch.randint(3, cfg.vocab_size, (2, 16))
+    out = m(ids, labels=ids)
+    assert out.loss.isfinite()
+    out.loss.backward()
+    m2 = LlamaForCausalLM(llama_tiny_config())
+    assert not m2.model.layers[0].parallel_residual
