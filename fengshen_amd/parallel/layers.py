@@ -9,6 +9,7 @@ from __future__ import annotations
 
 from typing import Callable, Optional
 
+import os
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -92,6 +93,14 @@ class VocabParallelEmbedding(nn.Module):
                         full_shape=(num_embeddings, embedding_dim), partition_dim=0)
 
     def forward(self, input_: torch.Tensor) -> torch.Tensor:
+        if os.environ.get("FENGSHEN_CHECK_IDS") == "1":
+            # opt-in host sync: an out-of-range id otherwise surfaces as
+            # an opaque HSA_STATUS_ERROR_EXCEPTION device fault on ROCm
+            mx, mn = int(input_.max()), int(input_.min())
+            if mx >= self.num_embeddings or mn < 0:
+                raise IndexError(
+                    f"token id range [{mn}, {mx}] outside vocab "
+                    f"[0, {self.num_embeddings})")
         tp = groups.get_tensor_model_parallel_world_size()
         if tp > 1:
             input_mask = (input_ < self.vocab_start_index) | (input_ >= self.vocab_end_index)

@@ -250,3 +250,17 @@ def _parallel_residual_worker(rank, world_size):
 def test_parallel_residual_deferred_reduce_tp2():
     res = run_distributed(_parallel_residual_worker, world_size=2)
     assert all(v < 1e-5 for v in res)
+
+
+def test_vocab_embedding_id_range_check(monkeypatch):
+    """FENGSHEN_CHECK_IDS=1 turns ROCm's opaque device fault on
+    out-of-range ids into a clear IndexError."""
+    import pytest
+    from fengshen_amd.parallel.layers import VocabParallelEmbedding
+    monkeypatch.setenv("FENGSHEN_CHECK_IDS", "1")
+    emb = VocabParallelEmbedding(50, 8)
+    assert emb(torch.tensor([[0, 49]])).shape == (1, 2, 8)
+    with pytest.raises(IndexError, match="outside vocab"):
+        emb(torch.tensor([[50]]))
+    with pytest.raises(IndexError):
+        emb(torch.tensor([[-1]]))
