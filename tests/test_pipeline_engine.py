@@ -88,3 +88,9 @@ def test_pipeline_split_contiguous():
     s1 = split_module_for_pipeline(layers, 3, 1)
     s2 = split_module_for_pipeline(layers, 3, 2)
     assert len(s0) == 3 and len(s1) == 3 and len(s2) == 1
+
+
+def test_pipeline_1f1b_world4():
+    """4-stage 1F1B (the driver's N=4 shape): gradient-equivalence vs
+    single-process reference, warmup depth 3 on stage 0."""
+    _check(4)
