@@ -94,3 +94,88 @@ def test_pipeline_1f1b_world4():
     """4-stage 1F1B (the driver's N=4 shape): gradient-equivalence vs
     single-process reference, warmup depth 3 on stage 0."""
     _check(4)
+
+
+class _TPBlock(torch.nn.Module):
+    """Column(d->2d, gelu) -> Row(2d->d): the minimal TP layer pair."""
+
+    def __init__(self, d):
+        super().__init__()
+        from fengshen_amd.parallel.layers import (
+            ColumnParallelLinear, RowParallelLinear)
+        self.col = ColumnParallelLinear(d, 2 * d, bias=True,
+                                        gather_output=False,
+                                        dtype=torch.float32)
+        self.row = RowParallelLinear(2 * d, d, bias=True,
+                                     input_is_parallel=True,
+                                     dtype=torch.float32)
+
+    def forward(self, x):
+        return self.row(torch.nn.functional.gelu(self.col(x)))
+
+
+def _tp_pp_worker(rank, world_size):
+    """world 4 = tp2 x pp2: TP blocks inside pipeline stages must match
+    the dense single-process model's loss and (sharded) gradients."""
+    import torch.distributed as dist
+    from fengshen_amd.parallel import groups
+    from fengshen_amd.parallel.groups import (
+        init_distributed, initialize_model_parallel)
+    from fengshen_amd.parallel.pipeline import PipelineEngine
+
+    d, num_micro = 16, 2
+    init_distributed(backend="gloo")
+    initialize_model_parallel(2, pipeline_model_parallel_size=2)
+    tp_rank = groups.get_tensor_model_parallel_rank()
+    pp_rank = groups.get_pipeline_model_parallel_rank()
+
+    # dense reference weights, identical on every rank
+    torch.manual_seed(21)
+    ref = [torch.nn.ModuleDict({
+        "col": torch.nn.Linear(d, 2 * d), "row": torch.nn.Linear(2 * d, d)})
+        for _ in range(2)]
+    x, y = _data(b=4, d=d, seed=23)
+
+    blk = _TPBlock(d)
+    r = ref[pp_rank]
+    per = 2 * d // 2  # column shard rows / row shard cols
+    with torch.no_grad():
+        blk.col.weight.copy_(r["col"].weight[tp_rank * per:(tp_rank + 1) * per])
+        blk.col.bias.copy_(r["col"].bias[tp_rank * per:(tp_rank + 1) * per])
+        blk.row.weight.copy_(r["row"].weight[:, tp_rank * per:(tp_rank + 1) * per])
+        blk.row.bias.copy_(r["row"].bias)
+
+    eng = PipelineEngine(
+        blk, lambda out, tg: torch.nn.functional.mse_loss(out, tg),
+        num_microbatches=num_micro, act_shape=(x.shape[0] // num_micro, d),
+        act_dtype=torch.float32)
+    mbs = list(x.chunk(num_micro)) if pp_rank == 0 else None
+    tgs = list(y.chunk(num_micro)) if pp_rank == 1 else None
+    loss = eng.train_batch(mbs, tgs)
+
+    # dense reference forward/backward
+    def dense(x):
+        for m in ref:
+            x = m["row"](torch.nn.functional.gelu(m["col"](x)))
+        return x
+    total = 0.0
+    for mb, tg in zip(x.chunk(num_micro), y.chunk(num_micro)):
+        l = torch.nn.functional.mse_loss(dense(mb), tg)
+        (l / num_micro).backward()
+        total += float(l) / num_micro
+    rg = ref[pp_rank]
+    col_g = rg["col"].weight.grad[tp_rank * per:(tp_rank + 1) * per]
+    row_g = rg["row"].weight.grad[:, tp_rank * per:(tp_rank + 1) * per]
+    err = max(float((blk.col.weight.grad - col_g).abs().max()),
+              float((blk.row.weight.grad - row_g).abs().max()))
+    dist.destroy_process_group()
+    return {"pp": pp_rank, "tp": tp_rank, "loss": loss, "ref_loss": total,
+            "err": err}
+
+
+def test_pipeline_tp2_pp2_grads():
+    results = run_distributed(_tp_pp_worker, world_size=4)
+    for r in results:
+        assert r["err"] < 1e-5, r
+        if r["loss"] is not None:  # last pp stage reports loss
+            assert abs(r["loss"] - r["ref_loss"]) < 1e-5, r
