@@ -179,3 +179,54 @@ def test_pipeline_tp2_pp2_grads():
         assert r["err"] < 1e-5, r
         if r["loss"] is not None:  # last pp stage reports loss
             assert abs(r["loss"] - r["ref_loss"]) < 1e-5, r
+
+
+def _dp_pp_worker(rank, world_size):
+    """world 4 = dp2 x pp2: pipeline stages wrapped in the DP GradReducer
+    must produce grads equal to the dense model over the FULL batch."""
+    import torch.distributed as dist
+    from fengshen_amd.parallel import groups
+    from fengshen_amd.parallel.ddp import GradReducer
+    from fengshen_amd.parallel.groups import (
+        init_distributed, initialize_model_parallel)
+    from fengshen_amd.parallel.pipeline import (
+        PipelineEngine, split_module_for_pipeline)
+
+    num_micro = 2
+    init_distributed(backend="gloo")
+    initialize_model_parallel(1, pipeline_model_parallel_size=2)
+    dp_rank = groups.get_data_parallel_rank()
+    pp_rank = groups.get_pipeline_model_parallel_rank()
+
+    layers = _layers(n=4, seed=31)
+    stage = split_module_for_pipeline(layers, 2, pp_rank)
+    reducer = GradReducer(stage, process_group=groups.get_data_parallel_group())
+    x, y = _data(b=8, seed=33)
+    # each DP replica consumes its half of the batch
+    xr, yr = x.chunk(2)[dp_rank], y.chunk(2)[dp_rank]
+    eng = PipelineEngine(
+        stage, lambda out, tg: torch.nn.functional.mse_loss(out, tg),
+        num_microbatches=num_micro, act_shape=(xr.shape[0] // num_micro, 16),
+        act_dtype=torch.float32)
+    reducer.set_sync(False)  # accumulate across microbatches
+    eng.train_batch(list(xr.chunk(num_micro)) if pp_rank == 0 else None,
+                    list(yr.chunk(num_micro)) if pp_rank == 1 else None)
+    reducer.set_sync(True)
+    reducer.finalize()
+
+    # dense reference over the FULL batch (mean of the two replicas)
+    model = torch.nn.Sequential(*_layers(n=4, seed=31))
+    for mb, tg in zip(x.chunk(2 * num_micro), y.chunk(2 * num_micro)):
+        (torch.nn.functional.mse_loss(model(mb), tg) / num_micro / 2).backward()
+    ref = torch.nn.Sequential(*split_module_for_pipeline(
+        list(model), 2, pp_rank))
+    err = max(float((p.grad - q.grad).abs().max())
+              for p, q in zip(stage.parameters(), ref.parameters()))
+    dist.destroy_process_group()
+    return {"pp": pp_rank, "dp": dp_rank, "err": err}
+
+
+def test_pipeline_dp2_pp2_grad_average():
+    results = run_distributed(_dp_pp_worker, world_size=4)
+    for r in results:
+        assert r["err"] < 1e-5, r
