@@ -230,3 +230,71 @@ def test_pipeline_dp2_pp2_grad_average():
     results = run_distributed(_dp_pp_worker, world_size=4)
     for r in results:
         assert r["err"] < 1e-5, r
+
+
+def _dp_tp_pp_worker(rank, world_size):
+    """world 8 = dp2 x tp2 x pp2 — the full 3D composition: TP blocks
+    inside pipeline stages, DP gradient averaging across replicas, all
+    exact vs the dense full-batch reference."""
+    import torch.distributed as dist
+    from fengshen_amd.parallel import groups
+    from fengshen_amd.parallel.ddp import GradReducer
+    from fengshen_amd.parallel.groups import (
+        init_distributed, initialize_model_parallel)
+    from fengshen_amd.parallel.pipeline import PipelineEngine
+
+    d, num_micro = 16, 2
+    init_distributed(backend="gloo")
+    initialize_model_parallel(2, pipeline_model_parallel_size=2)
+    tp_rank = groups.get_tensor_model_parallel_rank()
+    pp_rank = groups.get_pipeline_model_parallel_rank()
+    dp_rank = groups.get_data_parallel_rank()
+
+    torch.manual_seed(41)
+    ref = [torch.nn.ModuleDict({
+        "col": torch.nn.Linear(d, 2 * d), "row": torch.nn.Linear(2 * d, d)})
+        for _ in range(2)]
+    x, y = _data(b=8, d=d, seed=43)
+
+    blk = _TPBlock(d)
+    r = ref[pp_rank]
+    per = d  # 2d // tp2
+    with torch.no_grad():
+        blk.col.weight.copy_(r["col"].weight[tp_rank * per:(tp_rank + 1) * per])
+        blk.col.bias.copy_(r["col"].bias[tp_rank * per:(tp_rank + 1) * per])
+        blk.row.weight.copy_(r["row"].weight[:, tp_rank * per:(tp_rank + 1) * per])
+        blk.row.bias.copy_(r["row"].bias)
+    reducer = GradReducer(blk, process_group=groups.get_data_parallel_group())
+
+    xr, yr = x.chunk(2)[dp_rank], y.chunk(2)[dp_rank]
+    eng = PipelineEngine(
+        blk, lambda out, tg: torch.nn.functional.mse_loss(out, tg),
+        num_microbatches=num_micro, act_shape=(xr.shape[0] // num_micro, d),
+        act_dtype=torch.float32)
+    reducer.set_sync(False)
+    eng.train_batch(list(xr.chunk(num_micro)) if pp_rank == 0 else None,
+                    list(yr.chunk(num_micro)) if pp_rank == 1 else None)
+    reducer.set_sync(True)
+    reducer.finalize()
+
+    def dense(x):
+        for m in ref:
+            x = m["row"](torch.nn.functional.gelu(m["col"](x)))
+        return x
+    for mb, tg in zip(x.chunk(2 * num_micro), y.chunk(2 * num_micro)):
+        (torch.nn.functional.mse_loss(dense(mb), tg) / num_micro / 2).backward()
+    rg = ref[pp_rank]
+    col_g = rg["col"].weight.grad[tp_rank * per:(tp_rank + 1) * per]
+    row_g = rg["row"].weight.grad[:, tp_rank * per:(tp_rank + 1) * per]
+    err = max(float((blk.col.weight.grad - col_g).abs().max()),
+              float((blk.row.weight.grad - row_g).abs().max()),
+              float((blk.row.bias.grad - rg["row"].bias.grad).abs().max()))
+    dist.destroy_process_group()
+    return {"dp": dp_rank, "tp": tp_rank, "pp": pp_rank, "err": err}
+
+
+def test_pipeline_3d_dp2_tp2_pp2():
+    results = run_distributed(_dp_tp_pp_worker, world_size=8)
+    assert len(results) == 8
+    for r in results:
+        assert r["err"] < 1e-5, r
