@@ -233,7 +233,16 @@ class ZeroOptimizer(torch.optim.Optimizer):
 
     # ------------------------------------------------------------------
     def set_sync(self, flag: bool):
-        """False during gradient-accumulation micro-steps (no comm)."""
+        """False during gradient-accumulation micro-steps (no comm).
+
+        This is a hard contract, not an optimization: with sync on and
+        overlap_comm, the backward hook launches the bucket's async
+        reduction as soon as every param has accumulated ONCE — a later
+        micro-backward would then race the in-flight collective on the
+        same flat grad buffer.  step() launches any pending reductions
+        itself (trailing flush), so the pattern is:
+        set_sync(False) -> N micro-backwards -> set_sync(True) -> step().
+        """
         self._sync_grads = flag
 
     @torch.no_grad()
