@@ -23,5 +23,6 @@ setup(
         ],
     },
     package_data={"fengshen_amd": ["ops/_C.so", "data/_helpers.so",
-                                   "ops/csrc/*", "data/csrc/*"]},
+                                   "ops/csrc/*", "data/csrc/*",
+                                   "workspace/*/*"]},
 )
